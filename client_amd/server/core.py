@@ -1,0 +1,456 @@
+"""Protocol-agnostic inference core for the client_amd server.
+
+Shared by the HTTP and gRPC frontends: model repository, shared-memory
+registry (system POSIX shm + HIP-IPC device regions), request execution
+with per-model statistics in the KServe-v2 stats schema
+(reference schema: grpc_service.proto:881-1235).
+"""
+
+import threading
+import time
+
+import numpy as np
+
+from ..utils import (
+    deserialize_bytes_tensor,
+    deserialize_bf16_tensor,
+    serialize_byte_tensor,
+    serialize_bf16_tensor,
+    triton_to_np_dtype,
+    np_to_triton_dtype,
+    TRITON_DTYPE_SIZES,
+)
+
+
+class ShmRegistry:
+    """Server-side registry of client-shared memory regions.
+
+    System regions are mapped via POSIX shm_open/mmap; HIP regions are
+    opened from the 64-byte hipIpcMemHandle_t with hipIpcOpenMemHandle
+    (via client_amd.ops) and stay resident in HBM — tensors never cross
+    PCIe.
+    """
+
+    def __init__(self):
+        self._system = {}
+        self._hip = {}
+        self._lock = threading.Lock()
+
+    # ---- system shm ----
+    def register_system(self, name, key, offset, byte_size):
+        import mmap
+        import os
+
+        fd = os.open("/dev/shm/" + key.lstrip("/"), os.O_RDWR)
+        try:
+            mem = mmap.mmap(fd, byte_size + offset)
+        finally:
+            os.close(fd)
+        with self._lock:
+            if name in self._system:
+                self._system.pop(name)
+            self._system[name] = {
+                "name": name,
+                "key": key,
+                "offset": offset,
+                "byte_size": byte_size,
+                "mmap": mem,
+            }
+
+    def unregister_system(self, name=None):
+        with self._lock:
+            names = [name] if name else list(self._system)
+            for n in names:
+                region = self._system.pop(n, None)
+                if region is not None:
+                    region["mmap"].close()
+
+    def system_status(self, name=None):
+        with self._lock:
+            regions = (
+                [self._system[name]] if name and name in self._system
+                else list(self._system.values()) if not name else []
+            )
+            return [
+                {"name": r["name"], "key": r["key"], "offset": r["offset"],
+                 "byte_size": r["byte_size"]}
+                for r in regions
+            ]
+
+    # ---- HIP shm ----
+    def register_hip(self, name, raw_handle, device_id, byte_size):
+        """raw_handle: 64 bytes of hipIpcMemHandle_t."""
+        from ..ops import hip_runtime as hr
+
+        ptr = hr.ipc_open_mem_handle(raw_handle)
+        with self._lock:
+            if name in self._hip:
+                old = self._hip.pop(name)
+                hr.ipc_close_mem_handle(old["ptr"])
+            self._hip[name] = {
+                "name": name,
+                "device_id": device_id,
+                "byte_size": byte_size,
+                "ptr": ptr,
+            }
+
+    def unregister_hip(self, name=None):
+        from ..ops import hip_runtime as hr
+
+        with self._lock:
+            names = [name] if name else list(self._hip)
+            for n in names:
+                region = self._hip.pop(n, None)
+                if region is not None:
+                    hr.ipc_close_mem_handle(region["ptr"])
+
+    def hip_status(self, name=None):
+        with self._lock:
+            regions = (
+                [self._hip[name]] if name and name in self._hip
+                else list(self._hip.values()) if not name else []
+            )
+            return [
+                {"name": r["name"], "device_id": r["device_id"],
+                 "byte_size": r["byte_size"]}
+                for r in regions
+            ]
+
+    def get_system(self, name):
+        with self._lock:
+            return self._system.get(name)
+
+    def get_hip(self, name):
+        with self._lock:
+            return self._hip.get(name)
+
+
+class ModelStats:
+    def __init__(self):
+        self.inference_count = 0
+        self.execution_count = 0
+        self.last_inference = 0
+        self.success_count = 0
+        self.success_ns = 0
+        self.fail_count = 0
+        self.fail_ns = 0
+        self.queue_ns = 0
+        self.compute_input_ns = 0
+        self.compute_infer_ns = 0
+        self.compute_output_ns = 0
+
+    def to_dict(self, name, version="1"):
+        def d(count, ns):
+            return {"count": count, "ns": ns}
+
+        return {
+            "name": name,
+            "version": version,
+            "last_inference": self.last_inference,
+            "inference_count": self.inference_count,
+            "execution_count": self.execution_count,
+            "inference_stats": {
+                "success": d(self.success_count, self.success_ns),
+                "fail": d(self.fail_count, self.fail_ns),
+                "queue": d(self.success_count, self.queue_ns),
+                "compute_input": d(self.success_count, self.compute_input_ns),
+                "compute_infer": d(self.success_count, self.compute_infer_ns),
+                "compute_output": d(self.success_count, self.compute_output_ns),
+                "cache_hit": d(0, 0),
+                "cache_miss": d(0, 0),
+            },
+            "batch_stats": [],
+            "memory_usage": [],
+        }
+
+
+class InferenceError(Exception):
+    def __init__(self, msg, status=400):
+        super().__init__(msg)
+        self.status = status
+
+
+class InferenceCore:
+    """Executes KServe-v2 inference requests against the model repository."""
+
+    def __init__(self, server_name="client_amd_server", version="0.1.0"):
+        self.server_name = server_name
+        self.version = version
+        self.models = {}
+        self.model_state = {}
+        self.stats = {}
+        self.shm = ShmRegistry()
+        self.trace_settings = {
+            "trace_file": "",
+            "trace_level": ["OFF"],
+            "trace_rate": "1000",
+            "trace_count": "-1",
+            "log_frequency": "0",
+        }
+        self.log_settings = {
+            "log_file": "",
+            "log_info": True,
+            "log_warning": True,
+            "log_error": True,
+            "log_verbose_level": 0,
+            "log_format": "default",
+        }
+        self.live = True
+        self.ready = True
+
+    # ---- repository ----
+    def add_model(self, model, ready=True):
+        self.models[model.name] = model
+        self.model_state[model.name] = "READY" if ready else "UNAVAILABLE"
+        self.stats.setdefault(model.name, ModelStats())
+
+    def get_model(self, name, must_be_ready=True):
+        model = self.models.get(name)
+        if model is None:
+            raise InferenceError(f"Request for unknown model: '{name}' is not found",
+                                 status=404)
+        if must_be_ready and self.model_state.get(name) != "READY":
+            raise InferenceError(
+                f"Request for unknown model: '{name}' is not ready", status=400
+            )
+        return model
+
+    def load_model(self, name):
+        if name not in self.models:
+            raise InferenceError(f"failed to load '{name}', no such model", status=400)
+        self.model_state[name] = "READY"
+
+    def unload_model(self, name):
+        if name not in self.models:
+            raise InferenceError(f"failed to unload '{name}', no such model",
+                                 status=400)
+        self.model_state[name] = "UNAVAILABLE"
+
+    def repository_index(self):
+        return [
+            {"name": name, "version": "1", "state": self.model_state[name],
+             "reason": ""}
+            for name in self.models
+        ]
+
+    def statistics(self, model_name=None):
+        if model_name:
+            self.get_model(model_name, must_be_ready=False)
+            names = [model_name]
+        else:
+            names = list(self.models)
+        return {"model_stats": [self.stats[n].to_dict(n) for n in names]}
+
+    # ---- input materialization ----
+    def _input_array(self, inp, binary_buf, binary_cursor):
+        """Returns (numpy array, new_cursor). inp is the request-JSON dict."""
+        name = inp["name"]
+        datatype = inp["datatype"]
+        shape = inp["shape"]
+        params = inp.get("parameters", {})
+        elem_count = int(np.prod(shape)) if shape else 1
+
+        shm_name = params.get("shared_memory_region")
+        if shm_name is not None:
+            byte_size = params["shared_memory_byte_size"]
+            offset = params.get("shared_memory_offset", 0)
+            region = self.shm.get_system(shm_name)
+            if region is not None:
+                base = region["offset"] + offset
+                raw = bytes(region["mmap"][base : base + byte_size])
+                return self._decode_raw(raw, datatype, shape), binary_cursor
+            hip_region = self.shm.get_hip(shm_name)
+            if hip_region is not None:
+                from ..ops import hip_runtime as hr
+
+                raw = hr.d2h_bytes(hip_region["ptr"], offset, byte_size)
+                return self._decode_raw(raw, datatype, shape), binary_cursor
+            raise InferenceError(
+                f"Unable to find shared memory region: '{shm_name}'"
+            )
+
+        bsize = params.get("binary_data_size")
+        if bsize is not None:
+            raw = binary_buf[binary_cursor : binary_cursor + bsize]
+            if len(raw) != bsize:
+                raise InferenceError(
+                    f"expected {bsize} bytes of binary data for input '{name}'"
+                )
+            return self._decode_raw(raw, datatype, shape), binary_cursor + bsize
+
+        data = inp.get("data")
+        if data is None:
+            raise InferenceError(f"no data supplied for input '{name}'")
+        if datatype == "BYTES":
+            arr = np.array(
+                [v.encode("utf-8") if isinstance(v, str) else v for v in data],
+                dtype=np.object_,
+            ).reshape(shape)
+        elif datatype == "BF16":
+            raise InferenceError("BF16 data must be sent as binary")
+        else:
+            arr = np.array(data, dtype=triton_to_np_dtype(datatype)).reshape(shape)
+        return arr, binary_cursor
+
+    @staticmethod
+    def _decode_raw(raw, datatype, shape):
+        if datatype == "BYTES":
+            arr = deserialize_bytes_tensor(raw)
+        elif datatype == "BF16":
+            arr = deserialize_bf16_tensor(raw)
+        else:
+            arr = np.frombuffer(raw, dtype=triton_to_np_dtype(datatype)).copy()
+        return arr.reshape(shape)
+
+    @staticmethod
+    def _encode_raw(arr, datatype):
+        if datatype == "BYTES":
+            s = serialize_byte_tensor(arr)
+            return s.item() if s.size > 0 else b""
+        if datatype == "BF16":
+            return serialize_bf16_tensor(arr).tobytes()
+        return np.ascontiguousarray(arr).tobytes()
+
+    # ---- inference ----
+    def infer(self, model_name, request, binary_buf=b""):
+        """Execute one request.
+
+        request: the parsed KServe-v2 JSON dict; binary_buf: the raw bytes
+        that followed the JSON header. Returns (response_dict, binary_parts)
+        where binary_parts is a list of raw output buffers to append.
+        """
+        model = self.get_model(model_name)
+        stats = self.stats[model_name]
+        t0 = time.monotonic_ns()
+        parameters = dict(request.get("parameters", {}))
+
+        try:
+            inputs = {}
+            cursor = 0
+            for inp in request.get("inputs", []):
+                arr, cursor = self._input_array(inp, binary_buf, cursor)
+                inputs[inp["name"]] = arr
+            t1 = time.monotonic_ns()
+
+            result = model.execute(inputs, parameters)
+            t2 = time.monotonic_ns()
+
+            response, binary_parts = self._build_response(
+                model, request, result, parameters
+            )
+            t3 = time.monotonic_ns()
+
+            stats.inference_count += 1
+            stats.execution_count += 1
+            stats.success_count += 1
+            stats.last_inference = int(time.time() * 1000)
+            stats.compute_input_ns += t1 - t0
+            stats.compute_infer_ns += t2 - t1
+            stats.compute_output_ns += t3 - t2
+            stats.success_ns += t3 - t0
+            return response, binary_parts
+        except InferenceError:
+            stats.fail_count += 1
+            raise
+        except Exception as e:
+            stats.fail_count += 1
+            raise InferenceError(str(e)) from e
+
+    def _build_response(self, model, request, result, parameters):
+        requested = request.get("outputs")
+        binary_default = bool(
+            parameters.get("binary_data_output", False)
+        )
+        out_specs = []
+        if requested:
+            for out in requested:
+                out_specs.append((out["name"], out.get("parameters", {})))
+        else:
+            for name in result:
+                out_specs.append((name, {"binary_data": binary_default}))
+
+        model_dtypes = {n: d for n, d, _ in model.outputs}
+        response = {
+            "model_name": model.name,
+            "model_version": "1",
+            "outputs": [],
+        }
+        if "id" in request:
+            response["id"] = request["id"]
+        binary_parts = []
+        for name, params in out_specs:
+            if name not in result:
+                raise InferenceError(
+                    f"unexpected inference output '{name}' for model '{model.name}'"
+                )
+            arr = result[name]
+            datatype = model_dtypes.get(name) or np_to_triton_dtype(arr.dtype)
+            out_json = {
+                "name": name,
+                "datatype": datatype,
+                "shape": list(arr.shape),
+            }
+
+            shm_name = params.get("shared_memory_region")
+            class_count = params.get("classification", 0)
+            if class_count:
+                arr = self._classify(arr, class_count)
+                out_json["datatype"] = "BYTES"
+                out_json["shape"] = list(arr.shape)
+                datatype = "BYTES"
+
+            if shm_name is not None:
+                raw = self._encode_raw(arr, datatype)
+                byte_size = params["shared_memory_byte_size"]
+                offset = params.get("shared_memory_offset", 0)
+                if len(raw) > byte_size:
+                    raise InferenceError(
+                        f"shared memory region '{shm_name}' is too small for "
+                        f"output '{name}'"
+                    )
+                region = self.shm.get_system(shm_name)
+                if region is not None:
+                    base = region["offset"] + offset
+                    region["mmap"][base : base + len(raw)] = raw
+                else:
+                    hip_region = self.shm.get_hip(shm_name)
+                    if hip_region is None:
+                        raise InferenceError(
+                            f"Unable to find shared memory region: '{shm_name}'"
+                        )
+                    from ..ops import hip_runtime as hr
+
+                    hr.h2d_bytes(hip_region["ptr"], offset, raw)
+                out_json["parameters"] = {
+                    "shared_memory_region": shm_name,
+                    "shared_memory_byte_size": len(raw),
+                }
+                if offset:
+                    out_json["parameters"]["shared_memory_offset"] = offset
+            elif params.get("binary_data", binary_default):
+                raw = self._encode_raw(arr, datatype)
+                out_json["parameters"] = {"binary_data_size": len(raw)}
+                binary_parts.append(raw)
+            else:
+                if datatype == "BYTES":
+                    out_json["data"] = [
+                        v.decode("utf-8") if isinstance(v, bytes) else str(v)
+                        for v in arr.reshape(-1)
+                    ]
+                elif datatype == "BF16":
+                    raise InferenceError("BF16 outputs require binary_data")
+                else:
+                    out_json["data"] = [v.item() for v in arr.reshape(-1)]
+            response["outputs"].append(out_json)
+        return response, binary_parts
+
+    @staticmethod
+    def _classify(arr, k):
+        """Top-k classification extension: returns BYTES '<score>:<idx>'."""
+        flat = arr.reshape(arr.shape[0], -1) if arr.ndim > 1 else arr.reshape(1, -1)
+        rows = []
+        for row in flat:
+            idx = np.argsort(row)[::-1][:k]
+            rows.append([f"{row[i]}:{i}".encode("utf-8") for i in idx])
+        out = np.array(rows, dtype=np.object_)
+        return out

@@ -1,0 +1,226 @@
+"""Model runtime for the client_amd test/benchmark server.
+
+The reference repo is a client SDK with no server, but its test strategy
+relies on a live server with canonical fixture models
+(reference: src/c++/tests/cc_client_test.cc:42-129 uses
+onnx_int32_int32_int32-style models; SURVEY.md §4 calls for a
+fake-server fixture). This module provides:
+
+- IdentityModel       — echoes inputs (the ``identity_fp32`` fixture)
+- AddSubModel         — INPUT0+INPUT1 / INPUT0-INPUT1 (the simple_* fixture)
+- SequenceModel       — server-held per-sequence accumulator state
+- RepeatModel         — decoupled: N responses per request (repeat_int32)
+- TorchModel          — a torch.nn.Module executed on an MI355X via
+                        PyTorch-ROCm (bf16), used by the benchmark
+"""
+
+import time
+
+import numpy as np
+
+from ..utils import triton_to_np_dtype, np_to_triton_dtype
+
+
+class Model:
+    """Base model: subclasses implement execute(inputs, parameters)
+    returning a dict name -> numpy array."""
+
+    def __init__(self, name, inputs, outputs, max_batch_size=0, platform="python",
+                 decoupled=False):
+        # inputs/outputs: list of (name, datatype, shape) with -1 for dynamic dims
+        self.name = name
+        self.inputs = inputs
+        self.outputs = outputs
+        self.max_batch_size = max_batch_size
+        self.platform = platform
+        self.decoupled = decoupled
+        self.versions = ["1"]
+
+    def metadata(self):
+        return {
+            "name": self.name,
+            "versions": self.versions,
+            "platform": self.platform,
+            "inputs": [
+                {"name": n, "datatype": d, "shape": list(s)} for n, d, s in self.inputs
+            ],
+            "outputs": [
+                {"name": n, "datatype": d, "shape": list(s)} for n, d, s in self.outputs
+            ],
+        }
+
+    def config(self):
+        return {
+            "name": self.name,
+            "platform": self.platform,
+            "backend": self.platform,
+            "max_batch_size": self.max_batch_size,
+            "input": [
+                {"name": n, "data_type": "TYPE_" + d, "dims": list(s)}
+                for n, d, s in self.inputs
+            ],
+            "output": [
+                {"name": n, "data_type": "TYPE_" + d, "dims": list(s)}
+                for n, d, s in self.outputs
+            ],
+            "model_transaction_policy": {"decoupled": self.decoupled},
+        }
+
+    def execute(self, inputs, parameters):
+        raise NotImplementedError
+
+
+class IdentityModel(Model):
+    """Echo each INPUTi to OUTPUTi. identity_fp32 fixture analog."""
+
+    def __init__(self, name="identity_fp32", datatype="FP32", n_io=1):
+        ios = [("INPUT" + str(i) if n_io > 1 else "INPUT0", datatype, [-1])
+               for i in range(n_io)]
+        outs = [("OUTPUT" + str(i) if n_io > 1 else "OUTPUT0", datatype, [-1])
+                for i in range(n_io)]
+        super().__init__(name, ios, outs)
+
+    def execute(self, inputs, parameters):
+        out = {}
+        for (in_name, _, _), (out_name, _, _) in zip(self.inputs, self.outputs):
+            out[out_name] = inputs[in_name]
+        return out
+
+
+class AddSubModel(Model):
+    """OUTPUT0 = INPUT0 + INPUT1; OUTPUT1 = INPUT0 - INPUT1.
+
+    Matches the canonical simple/onnx_int32_int32_int32 fixture shape
+    used throughout the reference examples (e.g.
+    src/python/examples/simple_http_infer_client.py).
+    """
+
+    def __init__(self, name="simple", datatype="INT32", shape=(-1, 16)):
+        super().__init__(
+            name,
+            [("INPUT0", datatype, list(shape)), ("INPUT1", datatype, list(shape))],
+            [("OUTPUT0", datatype, list(shape)), ("OUTPUT1", datatype, list(shape))],
+        )
+
+    def execute(self, inputs, parameters):
+        a, b = inputs["INPUT0"], inputs["INPUT1"]
+        if a.dtype == np.object_:
+            # BYTES add/sub fixture: numeric strings
+            ai = np.array([int(x) for x in a.reshape(-1)]).reshape(a.shape)
+            bi = np.array([int(x) for x in b.reshape(-1)]).reshape(b.shape)
+            return {
+                "OUTPUT0": np.char.encode((ai + bi).astype(str), "utf-8").astype(np.object_),
+                "OUTPUT1": np.char.encode((ai - bi).astype(str), "utf-8").astype(np.object_),
+            }
+        return {"OUTPUT0": a + b, "OUTPUT1": a - b}
+
+
+class SequenceModel(Model):
+    """Stateful sequence accumulator: the server holds per-sequence_id
+    running sums; sequence_start resets, sequence_end finalizes
+    (server-side analog of the reference's sequence examples,
+    simple_grpc_sequence_stream_infer_client.cc)."""
+
+    def __init__(self, name="sequence_accumulate", datatype="INT32"):
+        super().__init__(name, [("INPUT", datatype, [1])], [("OUTPUT", datatype, [1])])
+        self._state = {}
+
+    def execute(self, inputs, parameters):
+        seq_id = parameters.get("sequence_id", 0)
+        start = parameters.get("sequence_start", False)
+        end = parameters.get("sequence_end", False)
+        val = inputs["INPUT"]
+        if start or seq_id not in self._state:
+            self._state[seq_id] = np.zeros_like(val)
+        self._state[seq_id] = self._state[seq_id] + val
+        out = self._state[seq_id].copy()
+        if end:
+            self._state.pop(seq_id, None)
+        return {"OUTPUT": out}
+
+
+class RepeatModel(Model):
+    """Decoupled model: for input IN (shape [n]) and DELAY, produces one
+    response per element (reference example:
+    simple_grpc_custom_repeat.cc:135-176 drives the repeat_int32 model)."""
+
+    def __init__(self, name="repeat_int32"):
+        super().__init__(
+            name,
+            [("IN", "INT32", [-1]), ("DELAY", "UINT32", [-1]), ("WAIT", "UINT32", [1])],
+            [("OUT", "INT32", [1]), ("IDX", "UINT32", [1])],
+            decoupled=True,
+        )
+
+    def execute_decoupled(self, inputs, parameters):
+        vals = inputs["IN"].reshape(-1)
+        delays = inputs.get("DELAY")
+        delays = delays.reshape(-1) if delays is not None else np.zeros(len(vals))
+        for idx, v in enumerate(vals):
+            if idx < len(delays) and delays[idx] > 0:
+                time.sleep(float(delays[idx]) / 1000.0)
+            yield {
+                "OUT": np.array([v], dtype=np.int32),
+                "IDX": np.array([idx], dtype=np.uint32),
+            }
+
+    def execute(self, inputs, parameters):
+        # Non-decoupled fallback: return the first response only.
+        for out in self.execute_decoupled(inputs, parameters):
+            return out
+        return {"OUT": np.zeros(1, np.int32), "IDX": np.zeros(1, np.uint32)}
+
+
+class TorchModel(Model):
+    """Executes a torch.nn.Module on the configured device.
+
+    On an MI355X the module runs in bf16 out of HBM3E; inputs arriving
+    via HIP-IPC shared memory are consumed as device tensors without any
+    host round-trip (see server/shm.py + ops/_hip extension).
+    """
+
+    def __init__(self, name, module, inputs, outputs, device="cuda:0",
+                 dtype=None, max_batch_size=0):
+        super().__init__(name, inputs, outputs, max_batch_size, platform="pytorch")
+        import torch
+
+        self._torch = torch
+        self.device = device
+        self.module = module.to(device)
+        if dtype is not None:
+            self.module = self.module.to(dtype)
+        self.dtype = dtype
+        self.module.eval()
+
+    def execute(self, inputs, parameters):
+        torch = self._torch
+        with torch.inference_mode():
+            tensors = []
+            for name, datatype, _ in self.inputs:
+                arr = inputs[name]
+                t = torch.from_numpy(np.ascontiguousarray(arr)).to(self.device)
+                if self.dtype is not None and t.is_floating_point():
+                    t = t.to(self.dtype)
+                tensors.append(t)
+            result = self.module(*tensors)
+            if not isinstance(result, (tuple, list)):
+                result = (result,)
+            out = {}
+            for (name, datatype, _), t in zip(self.outputs, result):
+                npdt = triton_to_np_dtype(datatype)
+                tt = t
+                if tt.dtype == torch.bfloat16:
+                    tt = tt.float()
+                out[name] = tt.detach().cpu().numpy().astype(npdt, copy=False)
+            return out
+
+    def execute_torch(self, device_tensors):
+        """Device-resident fast path: takes a list of torch tensors already
+        on self.device, returns the module outputs as device tensors.
+        Used by the HIP-shm data plane (no numpy, no host copies)."""
+        torch = self._torch
+        with torch.inference_mode():
+            result = self.module(*device_tensors)
+            if not isinstance(result, (tuple, list)):
+                result = (result,)
+            return list(result)
