@@ -1,0 +1,106 @@
+"""gRPC InferInput (reference: tritonclient/grpc/_infer_input.py).
+
+Tensor bytes ride in ModelInferRequest.raw_input_contents — the wire
+form the protocol chose explicitly for performance
+(grpc_service.proto:683-706).
+"""
+
+import numpy as np
+
+from ..utils import (
+    np_to_triton_dtype,
+    raise_error,
+    serialize_bf16_tensor,
+    serialize_byte_tensor,
+)
+from ._proto import service_pb2
+
+
+class InferInput:
+    def __init__(self, name, shape, datatype):
+        self._input = service_pb2.ModelInferRequest.InferInputTensor()
+        self._input.name = name
+        self._input.ClearField("shape")
+        self._input.shape.extend(shape)
+        self._input.datatype = datatype
+        self._raw_content = None
+
+    def name(self):
+        return self._input.name
+
+    def datatype(self):
+        return self._input.datatype
+
+    def shape(self):
+        return list(self._input.shape)
+
+    def set_shape(self, shape):
+        self._input.ClearField("shape")
+        self._input.shape.extend(shape)
+        return self
+
+    def set_data_from_numpy(self, input_tensor):
+        if not isinstance(input_tensor, (np.ndarray,)):
+            raise_error("input_tensor must be a numpy array")
+        dtype = np_to_triton_dtype(input_tensor.dtype)
+        if self._input.datatype != dtype:
+            if self._input.datatype == "BF16":
+                if input_tensor.dtype not in (np.float16, np.float32):
+                    raise_error(
+                        "got unexpected datatype {} from numpy array, expected "
+                        "float16/float32 for BF16 input".format(dtype)
+                    )
+            else:
+                raise_error(
+                    "got unexpected datatype {} from numpy array, expected {}".format(
+                        dtype, self._input.datatype
+                    )
+                )
+        valid_shape = True
+        if len(self._input.shape) != len(input_tensor.shape):
+            valid_shape = False
+        else:
+            for i in range(len(self._input.shape)):
+                if self._input.shape[i] != input_tensor.shape[i]:
+                    valid_shape = False
+        if not valid_shape:
+            raise_error(
+                "got unexpected numpy array shape [{}], expected [{}]".format(
+                    str(input_tensor.shape)[1:-1], str(list(self._input.shape))[1:-1]
+                )
+            )
+        self._input.parameters.pop("shared_memory_region", None)
+        self._input.parameters.pop("shared_memory_byte_size", None)
+        self._input.parameters.pop("shared_memory_offset", None)
+        if self._input.datatype == "BYTES":
+            serialized_output = serialize_byte_tensor(input_tensor)
+            if serialized_output.size > 0:
+                self._raw_content = serialized_output.item()
+            else:
+                self._raw_content = b""
+        elif self._input.datatype == "BF16":
+            self._raw_content = serialize_bf16_tensor(input_tensor).tobytes()
+        else:
+            self._raw_content = np.ascontiguousarray(input_tensor).tobytes()
+        return self
+
+    def set_raw_bytes(self, raw_bytes):
+        """Attach pre-serialized tensor bytes directly (zero extra copies);
+        used by the device pack path and the load generator."""
+        self._raw_content = raw_bytes
+        return self
+
+    def set_shared_memory(self, region_name, byte_size, offset=0):
+        self._input.ClearField("contents")
+        self._raw_content = None
+        self._input.parameters["shared_memory_region"].string_param = region_name
+        self._input.parameters["shared_memory_byte_size"].int64_param = byte_size
+        if offset != 0:
+            self._input.parameters["shared_memory_offset"].int64_param = offset
+        return self
+
+    def _get_tensor(self):
+        return self._input
+
+    def _get_content(self):
+        return self._raw_content

@@ -271,12 +271,18 @@ class InferenceCore:
 
         bsize = params.get("binary_data_size")
         if bsize is not None:
-            raw = binary_buf[binary_cursor : binary_cursor + bsize]
+            if isinstance(binary_buf, (list, tuple)):
+                # gRPC path: one buffer per input (raw_input_contents)
+                raw = binary_buf[binary_cursor]
+                cursor = binary_cursor + 1
+            else:
+                raw = binary_buf[binary_cursor : binary_cursor + bsize]
+                cursor = binary_cursor + bsize
             if len(raw) != bsize:
                 raise InferenceError(
                     f"expected {bsize} bytes of binary data for input '{name}'"
                 )
-            return self._decode_raw(raw, datatype, shape), binary_cursor + bsize
+            return self._decode_raw(raw, datatype, shape), cursor
 
         data = inp.get("data")
         if data is None:

@@ -1,0 +1,248 @@
+"""gRPC client <-> server integration tests (CPU only).
+
+Mirrors the reference's typed client test matrix
+(cc_client_test.cc:42-129 runs the same suite against HTTP and gRPC)
+plus streaming/decoupled coverage.
+"""
+
+import queue
+import threading
+
+import numpy as np
+import pytest
+
+import client_amd.grpc as grpcclient
+from client_amd.utils import InferenceServerException
+
+
+@pytest.fixture()
+def client(grpc_fixture_server):
+    host, port, _ = grpc_fixture_server
+    c = grpcclient.InferenceServerClient(f"{host}:{port}")
+    yield c
+    c.close()
+
+
+def test_health(client):
+    assert client.is_server_live()
+    assert client.is_server_ready()
+    assert client.is_model_ready("identity_fp32")
+    assert not client.is_model_ready("nope")
+
+
+def test_metadata_config(client):
+    meta = client.get_server_metadata()
+    assert meta.name == "client_amd_server"
+    as_json = client.get_server_metadata(as_json=True)
+    assert as_json["name"] == "client_amd_server"
+    mm = client.get_model_metadata("simple")
+    assert mm.inputs[0].name == "INPUT0"
+    cfg = client.get_model_config("simple")
+    assert cfg.config.name == "simple"
+    assert cfg.config.input[0].data_type == 8  # TYPE_INT32
+
+
+def test_repository(client):
+    index = client.get_model_repository_index()
+    names = {m.name for m in index.models}
+    assert "simple" in names
+    client.unload_model("simple")
+    assert not client.is_model_ready("simple")
+    client.load_model("simple")
+    assert client.is_model_ready("simple")
+
+
+def test_infer(client):
+    a = np.arange(16, dtype=np.int32).reshape(1, 16)
+    b = np.full((1, 16), 3, dtype=np.int32)
+    inputs = [
+        grpcclient.InferInput("INPUT0", [1, 16], "INT32"),
+        grpcclient.InferInput("INPUT1", [1, 16], "INT32"),
+    ]
+    inputs[0].set_data_from_numpy(a)
+    inputs[1].set_data_from_numpy(b)
+    outputs = [
+        grpcclient.InferRequestedOutput("OUTPUT0"),
+        grpcclient.InferRequestedOutput("OUTPUT1"),
+    ]
+    result = client.infer("simple", inputs, outputs=outputs, request_id="7")
+    np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), a + b)
+    np.testing.assert_array_equal(result.as_numpy("OUTPUT1"), a - b)
+    assert result.get_response().id == "7"
+    out_json = result.get_output("OUTPUT0", as_json=True)
+    assert out_json["datatype"] == "INT32"
+
+
+def test_infer_bf16_and_bytes(client):
+    x = np.array([[0.5, -1.5, 2.0, 8.0]], dtype=np.float32)
+    inp = grpcclient.InferInput("INPUT0", list(x.shape), "BF16")
+    inp.set_data_from_numpy(x)
+    result = client.infer("identity_bf16", [inp])
+    np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
+
+    s = np.array([b"ab", b"\xff\x00"], dtype=np.object_)
+    inp = grpcclient.InferInput("INPUT0", [2], "BYTES")
+    inp.set_data_from_numpy(s)
+    result = client.infer("identity_bytes", [inp])
+    np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), s)
+
+
+def test_async_infer(client):
+    a = np.arange(16, dtype=np.int32).reshape(1, 16)
+    inputs = [
+        grpcclient.InferInput("INPUT0", [1, 16], "INT32"),
+        grpcclient.InferInput("INPUT1", [1, 16], "INT32"),
+    ]
+    inputs[0].set_data_from_numpy(a)
+    inputs[1].set_data_from_numpy(a)
+    done = queue.Queue()
+
+    def callback(result, error):
+        done.put((result, error))
+
+    ctx = client.async_infer("simple", inputs, callback)
+    assert ctx is not None
+    result, error = done.get(timeout=10)
+    assert error is None
+    np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), a + a)
+
+
+def test_async_infer_error(client):
+    a = np.zeros((1, 16), dtype=np.int32)
+    inputs = [grpcclient.InferInput("INPUT0", [1, 16], "INT32")]
+    inputs[0].set_data_from_numpy(a)
+    done = queue.Queue()
+    client.async_infer("no_such_model", inputs, lambda result, error: done.put(error))
+    error = done.get(timeout=10)
+    assert isinstance(error, InferenceServerException)
+
+
+def test_stream_sequence(client):
+    """Sequence inference over the bi-di stream (reference example:
+    simple_grpc_sequence_stream_infer_client)."""
+    results = queue.Queue()
+    client.start_stream(callback=lambda result, error: results.put((result, error)))
+    try:
+        for i, (start, end) in enumerate([(True, False), (False, False),
+                                          (False, True)]):
+            inp = grpcclient.InferInput("INPUT", [1], "INT32")
+            inp.set_data_from_numpy(np.array([i + 1], dtype=np.int32))
+            client.async_stream_infer(
+                "sequence_accumulate", [inp], sequence_id=5,
+                sequence_start=start, sequence_end=end,
+            )
+        vals = []
+        for _ in range(3):
+            result, error = results.get(timeout=10)
+            assert error is None
+            vals.append(int(result.as_numpy("OUTPUT")[0]))
+        assert vals == [1, 3, 6]
+    finally:
+        client.stop_stream()
+
+
+def test_stream_decoupled_repeat(client):
+    """Decoupled model: N responses per request, final-response flag
+    (reference simple_grpc_custom_repeat.cc:135-176)."""
+    results = queue.Queue()
+    client.start_stream(callback=lambda result, error: results.put((result, error)))
+    try:
+        n = 4
+        in_vals = np.arange(n, dtype=np.int32)
+        delays = np.zeros(n, dtype=np.uint32)
+        wait = np.zeros(1, dtype=np.uint32)
+        inputs = [
+            grpcclient.InferInput("IN", [n], "INT32"),
+            grpcclient.InferInput("DELAY", [n], "UINT32"),
+            grpcclient.InferInput("WAIT", [1], "UINT32"),
+        ]
+        inputs[0].set_data_from_numpy(in_vals)
+        inputs[1].set_data_from_numpy(delays)
+        inputs[2].set_data_from_numpy(wait)
+        client.async_stream_infer(
+            "repeat_int32", inputs, enable_empty_final_response=True
+        )
+        seen = []
+        while True:
+            result, error = results.get(timeout=10)
+            assert error is None
+            if result.is_final_response():
+                break
+            seen.append(int(result.as_numpy("OUT")[0]))
+        assert seen == list(range(n))
+    finally:
+        client.stop_stream()
+
+
+def test_stream_error_in_band(client):
+    """A bad request on the stream reports via callback error and the
+    stream stays usable."""
+    results = queue.Queue()
+    client.start_stream(callback=lambda result, error: results.put((result, error)))
+    try:
+        inp = grpcclient.InferInput("INPUT", [1], "INT32")
+        inp.set_data_from_numpy(np.array([1], dtype=np.int32))
+        client.async_stream_infer("no_such_model", [inp])
+        result, error = results.get(timeout=10)
+        assert isinstance(error, InferenceServerException)
+        # stream still active
+        client.async_stream_infer(
+            "sequence_accumulate", [inp], sequence_id=6,
+            sequence_start=True, sequence_end=True,
+        )
+        result, error = results.get(timeout=10)
+        assert error is None
+    finally:
+        client.stop_stream()
+
+
+def test_statistics(client):
+    x = np.random.rand(1, 4).astype(np.float32)
+    inp = grpcclient.InferInput("INPUT0", [1, 4], "FP32")
+    inp.set_data_from_numpy(x)
+    client.infer("identity_fp32", [inp])
+    stats = client.get_inference_statistics("identity_fp32")
+    assert stats.model_stats[0].inference_count >= 1
+    assert stats.model_stats[0].inference_stats.success.count >= 1
+
+
+def test_trace_log_settings(client):
+    settings = client.get_trace_settings()
+    assert "trace_rate" in settings.settings
+    updated = client.update_trace_settings(settings={"trace_rate": "250"})
+    assert updated.settings["trace_rate"].value[0] == "250"
+    log = client.get_log_settings(as_json=True)
+    assert "log_info" in log["settings"]
+    updated = client.update_log_settings({"log_verbose_level": 3})
+    assert updated.settings["log_verbose_level"].uint32_param == 3
+
+
+def test_error_mapping(client):
+    with pytest.raises(InferenceServerException) as exc:
+        client.get_model_metadata("not_a_model")
+    assert "not_a_model" in str(exc.value)
+
+
+def test_keepalive_and_timeout():
+    opts = grpcclient.KeepAliveOptions(keepalive_time_ms=10000)
+    c = grpcclient.InferenceServerClient("127.0.0.1:1", keepalive_options=opts)
+    with pytest.raises(InferenceServerException):
+        c.is_server_live(client_timeout=0.2)
+    c.close()
+
+
+def test_infer_timeout(client, grpc_fixture_server):
+    # microscopic timeout -> Deadline Exceeded surface
+    # (reference client_timeout_test.cc drives every API this way)
+    a = np.zeros((1, 16), dtype=np.int32)
+    inputs = [
+        grpcclient.InferInput("INPUT0", [1, 16], "INT32"),
+        grpcclient.InferInput("INPUT1", [1, 16], "INT32"),
+    ]
+    inputs[0].set_data_from_numpy(a)
+    inputs[1].set_data_from_numpy(a)
+    try:
+        client.infer("simple", inputs, client_timeout=0.000001)
+        raise AssertionError("expected timeout")
+    except InferenceServerException as e:
+        assert "DEADLINE" in str(e.status()).upper()
