@@ -1,0 +1,297 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: perf_analyzer-style serving throughput.
+
+Measures the BASELINE.json north-star config: ResNet50 bs=8, bf16,
+HIP-IPC shared-memory I/O, gRPC async clients, one MI355X-backed server
+replica per GPU (one rank per GPU under torch.distributed.run).
+
+Each rank: spawns a server process on its GPU, creates per-slot HIP-shm
+input/output regions, registers them over gRPC, and drives closed-loop
+async inference. A step = `reqs_per_step` requests at concurrency
+`concurrency`; every step re-runs the CDNA4 pack kernel (fp32 -> bf16
+into the slot-0 input region) and, when world_size > 1, RCCL-broadcasts
+rank 0's staged input to every rank's region over xGMI before issuing
+requests (BASELINE.md config 3/5 shape).
+
+Rank 0 prints one JSON line (whole-job aggregate; weak scaling).
+"""
+
+import argparse
+import json
+import os
+import subprocess
+import sys
+import threading
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+BATCH = 8
+IN_SHAPE = (BATCH, 3, 224, 224)
+OUT_SHAPE = (BATCH, 1000)
+
+
+def start_server(device_index, port):
+    env = dict(os.environ)
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "client_amd.server", "--grpc-port", str(port),
+         "--models", "resnet50", "--device", f"cuda:{device_index}",
+         "--dtype", "bf16", "--grpc-workers", "8"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True, env=env,
+        cwd=os.path.dirname(os.path.abspath(__file__)),
+    )
+    deadline = time.time() + 300
+    port_actual = None
+    while time.time() < deadline:
+        line = proc.stdout.readline()
+        if not line:
+            if proc.poll() is not None:
+                raise RuntimeError("server exited early")
+            time.sleep(0.05)
+            continue
+        if line.startswith("GRPC_READY"):
+            port_actual = int(line.split()[1])
+            break
+    if port_actual is None:
+        proc.terminate()
+        raise RuntimeError("server did not become ready")
+    # drain server stdout in the background so it never blocks
+    threading.Thread(target=lambda: [None for _ in proc.stdout],
+                     daemon=True).start()
+    return proc, port_actual
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--concurrency", type=int, default=4)
+    ap.add_argument("--reqs-per-step", type=int, default=8)
+    args = ap.parse_args()
+
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+
+    if not torch.cuda.is_available():
+        print(json.dumps({"error": "no HIP device available"}))
+        return 1
+
+    torch.cuda.set_device(local_rank)
+    if distributed:
+        import torch.distributed as dist
+
+        dist.init_process_group("nccl")
+
+    import client_amd.grpc as grpcclient
+    import client_amd.utils.hip_shared_memory as hipshm
+    import torch.utils.dlpack
+
+    server_proc, port = start_server(local_rank, 8101 + local_rank)
+    try:
+        client = grpcclient.InferenceServerClient(f"127.0.0.1:{port}")
+        for _ in range(120):
+            try:
+                if client.is_server_ready():
+                    break
+            except Exception:
+                time.sleep(0.25)
+
+        # per-slot region pairs (bf16 in HBM3E)
+        in_elems = int(np.prod(IN_SHAPE))
+        out_elems = int(np.prod(OUT_SHAPE))
+        slots = []
+        for s in range(args.concurrency):
+            in_r = hipshm.create_shared_memory_region(
+                f"bench_in_{s}", in_elems * 2, local_rank
+            )
+            out_r = hipshm.create_shared_memory_region(
+                f"bench_out_{s}", out_elems * 2, local_rank
+            )
+            client.register_cuda_shared_memory(
+                f"bench_in_{s}", hipshm.get_raw_handle_bytes(in_r), local_rank,
+                in_elems * 2,
+            )
+            client.register_cuda_shared_memory(
+                f"bench_out_{s}", hipshm.get_raw_handle_bytes(out_r),
+                local_rank, out_elems * 2,
+            )
+            inp = grpcclient.InferInput("INPUT0", list(IN_SHAPE), "BF16")
+            inp.set_shared_memory(f"bench_in_{s}", in_elems * 2)
+            out = grpcclient.InferRequestedOutput("OUTPUT0")
+            out.set_shared_memory(f"bench_out_{s}", out_elems * 2)
+            slots.append({"in": in_r, "out": out_r, "inputs": [inp],
+                          "outputs": [out]})
+
+        # synthetic fp32 input, packed on-device to bf16
+        host_x = np.random.rand(*IN_SHAPE).astype(np.float32)
+        for slot in slots:
+            hipshm.set_shared_memory_region_cast(slot["in"], host_x, "BF16")
+
+        # torch views of slot-0 input region for the RCCL broadcast path
+        bcast_tensor = None
+        if distributed:
+            smt = hipshm.as_shared_memory_tensor(
+                slots[0]["in"], "BF16", list(IN_SHAPE)
+            )
+            bcast_tensor = torch.from_dlpack(smt)
+
+        latencies = []
+        lat_lock = threading.Lock()
+
+        def run_step(record=False):
+            """pack -> (broadcast) -> reqs_per_step requests at the
+            configured concurrency, closed loop."""
+            hipshm.set_shared_memory_region_cast(
+                slots[0]["in"], host_x, "BF16", sync=not distributed
+            )
+            if distributed:
+                # fan the staged input out to every replica over xGMI
+                import torch.distributed as dist
+
+                torch.cuda.synchronize()
+                dist.broadcast(bcast_tensor, src=0)
+                torch.cuda.synchronize()
+            remaining = args.reqs_per_step
+            done_evt = threading.Event()
+            state = {"outstanding": 0, "issued": 0}
+            lock = threading.Lock()
+
+            def issue_one(slot_idx):
+                slot = slots[slot_idx % len(slots)]
+                t0 = time.monotonic_ns()
+
+                def cb(result, error):
+                    if error is not None:
+                        raise SystemExit(f"infer error: {error}")
+                    if record:
+                        with lat_lock:
+                            latencies.append(time.monotonic_ns() - t0)
+                    with lock:
+                        state["outstanding"] -= 1
+                        if state["issued"] < args.reqs_per_step:
+                            idx = state["issued"]
+                            state["issued"] += 1
+                            state["outstanding"] += 1
+                            should_issue = idx
+                        elif state["outstanding"] == 0:
+                            done_evt.set()
+                            return
+                        else:
+                            return
+                    issue_one(should_issue)
+
+                client.async_infer(
+                    "resnet50", slot["inputs"], callback=cb,
+                    outputs=slot["outputs"],
+                )
+
+            first = min(args.concurrency, args.reqs_per_step)
+            with lock:
+                state["issued"] = first
+                state["outstanding"] = first
+            for i in range(first):
+                issue_one(i)
+            if not done_evt.wait(timeout=300):
+                raise SystemExit("step timed out")
+
+        # warmup
+        for _ in range(args.warmup):
+            run_step(record=False)
+
+        # timed region
+        if distributed:
+            import torch.distributed as dist
+
+            dist.barrier()
+        torch.cuda.synchronize()
+        t_start = time.monotonic()
+        for _ in range(args.steps):
+            run_step(record=True)
+        torch.cuda.synchronize()
+        if distributed:
+            dist.barrier()
+        t_end = time.monotonic()
+        elapsed = t_end - t_start
+
+        # max elapsed over ranks -> whole-job throughput
+        if distributed:
+            e = torch.tensor([elapsed], device="cuda")
+            dist.all_reduce(e, op=dist.ReduceOp.MAX)
+            elapsed = float(e.item())
+
+        total_requests = args.steps * args.reqs_per_step * world
+        total_inferences = total_requests * BATCH
+        value = total_inferences / elapsed
+        ms_per_step = elapsed / args.steps * 1000.0
+
+        lat_sorted = sorted(latencies)
+
+        def pct(q):
+            if not lat_sorted:
+                return 0
+            return lat_sorted[min(len(lat_sorted) - 1,
+                                  int(round(q / 100 * (len(lat_sorted) - 1))))]
+
+        # sanity: output region has live finite data
+        logits = hipshm.get_contents_cast(slots[0]["out"], "BF16",
+                                          list(OUT_SHAPE))
+        assert np.all(np.isfinite(logits)), "non-finite output"
+
+        if rank == 0:
+            print(json.dumps({
+                "metric": "perf_analyzer inferences/sec + p99 latency, "
+                          "ResNet50 bs=8 HIP-shm at 1/2/4/8 GPU",
+                "value": round(value, 2),
+                "unit": "inferences/sec",
+                "n_gpus": world,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": round(ms_per_step, 3),
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "bf16",
+                "data": "synthetic",
+                "config": {
+                    "model": "resnet50",
+                    "global_batch": BATCH * args.reqs_per_step * world,
+                    "batch_size": BATCH,
+                    "seq_len": None,
+                    "parallelism": f"replicated-serving x{world}, one gRPC "
+                                   f"client+server pair per GPU, HIP-IPC shm "
+                                   f"I/O, RCCL bcast fan-out",
+                    "concurrency": args.concurrency,
+                    "reqs_per_step": args.reqs_per_step,
+                    "requests_per_sec": round(total_requests / elapsed, 2),
+                    "latency_ms": {
+                        "p50": round(pct(50) / 1e6, 3),
+                        "p90": round(pct(90) / 1e6, 3),
+                        "p99": round(pct(99) / 1e6, 3),
+                    },
+                },
+            }))
+
+        for slot in slots:
+            hipshm.destroy_shared_memory_region(slot["in"])
+            hipshm.destroy_shared_memory_region(slot["out"])
+        client.close()
+        if distributed:
+            dist.destroy_process_group()
+        return 0
+    finally:
+        server_proc.terminate()
+        try:
+            server_proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            server_proc.kill()
+
+
+if __name__ == "__main__":
+    sys.exit(main())

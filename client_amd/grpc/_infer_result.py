@@ -37,6 +37,9 @@ class InferResult:
         index = 0
         for output in self._result.outputs:
             if output.name == name:
+                if "shared_memory_region" in output.parameters:
+                    # data lives in the client's shared-memory region
+                    return None
                 shape = list(output.shape)
                 datatype = output.datatype
                 if self._uses_raw_slot(output) and index < len(

@@ -1,0 +1,362 @@
+"""perf_analyzer-class load generator.
+
+The reference repo ships only a relocation stub for perf_analyzer
+(src/c++/perf_analyzer/README.md:29-30) but the north-star metric is
+perf_analyzer-style inferences/sec + p99 latency, so this
+is a from-scratch implementation designed from the client-stat substrate
+the reference exposes (RequestTimers/InferStat, common.h:93-114,
+568-648): closed-loop concurrency driving, warmup, fixed measurement
+windows with a stability check, latency percentiles from
+REQUEST_START -> REQUEST_END nanosecond timestamps.
+
+Input modes:
+  - wire:   synthetic tensors serialized once, bytes reused per request
+  - hipshm: per-concurrency-slot HIP-IPC region pairs; the CDNA4 pack
+    kernel stages input once (or per request with --repack); requests
+    carry only region references — no tensor bytes on the wire
+"""
+
+import threading
+import time
+
+import numpy as np
+
+from ..utils import triton_to_np_dtype, TRITON_DTYPE_SIZES
+
+
+class LatencyRecorder:
+    def __init__(self):
+        self._lock = threading.Lock()
+        self.latencies_ns = []
+        self.errors = 0
+
+    def record(self, start_ns, end_ns, error=None):
+        with self._lock:
+            if error is not None:
+                self.errors += 1
+            else:
+                self.latencies_ns.append(end_ns - start_ns)
+
+    def snapshot_and_reset(self):
+        with self._lock:
+            lat, self.latencies_ns = self.latencies_ns, []
+            err, self.errors = self.errors, 0
+        return lat, err
+
+
+def percentile(sorted_ns, q):
+    if not sorted_ns:
+        return 0.0
+    idx = min(len(sorted_ns) - 1, int(round(q / 100.0 * (len(sorted_ns) - 1))))
+    return sorted_ns[idx]
+
+
+class ConcurrencyDriver:
+    """Closed-loop load: ``concurrency`` slots, each slot re-issues as
+    soon as its previous request completes (perf_analyzer's concurrency
+    model)."""
+
+    def __init__(self, issue_fn, concurrency):
+        self._issue_fn = issue_fn
+        self._concurrency = concurrency
+        self._recorder = LatencyRecorder()
+        self._stop = threading.Event()
+        self._inflight = threading.Semaphore(0)
+        self._done = threading.Event()
+        self._active = 0
+        self._lock = threading.Lock()
+
+    def _launch(self, slot):
+        if self._stop.is_set():
+            with self._lock:
+                self._active -= 1
+                if self._active == 0:
+                    self._done.set()
+            return
+        start = time.monotonic_ns()
+
+        def on_complete(error=None):
+            self._recorder.record(start, time.monotonic_ns(), error)
+            self._launch(slot)
+
+        try:
+            self._issue_fn(slot, on_complete)
+        except Exception as e:
+            self._recorder.record(start, time.monotonic_ns(), e)
+            with self._lock:
+                self._active -= 1
+                if self._active == 0:
+                    self._done.set()
+
+    def run(self, warmup_s, window_s, max_windows, stability_pct=10.0,
+            min_stable=3):
+        """Returns (throughput_req_s, latencies_ns_sorted, errors, windows)."""
+        self._active = self._concurrency
+        for slot in range(self._concurrency):
+            self._launch(slot)
+        time.sleep(warmup_s)
+        self._recorder.snapshot_and_reset()
+        window_results = []
+        all_lat = []
+        stable = 0
+        for _ in range(max_windows):
+            time.sleep(window_s)
+            lat, err = self._recorder.snapshot_and_reset()
+            thr = len(lat) / window_s
+            window_results.append((thr, lat, err))
+            all_lat.extend(lat)
+            if len(window_results) >= min_stable:
+                recent = [w[0] for w in window_results[-min_stable:]]
+                mean = sum(recent) / len(recent)
+                if mean > 0 and all(
+                    abs(r - mean) / mean * 100 <= stability_pct for r in recent
+                ):
+                    stable += 1
+                    break
+        self._stop.set()
+        self._done.wait(timeout=60)
+        total_s = window_s * len(window_results)
+        total_req = sum(len(w[1]) for w in window_results)
+        total_err = sum(w[2] for w in window_results)
+        throughput = total_req / total_s if total_s else 0.0
+        all_lat.sort()
+        return throughput, all_lat, total_err, len(window_results)
+
+
+class PerfAnalyzer:
+    def __init__(self, url, protocol="grpc", model_name="identity_fp32",
+                 batch_size=1, shared_memory="none", input_dtype=None,
+                 repack=False, verbose=False, device_id=0):
+        self.url = url
+        self.protocol = protocol
+        self.model_name = model_name
+        self.batch_size = batch_size
+        self.shared_memory = shared_memory
+        self.repack = repack
+        self.verbose = verbose
+        self.device_id = device_id
+        self.input_dtype = input_dtype
+        self._client = None
+        self._slots = []
+
+    # ---- setup ----
+
+    def _make_client(self, concurrency):
+        if self.protocol == "grpc":
+            import client_amd.grpc as grpcclient
+
+            return grpcclient.InferenceServerClient(self.url), grpcclient
+        else:
+            import client_amd.http as httpclient
+
+            return (
+                httpclient.InferenceServerClient(
+                    self.url, concurrency=concurrency
+                ),
+                httpclient,
+            )
+
+    def _model_io(self, client):
+        meta = client.get_model_metadata(self.model_name)
+        if self.protocol == "grpc":
+            inputs = [(i.name, i.datatype, list(i.shape)) for i in meta.inputs]
+            outputs = [(o.name, o.datatype, list(o.shape)) for o in meta.outputs]
+        else:
+            inputs = [(i["name"], i["datatype"], list(i["shape"]))
+                      for i in meta["inputs"]]
+            outputs = [(o["name"], o["datatype"], list(o["shape"]))
+                       for o in meta["outputs"]]
+
+        def concrete(shape):
+            s = [d if d > 0 else (self.batch_size if i == 0 else 16)
+                 for i, d in enumerate(shape)]
+            return s
+
+        inputs = [(n, d, concrete(s)) for n, d, s in inputs]
+        outputs = [(n, d, concrete(s)) for n, d, s in outputs]
+        return inputs, outputs
+
+    def _synth_array(self, datatype, shape):
+        np_dt = triton_to_np_dtype(datatype)
+        if np_dt == np.object_:
+            return np.array(
+                [b"x" * 8] * int(np.prod(shape)), dtype=np.object_
+            ).reshape(shape)
+        if np_dt in (np.float16, np.float32, np.float64):
+            return np.random.rand(*shape).astype(np_dt)
+        return np.random.randint(0, 127, size=shape).astype(np_dt)
+
+    def _setup_wire_slots(self, mod, inputs, outputs, concurrency):
+        """Pre-serialize request objects, one set per slot (reused —
+        mirrors the C++ client's protobuf recycling)."""
+        slots = []
+        for _ in range(concurrency):
+            infer_inputs = []
+            for name, datatype, shape in inputs:
+                ii = mod.InferInput(name, shape, datatype)
+                ii.set_data_from_numpy(self._synth_array(datatype, shape))
+                infer_inputs.append(ii)
+            infer_outputs = [mod.InferRequestedOutput(n) for n, _, _ in outputs]
+            slots.append((infer_inputs, infer_outputs, []))
+        return slots
+
+    def _setup_hipshm_slots(self, client, mod, inputs, outputs, concurrency):
+        import client_amd.utils.hip_shared_memory as hipshm
+
+        slots = []
+        self._regions = []
+        for slot in range(concurrency):
+            infer_inputs = []
+            regions = []
+            staged = []
+            for name, datatype, shape in inputs:
+                elem = TRITON_DTYPE_SIZES.get(datatype, 4)
+                nbytes = int(np.prod(shape)) * elem
+                rname = f"pa_in_{slot}_{name}"
+                region = hipshm.create_shared_memory_region(
+                    rname, nbytes, self.device_id
+                )
+                client.register_cuda_shared_memory(
+                    rname,
+                    hipshm.get_raw_handle_bytes(region)
+                    if self.protocol == "grpc"
+                    else hipshm.get_raw_handle(region),
+                    self.device_id,
+                    nbytes,
+                )
+                data = self._synth_array(
+                    "FP32" if datatype in ("BF16", "FP16") else datatype, shape
+                )
+                if datatype in ("BF16",):
+                    hipshm.set_shared_memory_region_cast(region, data, "BF16")
+                else:
+                    hipshm.set_shared_memory_region(region, [data])
+                ii = mod.InferInput(name, shape, datatype)
+                ii.set_shared_memory(rname, nbytes)
+                infer_inputs.append(ii)
+                regions.append(region)
+                staged.append((region, data, datatype))
+            infer_outputs = []
+            for name, datatype, shape in outputs:
+                elem = TRITON_DTYPE_SIZES.get(datatype, 4)
+                nbytes = int(np.prod(shape)) * elem
+                rname = f"pa_out_{slot}_{name}"
+                region = hipshm.create_shared_memory_region(
+                    rname, nbytes, self.device_id
+                )
+                client.register_cuda_shared_memory(
+                    rname,
+                    hipshm.get_raw_handle_bytes(region)
+                    if self.protocol == "grpc"
+                    else hipshm.get_raw_handle(region),
+                    self.device_id,
+                    nbytes,
+                )
+                io = mod.InferRequestedOutput(name)
+                io.set_shared_memory(rname, nbytes)
+                infer_outputs.append(io)
+                regions.append(region)
+            self._regions.extend(regions)
+            slots.append((infer_inputs, infer_outputs, staged))
+        return slots
+
+    def _teardown_hipshm(self, client):
+        import client_amd.utils.hip_shared_memory as hipshm
+
+        try:
+            client.unregister_cuda_shared_memory()
+        except Exception:
+            pass
+        for region in getattr(self, "_regions", []):
+            try:
+                hipshm.destroy_shared_memory_region(region)
+            except Exception:
+                pass
+        self._regions = []
+
+    # ---- measurement ----
+
+    def run(self, concurrency_list, warmup_s=1.0, window_s=2.0, max_windows=6,
+            stability_pct=10.0):
+        """Sweep concurrency; returns list of result dicts."""
+        results = []
+        for concurrency in concurrency_list:
+            client, mod = self._make_client(concurrency)
+            try:
+                inputs, outputs = self._model_io(client)
+                if self.shared_memory in ("cuda", "hip"):
+                    slots = self._setup_hipshm_slots(
+                        client, mod, inputs, outputs, concurrency
+                    )
+                else:
+                    slots = self._setup_wire_slots(
+                        mod, inputs, outputs, concurrency
+                    )
+
+                if self.protocol == "grpc":
+                    def issue(slot_idx, on_complete):
+                        infer_inputs, infer_outputs, staged = slots[slot_idx]
+                        if self.repack and staged:
+                            import client_amd.utils.hip_shared_memory as hs
+
+                            for region, data, datatype in staged:
+                                if datatype == "BF16":
+                                    hs.set_shared_memory_region_cast(
+                                        region, data, "BF16"
+                                    )
+                                else:
+                                    hs.set_shared_memory_region(region, [data])
+                        client.async_infer(
+                            self.model_name, infer_inputs,
+                            callback=lambda result, error: on_complete(error),
+                            outputs=infer_outputs,
+                        )
+                else:
+                    def issue(slot_idx, on_complete):
+                        infer_inputs, infer_outputs, staged = slots[slot_idx]
+
+                        def _done(req):
+                            try:
+                                req.get_result()
+                                on_complete(None)
+                            except Exception as e:
+                                on_complete(e)
+
+                        req = client.async_infer(
+                            self.model_name, infer_inputs,
+                            outputs=infer_outputs,
+                        )
+                        threading.Thread(target=_done, args=(req,),
+                                         daemon=True).start()
+
+                driver = ConcurrencyDriver(issue, concurrency)
+                throughput, lat, errors, windows = driver.run(
+                    warmup_s, window_s, max_windows, stability_pct
+                )
+                result = {
+                    "concurrency": concurrency,
+                    "batch_size": self.batch_size,
+                    "request_rate_per_sec": round(throughput, 2),
+                    "inferences_per_sec": round(
+                        throughput * self.batch_size, 2
+                    ),
+                    "latency_us": {
+                        "avg": int(np.mean(lat) / 1000) if lat else 0,
+                        "p50": int(percentile(lat, 50) / 1000),
+                        "p90": int(percentile(lat, 90) / 1000),
+                        "p95": int(percentile(lat, 95) / 1000),
+                        "p99": int(percentile(lat, 99) / 1000),
+                    },
+                    "errors": errors,
+                    "measurement_windows": windows,
+                    "shared_memory": self.shared_memory,
+                    "protocol": self.protocol,
+                }
+                results.append(result)
+                if self.verbose:
+                    print(result)
+                if self.shared_memory in ("cuda", "hip"):
+                    self._teardown_hipshm(client)
+            finally:
+                client.close()
+        return results

@@ -1,0 +1,131 @@
+"""Launch a client_amd KServe-v2 server process.
+
+    python -m client_amd.server --grpc-port 8001 --http-port 8000 \
+        --models identity_fp32,simple,resnet50 --device cuda:0
+
+Used by the GPU tests and bench.py: HIP-IPC handles can only be opened
+by a *different* process, so the server always runs out-of-process.
+"""
+
+import argparse
+import signal
+import sys
+import threading
+
+
+def build_core(model_names, device="cuda:0", dtype="bf16"):
+    from . import (
+        AddSubModel,
+        IdentityModel,
+        InferenceCore,
+        RepeatModel,
+        SequenceModel,
+        TorchModel,
+    )
+
+    core = InferenceCore()
+    for name in model_names:
+        if name == "identity_fp32":
+            core.add_model(IdentityModel("identity_fp32", "FP32"))
+        elif name == "identity_bf16":
+            core.add_model(IdentityModel("identity_bf16", "BF16"))
+        elif name == "identity_bytes":
+            core.add_model(IdentityModel("identity_bytes", "BYTES"))
+        elif name == "simple":
+            core.add_model(AddSubModel("simple", "INT32", (-1, 16)))
+        elif name == "simple_string":
+            core.add_model(AddSubModel("simple_string", "BYTES", (-1, 16)))
+        elif name == "sequence_accumulate":
+            core.add_model(SequenceModel())
+        elif name == "repeat_int32":
+            core.add_model(RepeatModel())
+        elif name == "resnet50":
+            import torch
+
+            from ..models import resnet50
+
+            tdt = {"bf16": torch.bfloat16, "fp16": torch.float16,
+                   "fp32": torch.float32}[dtype]
+            io_dt = {"bf16": "BF16", "fp16": "FP16", "fp32": "FP32"}[dtype]
+            core.add_model(
+                TorchModel(
+                    "resnet50",
+                    resnet50(),
+                    inputs=[("INPUT0", io_dt, [-1, 3, 224, 224])],
+                    outputs=[("OUTPUT0", io_dt, [-1, 1000])],
+                    device=device,
+                    dtype=tdt,
+                )
+            )
+        elif name == "identity_gpu":
+            # GPU identity via TorchModel (device fast path test target)
+            import torch
+
+            core.add_model(
+                TorchModel(
+                    "identity_gpu",
+                    torch.nn.Identity(),
+                    inputs=[("INPUT0", "FP32", [-1])],
+                    outputs=[("OUTPUT0", "FP32", [-1])],
+                    device=device,
+                )
+            )
+        else:
+            raise SystemExit(f"unknown model '{name}'")
+    return core
+
+
+def main(argv=None):
+    parser = argparse.ArgumentParser("client_amd.server")
+    parser.add_argument("--http-port", type=int, default=0,
+                        help="0 disables HTTP")
+    parser.add_argument("--grpc-port", type=int, default=0,
+                        help="0 disables gRPC; -1 picks an ephemeral port")
+    parser.add_argument("--host", default="127.0.0.1")
+    parser.add_argument("--models", default="identity_fp32,simple")
+    parser.add_argument("--device", default="cuda:0")
+    parser.add_argument("--dtype", default="bf16",
+                        choices=["bf16", "fp16", "fp32"])
+    parser.add_argument("--grpc-workers", type=int, default=8)
+    args = parser.parse_args(argv)
+
+    core = build_core(
+        [m for m in args.models.split(",") if m], args.device, args.dtype
+    )
+
+    stoppers = []
+    if args.grpc_port != 0:
+        from .grpc_server import GrpcServer
+
+        port = max(args.grpc_port, 0)
+        gs = GrpcServer(core, host=args.host, port=port,
+                        max_workers=args.grpc_workers)
+        gs.start()
+        stoppers.append(lambda: gs.stop(grace=1))
+        print(f"GRPC_READY {gs.port}", flush=True)
+    if args.http_port != 0:
+        from .http_server import HttpServer
+
+        port = max(args.http_port, 0)
+        hs = HttpServer(core, host=args.host, port=port)
+        stop = hs.serve_forever_in_thread()
+        stoppers.append(stop)
+        print(f"HTTP_READY {hs.port}", flush=True)
+
+    done = threading.Event()
+
+    def _sig(*_):
+        done.set()
+
+    signal.signal(signal.SIGTERM, _sig)
+    signal.signal(signal.SIGINT, _sig)
+    done.wait()
+    for stop in stoppers:
+        try:
+            stop()
+        except Exception:
+            pass
+
+
+if __name__ == "__main__":
+    main()

@@ -263,7 +263,9 @@ class InferenceCore:
             if hip_region is not None:
                 from ..ops import hip_runtime as hr
 
-                raw = hr.d2h_bytes(hip_region["ptr"], offset, byte_size)
+                raw = hr.memcpy_d2h(
+                    hip_region["ptr"] + offset, byte_size, hip_region["device_id"]
+                )
                 return self._decode_raw(raw, datatype, shape), binary_cursor
             raise InferenceError(
                 f"Unable to find shared memory region: '{shm_name}'"
@@ -331,19 +333,25 @@ class InferenceCore:
         parameters = dict(request.get("parameters", {}))
 
         try:
-            inputs = {}
-            cursor = 0
-            for inp in request.get("inputs", []):
-                arr, cursor = self._input_array(inp, binary_buf, cursor)
-                inputs[inp["name"]] = arr
-            t1 = time.monotonic_ns()
+            device_result = self._try_device_infer(model, request, parameters)
+            if device_result is not None:
+                response, binary_parts, dt_input, dt_infer = device_result
+                t1 = t0 + dt_input
+                t2 = t1 + dt_infer
+            else:
+                inputs = {}
+                cursor = 0
+                for inp in request.get("inputs", []):
+                    arr, cursor = self._input_array(inp, binary_buf, cursor)
+                    inputs[inp["name"]] = arr
+                t1 = time.monotonic_ns()
 
-            result = model.execute(inputs, parameters)
-            t2 = time.monotonic_ns()
+                result = model.execute(inputs, parameters)
+                t2 = time.monotonic_ns()
 
-            response, binary_parts = self._build_response(
-                model, request, result, parameters
-            )
+                response, binary_parts = self._build_response(
+                    model, request, result, parameters
+                )
             t3 = time.monotonic_ns()
 
             stats.inference_count += 1
@@ -361,6 +369,94 @@ class InferenceCore:
         except Exception as e:
             stats.fail_count += 1
             raise InferenceError(str(e)) from e
+
+    def _try_device_infer(self, model, request, parameters):
+        """Zero-copy HIP-shm fast path: when every input
+        and every requested output lives in a registered HIP-IPC region
+        and the model executes on the GPU, wrap the regions as torch
+        tensors via DLPack (kDLROCM) and never touch the host. Returns
+        (response, binary_parts, t_input_ns, t_infer_ns) or None to fall
+        back to the generic path."""
+        if not hasattr(model, "execute_torch"):
+            return None
+        req_inputs = request.get("inputs", [])
+        req_outputs = request.get("outputs")
+        if not req_inputs or not req_outputs:
+            return None
+        for io in req_inputs + req_outputs:
+            params = io.get("parameters", {})
+            region_name = params.get("shared_memory_region")
+            if region_name is None or self.shm.get_hip(region_name) is None:
+                return None
+
+        import torch
+
+        from ..utils._dlpack import DLDeviceType
+        from ..utils._shared_memory_tensor import SharedMemoryTensor
+
+        def region_tensor(io, datatype, shape):
+            params = io["parameters"]
+            region = self.shm.get_hip(params["shared_memory_region"])
+            offset = params.get("shared_memory_offset", 0)
+            smt = SharedMemoryTensor(
+                datatype=datatype,
+                shape=shape,
+                base_addr=region["ptr"] + offset,
+                byte_offset=0,
+                device_type=DLDeviceType.kDLROCM,
+                device_id=region["device_id"],
+            )
+            return torch.from_dlpack(smt)
+
+        t0 = time.monotonic_ns()
+        tensors = []
+        for io in req_inputs:
+            tensors.append(region_tensor(io, io["datatype"], io["shape"]))
+        t1 = time.monotonic_ns()
+        results = model.execute_torch(tensors)
+        t2 = time.monotonic_ns()
+
+        model_dtypes = {n: d for n, d, _ in model.outputs}
+        response = {
+            "model_name": model.name,
+            "model_version": "1",
+            "outputs": [],
+        }
+        if "id" in request:
+            response["id"] = request["id"]
+        torch_dt = {
+            "FP32": torch.float32, "BF16": torch.bfloat16, "FP16": torch.float16,
+            "INT64": torch.int64, "INT32": torch.int32, "INT8": torch.int8,
+            "UINT8": torch.uint8, "FP64": torch.float64, "BOOL": torch.bool,
+        }
+        by_name = {n: r for (n, _, _), r in zip(model.outputs, results)}
+        for io in req_outputs:
+            name = io["name"]
+            if name not in by_name:
+                raise InferenceError(
+                    f"unexpected inference output '{name}' for model "
+                    f"'{model.name}'"
+                )
+            result = by_name[name]
+            datatype = model_dtypes[name]
+            out_view = region_tensor(
+                {"parameters": io["parameters"]}, datatype, list(result.shape)
+            )
+            out_view.copy_(result.to(torch_dt[datatype]))
+            params = io["parameters"]
+            response["outputs"].append({
+                "name": name,
+                "datatype": datatype,
+                "shape": list(result.shape),
+                "parameters": {
+                    "shared_memory_region": params["shared_memory_region"],
+                    "shared_memory_byte_size": params["shared_memory_byte_size"],
+                    **({"shared_memory_offset": params["shared_memory_offset"]}
+                       if params.get("shared_memory_offset") else {}),
+                },
+            })
+        torch.cuda.synchronize()
+        return response, [], t1 - t0, t2 - t1
 
     def _build_response(self, model, request, result, parameters):
         requested = request.get("outputs")
@@ -426,7 +522,10 @@ class InferenceCore:
                         )
                     from ..ops import hip_runtime as hr
 
-                    hr.h2d_bytes(hip_region["ptr"], offset, raw)
+                    hr.memcpy_h2d(
+                        hip_region["ptr"] + offset, raw, len(raw),
+                        hip_region["device_id"],
+                    )
                 out_json["parameters"] = {
                     "shared_memory_region": shm_name,
                     "shared_memory_byte_size": len(raw),

@@ -234,15 +234,18 @@ def test_keepalive_and_timeout():
 def test_infer_timeout(client, grpc_fixture_server):
     # microscopic timeout -> Deadline Exceeded surface
     # (reference client_timeout_test.cc drives every API this way)
-    a = np.zeros((1, 16), dtype=np.int32)
+    # drive a model that sleeps server-side (repeat_int32 DELAY) so the
+    # deadline reliably expires
     inputs = [
-        grpcclient.InferInput("INPUT0", [1, 16], "INT32"),
-        grpcclient.InferInput("INPUT1", [1, 16], "INT32"),
+        grpcclient.InferInput("IN", [1], "INT32"),
+        grpcclient.InferInput("DELAY", [1], "UINT32"),
+        grpcclient.InferInput("WAIT", [1], "UINT32"),
     ]
-    inputs[0].set_data_from_numpy(a)
-    inputs[1].set_data_from_numpy(a)
+    inputs[0].set_data_from_numpy(np.array([1], dtype=np.int32))
+    inputs[1].set_data_from_numpy(np.array([500], dtype=np.uint32))
+    inputs[2].set_data_from_numpy(np.array([0], dtype=np.uint32))
     try:
-        client.infer("simple", inputs, client_timeout=0.000001)
+        client.infer("repeat_int32", inputs, client_timeout=0.05)
         raise AssertionError("expected timeout")
     except InferenceServerException as e:
         assert "DEADLINE" in str(e.status()).upper()

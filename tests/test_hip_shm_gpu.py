@@ -1,0 +1,257 @@
+"""HIP shared-memory + CDNA4 kernel tests (require an MI355X).
+
+GPU tier of SURVEY.md §4: ports of the reference
+tests/test_cuda_shared_memory.py plus numerics checks of the CDNA4
+cast/pack/preprocess kernels against plain fp32 numpy references.
+"""
+
+import multiprocessing as mp
+
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def hipshm():
+    import client_amd.utils.hip_shared_memory as hipshm
+
+    from client_amd.ops import gpu_available
+
+    if not gpu_available():
+        pytest.skip("no HIP device")
+    return hipshm
+
+
+def test_create_destroy(hipshm):
+    h = hipshm.create_shared_memory_region("r0", 4096, 0)
+    assert ("r0", 4096, 0) in hipshm.allocated_shared_memory_regions()
+    hipshm.destroy_shared_memory_region(h)
+    assert ("r0", 4096, 0) not in hipshm.allocated_shared_memory_regions()
+
+
+def test_raw_handle(hipshm):
+    import base64
+
+    h = hipshm.create_shared_memory_region("r1", 4096, 0)
+    try:
+        b64 = hipshm.get_raw_handle(h)
+        raw = base64.b64decode(b64)
+        assert len(raw) == 64  # hipIpcMemHandle_t is 64 bytes
+        assert hipshm.get_raw_handle_bytes(h) == raw
+    finally:
+        hipshm.destroy_shared_memory_region(h)
+
+
+def test_numpy_roundtrip(hipshm):
+    h = hipshm.create_shared_memory_region("r2", 1 << 20, 0)
+    try:
+        x = np.random.rand(13, 7).astype(np.float32)
+        y = np.arange(11, dtype=np.int64)
+        hipshm.set_shared_memory_region(h, [x, y])
+        out_x = hipshm.get_contents_as_numpy(h, np.float32, [13, 7])
+        out_y = hipshm.get_contents_as_numpy(h, np.int64, [11], offset=x.nbytes)
+        np.testing.assert_array_equal(out_x, x)
+        np.testing.assert_array_equal(out_y, y)
+    finally:
+        hipshm.destroy_shared_memory_region(h)
+
+
+def test_bytes_roundtrip(hipshm):
+    h = hipshm.create_shared_memory_region("r3", 1 << 16, 0)
+    try:
+        s = np.array([b"hip", b"", b"\xff\x00shm"], dtype=np.object_)
+        hipshm.set_shared_memory_region(h, [s])
+        out = hipshm.get_contents_as_numpy(h, np.object_, [3])
+        np.testing.assert_array_equal(out, s)
+    finally:
+        hipshm.destroy_shared_memory_region(h)
+
+
+def test_cast_bf16_wire_exact(hipshm):
+    """Device pack kernel must be byte-exact with the CPU wire codec
+    (truncation semantics, reference utils/__init__.py:294-330)."""
+    from client_amd.utils import deserialize_bf16_tensor, serialize_bf16_tensor
+
+    n = 1 << 20
+    x = (np.random.randn(n) * 100).astype(np.float32)
+    h = hipshm.create_shared_memory_region("r4", n * 2, 0)
+    try:
+        nbytes = hipshm.set_shared_memory_region_cast(h, x, "BF16")
+        assert nbytes == n * 2
+        raw = hipshm.get_contents_as_numpy(h, np.uint8, [n * 2])
+        expected = serialize_bf16_tensor(x)
+        np.testing.assert_array_equal(raw, expected)
+        # device unpack matches CPU deserialize
+        back = hipshm.get_contents_cast(h, "BF16", [n])
+        np.testing.assert_array_equal(back, deserialize_bf16_tensor(expected))
+    finally:
+        hipshm.destroy_shared_memory_region(h)
+
+
+def test_cast_bf16_odd_sizes(hipshm):
+    from client_amd.utils import serialize_bf16_tensor
+
+    for n in (1, 7, 8, 9, 255, 1000003):
+        x = np.random.randn(n).astype(np.float32)
+        h = hipshm.create_shared_memory_region("r5", max(n * 2, 16), 0)
+        try:
+            hipshm.set_shared_memory_region_cast(h, x, "BF16")
+            raw = hipshm.get_contents_as_numpy(h, np.uint8, [n * 2])
+            np.testing.assert_array_equal(raw, serialize_bf16_tensor(x),
+                                          err_msg=f"n={n}")
+        finally:
+            hipshm.destroy_shared_memory_region(h)
+
+
+def test_cast_fp8_roundtrip(hipshm):
+    """fp8 e4m3 (OCP fn): verify against torch's float8_e4m3fn cast."""
+    torch = pytest.importorskip("torch")
+    if not hasattr(torch, "float8_e4m3fn"):
+        pytest.skip("torch has no float8_e4m3fn")
+    n = 1 << 16
+    x = (np.random.randn(n) * 4).astype(np.float32)
+    h = hipshm.create_shared_memory_region("r6", n * 4, 0)
+    try:
+        hipshm.set_shared_memory_region_cast(h, x, "FP8E4M3")
+        raw = hipshm.get_contents_as_numpy(h, np.uint8, [n])
+        expected = (
+            torch.from_numpy(x).to(torch.float8_e4m3fn).view(torch.uint8).numpy()
+        )
+        np.testing.assert_array_equal(raw, expected)
+        back = hipshm.get_contents_cast(h, "FP8E4M3", [n])
+        expected_f = torch.from_numpy(x).to(torch.float8_e4m3fn).float().numpy()
+        np.testing.assert_array_equal(back, expected_f)
+    finally:
+        hipshm.destroy_shared_memory_region(h)
+
+
+def test_dlpack_from_torch(hipshm):
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no torch GPU")
+    x = torch.randn(64, 32, device="cuda:0")
+    h = hipshm.create_shared_memory_region("r7", x.numel() * 4, 0)
+    try:
+        hipshm.set_shared_memory_region_from_dlpack(h, [x])
+        out = hipshm.get_contents_as_numpy(h, np.float32, [64, 32])
+        np.testing.assert_array_equal(out, x.cpu().numpy())
+        # non-contiguous device tensor -> gather_pack kernel path
+        xt = x.t()  # strided view (32, 64)
+        hipshm.set_shared_memory_region_from_dlpack(h, [xt])
+        out = hipshm.get_contents_as_numpy(h, np.float32, [32, 64])
+        np.testing.assert_array_equal(out, x.t().cpu().numpy())
+        # host tensor path
+        xc = torch.randn(16, 16)
+        hipshm.set_shared_memory_region_from_dlpack(h, [xc])
+        out = hipshm.get_contents_as_numpy(h, np.float32, [16, 16])
+        np.testing.assert_array_equal(out, xc.numpy())
+    finally:
+        hipshm.destroy_shared_memory_region(h)
+
+
+def test_dlpack_export_zero_copy(hipshm):
+    """Region exported as DLPack (kDLROCM) wraps into torch with no copy."""
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no torch GPU")
+    h = hipshm.create_shared_memory_region("r8", 1024 * 4, 0)
+    try:
+        smt = hipshm.as_shared_memory_tensor(h, "FP32", [1024])
+        t = torch.from_dlpack(smt)
+        assert t.is_cuda and t.shape == (1024,)
+        t.fill_(3.5)
+        torch.cuda.synchronize()
+        out = hipshm.get_contents_as_numpy(h, np.float32, [1024])
+        np.testing.assert_array_equal(out, np.full(1024, 3.5, np.float32))
+    finally:
+        hipshm.destroy_shared_memory_region(h)
+
+
+def _ref_preprocess(img, oh, ow, mode, mean, std):
+    """numpy reference of the kernel's pixel-center bilinear + normalize."""
+    ih, iw, _ = img.shape
+    out = np.empty((3, oh, ow), dtype=np.float32)
+    sy, sx = ih / oh, iw / ow
+    fy = (np.arange(oh) + 0.5) * sy - 0.5
+    fx = (np.arange(ow) + 0.5) * sx - 0.5
+    y0 = np.minimum(np.maximum(0, np.floor(fy).astype(int)), ih - 1)
+    x0 = np.minimum(np.maximum(0, np.floor(fx).astype(int)), iw - 1)
+    y1 = np.minimum(ih - 1, y0 + 1)
+    x1 = np.minimum(iw - 1, x0 + 1)
+    wy = np.where(fy < 0, 0.0, fy - np.floor(fy))[:, None]
+    wx = np.where(fx < 0, 0.0, fx - np.floor(fx))[None, :]
+    for c in range(3):
+        p = img[:, :, c].astype(np.float32)
+        v = ((1 - wy) * ((1 - wx) * p[y0][:, x0] + wx * p[y0][:, x1])
+             + wy * ((1 - wx) * p[y1][:, x0] + wx * p[y1][:, x1]))
+        if mode == 1:
+            v = v / 127.5 - 1.0
+        elif mode == 2:
+            v = v - mean[c]
+        else:
+            v = (v - mean[c]) * std[c]
+        out[c] = v
+    return out
+
+
+@pytest.mark.parametrize("mode", [0, 1, 2])
+def test_image_preprocess_kernel(hipshm, mode):
+    from client_amd.ops import hip_runtime as hr
+
+    ih, iw, oh, ow = 300, 451, 224, 224
+    img = np.random.randint(0, 256, (ih, iw, 3), dtype=np.uint8)
+    mean = [104.0, 117.0, 123.0]
+    std = [1.0, 1.0, 1.0]
+    src = hr.malloc(0, img.nbytes)
+    dst = hr.malloc(0, 3 * oh * ow * 4)
+    try:
+        hr.memcpy_h2d(src, img.reshape(-1), img.nbytes, 0, True)
+        hr.image_preprocess(src, dst, ih, iw, oh, ow, mode, False, mean, std,
+                            0, True)
+        out = np.empty(3 * oh * ow, dtype=np.float32)
+        hr.memcpy_d2h_into(dst, out.view(np.uint8), out.nbytes, 0)
+        ref = _ref_preprocess(img, oh, ow, mode, mean, std)
+        np.testing.assert_allclose(out.reshape(3, oh, ow), ref, atol=1e-3)
+    finally:
+        hr.free(src)
+        hr.free(dst)
+
+
+def _ipc_child(handle_bytes, n, conn):
+    """Child process: open the IPC handle, double the data in-place."""
+    try:
+        from client_amd.ops import hip_runtime as hr
+        import numpy as np
+
+        ptr = hr.ipc_open_mem_handle(handle_bytes)
+        data = np.frombuffer(hr.memcpy_d2h(ptr, n * 4, 0), dtype=np.float32)
+        hr.memcpy_h2d(ptr, (data * 2).view(np.uint8), n * 4, 0, True)
+        hr.ipc_close_mem_handle(ptr)
+        conn.send("ok")
+    except Exception as e:  # pragma: no cover
+        conn.send(f"error: {e}")
+
+
+def test_ipc_cross_process(hipshm):
+    """The actual IPC contract: a second process opens the region via the
+    64-byte handle and mutates it (the server side of SURVEY.md §3.5)."""
+    n = 4096
+    h = hipshm.create_shared_memory_region("ipc0", n * 4, 0)
+    try:
+        x = np.random.rand(n).astype(np.float32)
+        hipshm.set_shared_memory_region(h, [x])
+        raw = hipshm.get_raw_handle_bytes(h)
+        ctx = mp.get_context("spawn")
+        parent, child = ctx.Pipe()
+        p = ctx.Process(target=_ipc_child, args=(raw, n, child))
+        p.start()
+        assert parent.poll(120), "ipc child timed out"
+        msg = parent.recv()
+        p.join(30)
+        assert msg == "ok", msg
+        out = hipshm.get_contents_as_numpy(h, np.float32, [n])
+        np.testing.assert_allclose(out, x * 2, rtol=1e-6)
+    finally:
+        hipshm.destroy_shared_memory_region(h)
