@@ -158,10 +158,12 @@ def main():
                 torch.cuda.synchronize()
                 dist.broadcast(bcast_tensor, src=0)
                 torch.cuda.synchronize()
-            remaining = args.reqs_per_step
-            done_evt = threading.Event()
-            state = {"outstanding": 0, "issued": 0}
-            lock = threading.Lock()
+            # dispatcher model: the main thread issues every request,
+            # bounded by a concurrency semaphore; gRPC completion
+            # callbacks only record + release (never issue RPCs from
+            # completion threads).
+            sem = threading.Semaphore(args.concurrency)
+            errors = []
 
             def issue_one(slot_idx):
                 slot = slots[slot_idx % len(slots)]
@@ -169,37 +171,31 @@ def main():
 
                 def cb(result, error):
                     if error is not None:
-                        raise SystemExit(f"infer error: {error}")
-                    if record:
+                        errors.append(error)
+                    elif record:
                         with lat_lock:
                             latencies.append(time.monotonic_ns() - t0)
-                    with lock:
-                        state["outstanding"] -= 1
-                        if state["issued"] < args.reqs_per_step:
-                            idx = state["issued"]
-                            state["issued"] += 1
-                            state["outstanding"] += 1
-                            should_issue = idx
-                        elif state["outstanding"] == 0:
-                            done_evt.set()
-                            return
-                        else:
-                            return
-                    issue_one(should_issue)
+                    sem.release()
 
                 client.async_infer(
                     "resnet50", slot["inputs"], callback=cb,
                     outputs=slot["outputs"],
                 )
 
-            first = min(args.concurrency, args.reqs_per_step)
-            with lock:
-                state["issued"] = first
-                state["outstanding"] = first
-            for i in range(first):
+            for i in range(args.reqs_per_step):
+                if not sem.acquire(timeout=300):
+                    raise SystemExit("step timed out (issue)")
+                if errors:
+                    raise SystemExit(f"infer error: {errors[0]}")
                 issue_one(i)
-            if not done_evt.wait(timeout=300):
-                raise SystemExit("step timed out")
+            # drain
+            for _ in range(args.concurrency):
+                if not sem.acquire(timeout=300):
+                    raise SystemExit("step timed out (drain)")
+            for _ in range(args.concurrency):
+                sem.release()
+            if errors:
+                raise SystemExit(f"infer error: {errors[0]}")
 
         # warmup
         for _ in range(args.warmup):

@@ -52,48 +52,42 @@ def percentile(sorted_ns, q):
 
 
 class ConcurrencyDriver:
-    """Closed-loop load: ``concurrency`` slots, each slot re-issues as
-    soon as its previous request completes (perf_analyzer's concurrency
-    model)."""
+    """Closed-loop load: ``concurrency`` worker threads, each owning one
+    slot and re-issuing as soon as its previous request completes
+    (perf_analyzer's concurrency model). ``issue_fn(slot)`` BLOCKS until
+    the request completes — never issue new RPCs from completion
+    callbacks (grpcio completion threads must stay free)."""
 
     def __init__(self, issue_fn, concurrency):
         self._issue_fn = issue_fn
         self._concurrency = concurrency
         self._recorder = LatencyRecorder()
         self._stop = threading.Event()
-        self._inflight = threading.Semaphore(0)
         self._done = threading.Event()
         self._active = 0
         self._lock = threading.Lock()
 
-    def _launch(self, slot):
-        if self._stop.is_set():
-            with self._lock:
-                self._active -= 1
-                if self._active == 0:
-                    self._done.set()
-            return
-        start = time.monotonic_ns()
-
-        def on_complete(error=None):
-            self._recorder.record(start, time.monotonic_ns(), error)
-            self._launch(slot)
-
-        try:
-            self._issue_fn(slot, on_complete)
-        except Exception as e:
-            self._recorder.record(start, time.monotonic_ns(), e)
-            with self._lock:
-                self._active -= 1
-                if self._active == 0:
-                    self._done.set()
+    def _worker(self, slot):
+        while not self._stop.is_set():
+            start = time.monotonic_ns()
+            try:
+                self._issue_fn(slot)
+                self._recorder.record(start, time.monotonic_ns())
+            except Exception as e:
+                self._recorder.record(start, time.monotonic_ns(), e)
+                time.sleep(0.01)
+        with self._lock:
+            self._active -= 1
+            if self._active == 0:
+                self._done.set()
 
     def run(self, warmup_s, window_s, max_windows, stability_pct=10.0,
             min_stable=3):
         """Returns (throughput_req_s, latencies_ns_sorted, errors, windows)."""
         self._active = self._concurrency
         for slot in range(self._concurrency):
-            self._launch(slot)
+            threading.Thread(target=self._worker, args=(slot,),
+                             daemon=True).start()
         time.sleep(warmup_s)
         self._recorder.snapshot_and_reset()
         window_results = []
@@ -293,41 +287,45 @@ class PerfAnalyzer:
                         mod, inputs, outputs, concurrency
                     )
 
+                def repack_slot(staged):
+                    if self.repack and staged:
+                        import client_amd.utils.hip_shared_memory as hs
+
+                        for region, data, datatype in staged:
+                            if datatype == "BF16":
+                                hs.set_shared_memory_region_cast(
+                                    region, data, "BF16"
+                                )
+                            else:
+                                hs.set_shared_memory_region(region, [data])
+
                 if self.protocol == "grpc":
-                    def issue(slot_idx, on_complete):
+                    def issue(slot_idx):
                         infer_inputs, infer_outputs, staged = slots[slot_idx]
-                        if self.repack and staged:
-                            import client_amd.utils.hip_shared_memory as hs
+                        repack_slot(staged)
+                        done = threading.Event()
+                        box = {}
 
-                            for region, data, datatype in staged:
-                                if datatype == "BF16":
-                                    hs.set_shared_memory_region_cast(
-                                        region, data, "BF16"
-                                    )
-                                else:
-                                    hs.set_shared_memory_region(region, [data])
+                        def cb(result, error):
+                            box["error"] = error
+                            done.set()
+
                         client.async_infer(
-                            self.model_name, infer_inputs,
-                            callback=lambda result, error: on_complete(error),
+                            self.model_name, infer_inputs, callback=cb,
                             outputs=infer_outputs,
                         )
+                        if not done.wait(timeout=120):
+                            raise TimeoutError("request timed out")
+                        if box["error"] is not None:
+                            raise box["error"]
                 else:
-                    def issue(slot_idx, on_complete):
+                    def issue(slot_idx):
                         infer_inputs, infer_outputs, staged = slots[slot_idx]
-
-                        def _done(req):
-                            try:
-                                req.get_result()
-                                on_complete(None)
-                            except Exception as e:
-                                on_complete(e)
-
-                        req = client.async_infer(
+                        repack_slot(staged)
+                        client.infer(
                             self.model_name, infer_inputs,
                             outputs=infer_outputs,
                         )
-                        threading.Thread(target=_done, args=(req,),
-                                         daemon=True).start()
 
                 driver = ConcurrencyDriver(issue, concurrency)
                 throughput, lat, errors, windows = driver.run(
