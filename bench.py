@@ -35,31 +35,38 @@ OUT_SHAPE = (BATCH, 1000)
 
 def start_server(device_index, port):
     env = dict(os.environ)
+    repo = os.path.dirname(os.path.abspath(__file__))
+    log_dir = os.path.join(repo, "gpurun_out")
+    os.makedirs(log_dir, exist_ok=True)
+    log_path = os.path.join(log_dir, f"server_rank{device_index}.log")
+    log_f = open(log_path, "w")
     proc = subprocess.Popen(
         [sys.executable, "-m", "client_amd.server", "--grpc-port", str(port),
          "--models", "resnet50", "--device", f"cuda:{device_index}",
          "--dtype", "bf16", "--grpc-workers", "8"],
-        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True, env=env,
-        cwd=os.path.dirname(os.path.abspath(__file__)),
+        stdout=log_f, stderr=subprocess.STDOUT, text=True, env=env, cwd=repo,
     )
     deadline = time.time() + 300
     port_actual = None
     while time.time() < deadline:
-        line = proc.stdout.readline()
-        if not line:
-            if proc.poll() is not None:
-                raise RuntimeError("server exited early")
-            time.sleep(0.05)
-            continue
-        if line.startswith("GRPC_READY"):
-            port_actual = int(line.split()[1])
+        if proc.poll() is not None:
+            raise RuntimeError(
+                f"server exited early; log: {open(log_path).read()[-2000:]}"
+            )
+        try:
+            with open(log_path) as f:
+                for line in f:
+                    if line.startswith("GRPC_READY"):
+                        port_actual = int(line.split()[1])
+                        break
+        except FileNotFoundError:
+            pass
+        if port_actual is not None:
             break
+        time.sleep(0.1)
     if port_actual is None:
         proc.terminate()
         raise RuntimeError("server did not become ready")
-    # drain server stdout in the background so it never blocks
-    threading.Thread(target=lambda: [None for _ in proc.stdout],
-                     daemon=True).start()
     return proc, port_actual
 
 

@@ -15,6 +15,14 @@ from ._dlpack import (
 )
 
 
+# The DLManagedTensor struct, its shape array and the ctypes deleter
+# trampoline must outlive the consumer's tensor: the consumer (e.g.
+# torch) keeps the DLManagedTensor* and invokes ->deleter when ITS
+# tensor dies, which can be long after the SharedMemoryTensor is gone.
+# Entries are keyed by the struct address and removed by the deleter.
+_EXPORT_KEEPALIVE = {}
+
+
 class SharedMemoryTensor:
     def __init__(self, datatype, shape, base_addr, byte_offset, device_type,
                  device_id):
@@ -24,8 +32,6 @@ class SharedMemoryTensor:
         self._byte_offset = byte_offset
         self._device_type = device_type
         self._device_id = device_id
-        # keep ctypes arrays alive as long as exported capsules may live
-        self._live = []
 
     def __dlpack__(self, stream=None):
         dl_dtype = triton_to_dlpack_dtype(self._datatype)
@@ -42,14 +48,15 @@ class SharedMemoryTensor:
             byte_offset=self._byte_offset,
         )
         managed.manager_ctx = None
+        box = ctypes.pointer(managed)
+        key = ctypes.addressof(managed)
 
-        def _deleter(handle):
-            pass
+        def _deleter(handle, _key=key):
+            _EXPORT_KEEPALIVE.pop(_key, None)
 
         deleter = DLManagedTensorDeleter(_deleter)
         managed.deleter = deleter
-        box = ctypes.pointer(managed)
-        self._live.append((managed, shape_arr, deleter, box))
+        _EXPORT_KEEPALIVE[key] = (managed, shape_arr, deleter, box)
         capsule = ctypes.pythonapi.PyCapsule_New(
             ctypes.cast(box, ctypes.c_void_p), _c_str_dltensor, None
         )

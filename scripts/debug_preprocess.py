@@ -1,11 +1,16 @@
 """Diagnose image_preprocess kernel vs numpy reference: mismatch count,
 max diff, and the (c,y,x) locations of the worst offenders."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, os.path.join(
+    os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tests"))
+
 import numpy as np
 
 from client_amd.ops import hip_runtime as hr
-import sys
-sys.path.insert(0, "tests")
 from test_hip_shm_gpu import _ref_preprocess  # noqa: E402
 
 ih, iw, oh, ow = 300, 451, 224, 224
