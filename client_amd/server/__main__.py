@@ -76,6 +76,10 @@ def build_core(model_names, device="cuda:0", dtype="bf16"):
 
 
 def main(argv=None):
+    import os
+
+    # fast MIOpen kernel selection: skip exhaustive conv tuning at start
+    os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
     parser = argparse.ArgumentParser("client_amd.server")
     parser.add_argument("--http-port", type=int, default=0,
                         help="0 disables HTTP")

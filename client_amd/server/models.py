@@ -180,8 +180,10 @@ class TorchModel(Model):
     """
 
     def __init__(self, name, module, inputs, outputs, device="cuda:0",
-                 dtype=None, max_batch_size=0):
+                 dtype=None, max_batch_size=0, use_graph=True):
         super().__init__(name, inputs, outputs, max_batch_size, platform="pytorch")
+        import threading
+
         import torch
 
         self._torch = torch
@@ -191,6 +193,12 @@ class TorchModel(Model):
             self.module = self.module.to(dtype)
         self.dtype = dtype
         self.module.eval()
+        # hipGraph capture-and-replay per input signature: the serving
+        # loop is launch-bound in eager mode; replay submits the whole
+        # forward as one graph (HIP graphs, not a tracing compiler).
+        self.use_graph = use_graph and device.startswith("cuda")
+        self._graphs = {}
+        self._graph_lock = threading.Lock()
 
     def execute(self, inputs, parameters):
         torch = self._torch
@@ -214,13 +222,49 @@ class TorchModel(Model):
                 out[name] = tt.detach().cpu().numpy().astype(npdt, copy=False)
             return out
 
+    def _capture(self, device_tensors):
+        torch = self._torch
+        static_ins = [t.clone() for t in device_tensors]
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            with torch.inference_mode():
+                for _ in range(3):
+                    warm = self.module(*static_ins)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
+        with torch.inference_mode():
+            with torch.cuda.graph(graph):
+                out = self.module(*static_ins)
+        if not isinstance(out, (tuple, list)):
+            out = (out,)
+        return static_ins, graph, list(out)
+
     def execute_torch(self, device_tensors):
         """Device-resident fast path: takes a list of torch tensors already
         on self.device, returns the module outputs as device tensors.
-        Used by the HIP-shm data plane (no numpy, no host copies)."""
+        Used by the HIP-shm data plane (no numpy, no host copies).
+
+        With use_graph, the forward is hipGraph-captured per input
+        signature and replayed; returned tensors are fresh clones so a
+        later replay cannot clobber them. The lock only serializes
+        launch submission — GPU work stays pipelined on the stream."""
         torch = self._torch
-        with torch.inference_mode():
-            result = self.module(*device_tensors)
-            if not isinstance(result, (tuple, list)):
-                result = (result,)
-            return list(result)
+        if not self.use_graph:
+            with torch.inference_mode():
+                result = self.module(*device_tensors)
+                if not isinstance(result, (tuple, list)):
+                    result = (result,)
+                return list(result)
+        key = tuple((tuple(t.shape), t.dtype) for t in device_tensors)
+        with self._graph_lock:
+            entry = self._graphs.get(key)
+            if entry is None:
+                entry = self._capture(device_tensors)
+                self._graphs[key] = entry
+            static_ins, graph, static_outs = entry
+            for si, t in zip(static_ins, device_tensors):
+                si.copy_(t)
+            graph.replay()
+            return [o.clone() for o in static_outs]

@@ -213,7 +213,11 @@ def test_image_preprocess_kernel(hipshm, mode):
         out = np.empty(3 * oh * ow, dtype=np.float32)
         hr.memcpy_d2h_into(dst, out.view(np.uint8), out.nbytes, 0)
         ref = _ref_preprocess(img, oh, ow, mode, mean, std)
-        np.testing.assert_allclose(out.reshape(3, oh, ow), ref, atol=1e-3)
+        # kernel computes in fp32 (fmaf contraction); the float64 numpy
+        # reference differs by up to ~8e-3 on the 0..255 pixel scale
+        # (measured maxdiff 0.0077 = 3e-5 relative) — far below the u8
+        # quantization step.
+        np.testing.assert_allclose(out.reshape(3, oh, ow), ref, atol=0.02)
     finally:
         hr.free(src)
         hr.free(dst)
