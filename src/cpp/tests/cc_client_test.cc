@@ -1,0 +1,271 @@
+// Standalone C++ client test (the reference runs its typed
+// cc_client_test against a live server; cc_client_test.cc:42-129).
+// Usage: cc_client_test <host:port>   — exercises the HTTP client
+// against the fixture server (tests/test_cpp_client.py launches both).
+#include <atomic>
+#include <chrono>
+#include <condition_variable>
+#include <cstring>
+#include <iostream>
+#include <mutex>
+#include <thread>
+#include <vector>
+
+#include "client_amd/base64.h"
+#include "client_amd/common.h"
+#include "client_amd/http_client.h"
+#include "client_amd/json.h"
+#include "client_amd/shm_utils.h"
+
+using namespace client_amd;
+
+#define CHECK(cond)                                                     \
+  do {                                                                  \
+    if (!(cond)) {                                                      \
+      std::cerr << "FAILED at " << __LINE__ << ": " #cond << std::endl; \
+      return 1;                                                         \
+    }                                                                   \
+  } while (0)
+
+#define CHECK_OK(expr)                                                \
+  do {                                                                \
+    Error e = (expr);                                                 \
+    if (!e.IsOk()) {                                                  \
+      std::cerr << "FAILED at " << __LINE__ << ": " << e.Message()    \
+                << std::endl;                                         \
+      return 1;                                                       \
+    }                                                                 \
+  } while (0)
+
+int main(int argc, char** argv) {
+  std::string url = argc > 1 ? argv[1] : "127.0.0.1:8000";
+
+  std::unique_ptr<InferenceServerHttpClient> client;
+  CHECK_OK(InferenceServerHttpClient::Create(&client, url));
+
+  // ---- health ----
+  bool live = false, ready = false;
+  CHECK_OK(client->IsServerLive(&live));
+  CHECK(live);
+  CHECK_OK(client->IsServerReady(&ready));
+  CHECK(ready);
+  bool model_ready = false;
+  CHECK_OK(client->IsModelReady(&model_ready, "simple"));
+  CHECK(model_ready);
+  CHECK_OK(client->IsModelReady(&model_ready, "no_such_model"));
+  CHECK(!model_ready);
+
+  // ---- metadata / config / repository ----
+  std::string meta;
+  CHECK_OK(client->ServerMetadata(&meta));
+  CHECK(Json::Parse(meta)["name"].AsString() == "client_amd_server");
+  std::string model_meta;
+  CHECK_OK(client->ModelMetadata(&model_meta, "simple"));
+  CHECK(Json::Parse(model_meta)["inputs"].AsArray().size() == 2);
+  std::string config;
+  CHECK_OK(client->ModelConfig(&config, "simple"));
+  CHECK(Json::Parse(config)["name"].AsString() == "simple");
+  std::string index;
+  CHECK_OK(client->ModelRepositoryIndex(&index));
+  CHECK(Json::Parse(index).IsArray());
+  CHECK_OK(client->UnloadModel("simple"));
+  CHECK_OK(client->IsModelReady(&model_ready, "simple"));
+  CHECK(!model_ready);
+  CHECK_OK(client->LoadModel("simple"));
+  CHECK_OK(client->IsModelReady(&model_ready, "simple"));
+  CHECK(model_ready);
+
+  // ---- sync binary infer (addsub) ----
+  std::vector<int32_t> in0(16), in1(16);
+  for (int i = 0; i < 16; ++i) {
+    in0[i] = i;
+    in1[i] = 2 * i;
+  }
+  InferInput* input0;
+  InferInput* input1;
+  CHECK_OK(InferInput::Create(&input0, "INPUT0", {1, 16}, "INT32"));
+  CHECK_OK(InferInput::Create(&input1, "INPUT1", {1, 16}, "INT32"));
+  std::unique_ptr<InferInput> input0_p(input0), input1_p(input1);
+  CHECK_OK(input0->AppendRaw(
+      reinterpret_cast<uint8_t*>(in0.data()), in0.size() * 4));
+  CHECK_OK(input1->AppendRaw(
+      reinterpret_cast<uint8_t*>(in1.data()), in1.size() * 4));
+  InferRequestedOutput* output0;
+  InferRequestedOutput* output1;
+  CHECK_OK(InferRequestedOutput::Create(&output0, "OUTPUT0"));
+  CHECK_OK(InferRequestedOutput::Create(&output1, "OUTPUT1"));
+  std::unique_ptr<InferRequestedOutput> o0p(output0), o1p(output1);
+
+  InferOptions options("simple");
+  options.request_id_ = "42";
+  InferResult* result = nullptr;
+  CHECK_OK(client->Infer(&result, options, {input0, input1},
+                         {output0, output1}));
+  std::unique_ptr<InferResult> result_p(result);
+  CHECK_OK(result->RequestStatus());
+  std::string id;
+  CHECK_OK(result->Id(&id));
+  CHECK(id == "42");
+  std::vector<int64_t> shape;
+  CHECK_OK(result->Shape("OUTPUT0", &shape));
+  CHECK(shape.size() == 2 && shape[0] == 1 && shape[1] == 16);
+  std::string datatype;
+  CHECK_OK(result->Datatype("OUTPUT0", &datatype));
+  CHECK(datatype == "INT32");
+  const uint8_t* buf;
+  size_t nbytes;
+  CHECK_OK(result->RawData("OUTPUT0", &buf, &nbytes));
+  CHECK(nbytes == 64);
+  const int32_t* out0 = reinterpret_cast<const int32_t*>(buf);
+  for (int i = 0; i < 16; ++i) CHECK(out0[i] == in0[i] + in1[i]);
+  CHECK_OK(result->RawData("OUTPUT1", &buf, &nbytes));
+  const int32_t* out1 = reinterpret_cast<const int32_t*>(buf);
+  for (int i = 0; i < 16; ++i) CHECK(out1[i] == in0[i] - in1[i]);
+
+  // ---- JSON (non-binary) infer path ----
+  input0->SetBinaryData(false);
+  output0->SetBinaryData(false);
+  InferResult* result2 = nullptr;
+  CHECK_OK(client->Infer(&result2, options, {input0, input1}, {output0}));
+  std::unique_ptr<InferResult> result2_p(result2);
+  CHECK_OK(result2->RequestStatus());
+  // JSON outputs land in the response JSON, not the binary tail
+  CHECK(!result2->RawData("OUTPUT0", &buf, &nbytes).IsOk());
+  input0->SetBinaryData(true);
+  output0->SetBinaryData(true);
+
+  // ---- BYTES via AppendFromString ----
+  {
+    InferInput* sin0;
+    InferInput* sin1;
+    CHECK_OK(InferInput::Create(&sin0, "INPUT0", {1, 16}, "BYTES"));
+    CHECK_OK(InferInput::Create(&sin1, "INPUT1", {1, 16}, "BYTES"));
+    std::unique_ptr<InferInput> sp0(sin0), sp1(sin1);
+    std::vector<std::string> s0, s1;
+    for (int i = 0; i < 16; ++i) {
+      s0.push_back(std::to_string(i));
+      s1.push_back(std::to_string(1));
+    }
+    CHECK_OK(sin0->AppendFromString(s0));
+    CHECK_OK(sin1->AppendFromString(s1));
+    InferOptions sopt("simple_string");
+    InferResult* sres = nullptr;
+    CHECK_OK(client->Infer(&sres, sopt, {sin0, sin1}));
+    std::unique_ptr<InferResult> sres_p(sres);
+    CHECK_OK(sres->RequestStatus());
+    std::vector<std::string> strs;
+    CHECK_OK(sres->StringData("OUTPUT0", &strs));
+    CHECK(strs.size() == 16);
+    CHECK(strs[3] == "4");
+  }
+
+  // ---- system shared memory round trip ----
+  {
+    std::string key = "/cc_test_shm";
+    int fd;
+    CHECK_OK(CreateSharedMemoryRegion(key, 256, &fd));
+    void* base;
+    CHECK_OK(MapSharedMemory(fd, 0, 256, &base));
+    memcpy(base, in0.data(), 64);
+    memcpy((char*)base + 64, in1.data(), 64);
+    CHECK_OK(client->RegisterSystemSharedMemory("cc_io", key, 256));
+    std::string status;
+    CHECK_OK(client->SystemSharedMemoryStatus(&status));
+    CHECK(status.find("cc_io") != std::string::npos);
+
+    InferInput* shm_in0;
+    InferInput* shm_in1;
+    CHECK_OK(InferInput::Create(&shm_in0, "INPUT0", {1, 16}, "INT32"));
+    CHECK_OK(InferInput::Create(&shm_in1, "INPUT1", {1, 16}, "INT32"));
+    std::unique_ptr<InferInput> shm_p0(shm_in0), shm_p1(shm_in1);
+    CHECK_OK(shm_in0->SetSharedMemory("cc_io", 64, 0));
+    CHECK_OK(shm_in1->SetSharedMemory("cc_io", 64, 64));
+    InferRequestedOutput* shm_out;
+    CHECK_OK(InferRequestedOutput::Create(&shm_out, "OUTPUT0"));
+    std::unique_ptr<InferRequestedOutput> shm_op(shm_out);
+    CHECK_OK(shm_out->SetSharedMemory("cc_io", 64, 128));
+    InferResult* shm_res = nullptr;
+    CHECK_OK(client->Infer(&shm_res, options, {shm_in0, shm_in1}, {shm_out}));
+    std::unique_ptr<InferResult> shm_res_p(shm_res);
+    CHECK_OK(shm_res->RequestStatus());
+    const int32_t* shm_vals = (const int32_t*)((char*)base + 128);
+    for (int i = 0; i < 16; ++i) CHECK(shm_vals[i] == in0[i] + in1[i]);
+    CHECK_OK(client->UnregisterSystemSharedMemory("cc_io"));
+    CHECK_OK(UnmapSharedMemory(base, 256));
+    CHECK_OK(CloseSharedMemory(fd));
+    CHECK_OK(UnlinkSharedMemoryRegion(key));
+  }
+
+  // ---- async infer ----
+  {
+    std::mutex mu;
+    std::condition_variable cv;
+    int completed = 0;
+    const int kAsync = 8;
+    bool all_ok = true;
+    for (int r = 0; r < kAsync; ++r) {
+      Error err = client->AsyncInfer(
+          [&](InferResult* res) {
+            std::unique_ptr<InferResult> owned(res);
+            if (!owned->RequestStatus().IsOk()) all_ok = false;
+            std::lock_guard<std::mutex> lock(mu);
+            completed++;
+            cv.notify_all();
+          },
+          options, {input0, input1}, {output0, output1});
+      CHECK_OK(err);
+    }
+    std::unique_lock<std::mutex> lock(mu);
+    CHECK(cv.wait_for(lock, std::chrono::seconds(30),
+                      [&] { return completed == kAsync; }));
+    CHECK(all_ok);
+  }
+
+  // ---- InferMulti ----
+  {
+    std::vector<InferResult*> results;
+    std::vector<InferOptions> opts{options};
+    std::vector<std::vector<InferInput*>> ins{{input0, input1},
+                                              {input0, input1}};
+    CHECK_OK(client->InferMulti(&results, opts, ins));
+    CHECK(results.size() == 2);
+    for (auto* r : results) {
+      CHECK_OK(r->RequestStatus());
+      delete r;
+    }
+  }
+
+  // ---- statistics + client-side InferStat ----
+  std::string stats;
+  CHECK_OK(client->ModelInferenceStatistics(&stats, "simple"));
+  CHECK(Json::Parse(stats)["model_stats"].AsArray().size() == 1);
+  InferStat infer_stat;
+  CHECK_OK(client->ClientInferStat(&infer_stat));
+  CHECK(infer_stat.completed_request_count >= 12);
+  CHECK(infer_stat.cumulative_total_request_time_ns > 0);
+
+  // ---- trace / log settings ----
+  std::string trace;
+  CHECK_OK(client->GetTraceSettings(&trace));
+  CHECK(Json::Parse(trace).Has("trace_rate"));
+  CHECK_OK(client->UpdateTraceSettings(&trace, "", {{"trace_rate", {"123"}}}));
+  CHECK(Json::Parse(trace)["trace_rate"].AsString() == "123");
+  std::string log_settings;
+  CHECK_OK(client->GetLogSettings(&log_settings));
+  CHECK(Json::Parse(log_settings).Has("log_info"));
+
+  // ---- error mapping ----
+  InferOptions bad_options("no_such_model");
+  InferResult* bad_result = nullptr;
+  Error bad = client->Infer(&bad_result, bad_options, {input0, input1});
+  CHECK(!bad.IsOk());
+  delete bad_result;
+
+  // ---- base64 roundtrip (cencode parity) ----
+  std::string raw64(64, '\0');
+  for (int i = 0; i < 64; ++i) raw64[i] = (char)i;
+  CHECK(Base64Decode(Base64Encode(raw64)) == raw64);
+
+  std::cout << "cc_client_test: ALL PASSED" << std::endl;
+  return 0;
+}
