@@ -33,7 +33,7 @@ IN_SHAPE = (BATCH, 3, 224, 224)
 OUT_SHAPE = (BATCH, 1000)
 
 
-def start_server(device_index, port):
+def start_server(device_index, port, extra_args=()):
     env = dict(os.environ)
     repo = os.path.dirname(os.path.abspath(__file__))
     log_dir = os.path.join(repo, "gpurun_out")
@@ -43,7 +43,7 @@ def start_server(device_index, port):
     proc = subprocess.Popen(
         [sys.executable, "-m", "client_amd.server", "--grpc-port", str(port),
          "--models", "resnet50", "--device", f"cuda:{device_index}",
-         "--dtype", "bf16", "--grpc-workers", "8"],
+         "--dtype", "bf16", "--grpc-workers", "8", *extra_args],
         stdout=log_f, stderr=subprocess.STDOUT, text=True, env=env, cwd=repo,
     )
     deadline = time.time() + 300
@@ -77,6 +77,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--concurrency", type=int, default=4)
     ap.add_argument("--reqs-per-step", type=int, default=8)
+    ap.add_argument("--no-dynamic-batching", action="store_true")
     args = ap.parse_args()
 
     import torch
@@ -100,7 +101,12 @@ def main():
     import client_amd.utils.hip_shared_memory as hipshm
     import torch.utils.dlpack
 
-    server_proc, port = start_server(local_rank, 8101 + local_rank)
+    server_args = [] if args.no_dynamic_batching else [
+        "--dynamic-batching", "--preferred-batch-size", "32",
+        "--max-queue-delay-us", "400",
+    ]
+    server_proc, port = start_server(local_rank, 8101 + local_rank,
+                                     server_args)
     try:
         client = grpcclient.InferenceServerClient(f"127.0.0.1:{port}")
         for _ in range(120):

@@ -1,0 +1,174 @@
+"""Llama-3-8B-class decoder with KV cache and greedy generation
+(from scratch on torch.nn).
+
+BASELINE.md config 5: decoupled token streaming over gRPC — one
+response per generated token. Architecture follows the Llama-3 shape
+(RMSNorm, RoPE theta 500000, GQA 32 q / 8 kv heads, SwiGLU FFN 14336,
+vocab 128256); weights are random-init (no network for checkpoints) and
+inputs synthetic, which is exactly what the decode-path benchmark
+needs — the compute per token is the real thing.
+"""
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    dim: int = 4096
+    n_layers: int = 32
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    ffn_dim: int = 14336
+    max_seq: int = 4096
+    rope_theta: float = 500000.0
+    norm_eps: float = 1e-5
+
+
+def llama3_8b_config():
+    return LlamaConfig()
+
+
+def llama_tiny_config():
+    return LlamaConfig(vocab_size=256, dim=64, n_layers=2, n_heads=4,
+                       n_kv_heads=2, ffn_dim=128, max_seq=128)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim, eps):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        dt = x.dtype
+        x = x.float()
+        x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
+        return (x * self.weight.float()).to(dt)
+
+
+def precompute_rope(head_dim, max_seq, theta, device):
+    inv = 1.0 / (theta ** (torch.arange(0, head_dim, 2, device=device).float()
+                           / head_dim))
+    t = torch.arange(max_seq, device=device).float()
+    freqs = torch.outer(t, inv)
+    return torch.cos(freqs), torch.sin(freqs)
+
+
+def apply_rope(x, cos, sin, pos):
+    # x: [b, h, s, d]
+    b, h, s, d = x.shape
+    c = cos[pos : pos + s][None, None]  # [1,1,s,d/2]
+    si = sin[pos : pos + s][None, None]
+    x1, x2 = x[..., 0::2], x[..., 1::2]
+    out = torch.empty_like(x)
+    out[..., 0::2] = x1 * c - x2 * si
+    out[..., 1::2] = x1 * si + x2 * c
+    return out
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.head_dim = cfg.dim // cfg.n_heads
+        self.wq = nn.Linear(cfg.dim, cfg.n_heads * self.head_dim, bias=False)
+        self.wk = nn.Linear(cfg.dim, cfg.n_kv_heads * self.head_dim, bias=False)
+        self.wv = nn.Linear(cfg.dim, cfg.n_kv_heads * self.head_dim, bias=False)
+        self.wo = nn.Linear(cfg.n_heads * self.head_dim, cfg.dim, bias=False)
+        self.w1 = nn.Linear(cfg.dim, cfg.ffn_dim, bias=False)  # gate
+        self.w3 = nn.Linear(cfg.dim, cfg.ffn_dim, bias=False)  # up
+        self.w2 = nn.Linear(cfg.ffn_dim, cfg.dim, bias=False)  # down
+        self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.ffn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+
+    def forward(self, x, cos, sin, pos, kv_cache):
+        # x: [b, s, dim]; kv_cache: (k, v) preallocated
+        # [b, n_kv, max_seq, head_dim]
+        b, s, _ = x.shape
+        h = self.attn_norm(x)
+        q = self.wq(h).view(b, s, self.cfg.n_heads, self.head_dim).transpose(1, 2)
+        k = self.wk(h).view(b, s, self.cfg.n_kv_heads, self.head_dim).transpose(1, 2)
+        v = self.wv(h).view(b, s, self.cfg.n_kv_heads, self.head_dim).transpose(1, 2)
+        q = apply_rope(q, cos, sin, pos)
+        k = apply_rope(k, cos, sin, pos)
+        ck, cv = kv_cache
+        ck[:, :, pos : pos + s] = k
+        cv[:, :, pos : pos + s] = v
+        k_all = ck[:, :, : pos + s]
+        v_all = cv[:, :, : pos + s]
+        rep = self.cfg.n_heads // self.cfg.n_kv_heads
+        k_all = k_all.repeat_interleave(rep, dim=1)
+        v_all = v_all.repeat_interleave(rep, dim=1)
+        attn = F.scaled_dot_product_attention(
+            q, k_all, v_all, is_causal=(s > 1)
+        )
+        attn = attn.transpose(1, 2).reshape(b, s, -1)
+        x = x + self.wo(attn)
+        h = self.ffn_norm(x)
+        x = x + self.w2(F.silu(self.w1(h)) * self.w3(h))
+        return x
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.tok = nn.Embedding(cfg.vocab_size, cfg.dim)
+        self.blocks = nn.ModuleList(
+            [LlamaBlock(cfg) for _ in range(cfg.n_layers)]
+        )
+        self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        self._rope = None
+
+    def _get_rope(self, device):
+        if self._rope is None or self._rope[0].device != device:
+            self._rope = precompute_rope(
+                self.cfg.dim // self.cfg.n_heads, self.cfg.max_seq,
+                self.cfg.rope_theta, device,
+            )
+        return self._rope
+
+    def make_kv_cache(self, batch, device, dtype):
+        head_dim = self.cfg.dim // self.cfg.n_heads
+        return [
+            (
+                torch.zeros(batch, self.cfg.n_kv_heads, self.cfg.max_seq,
+                            head_dim, device=device, dtype=dtype),
+                torch.zeros(batch, self.cfg.n_kv_heads, self.cfg.max_seq,
+                            head_dim, device=device, dtype=dtype),
+            )
+            for _ in range(self.cfg.n_layers)
+        ]
+
+    def forward_step(self, input_ids, pos, kv_cache):
+        """input_ids [b, s] starting at position pos; returns logits of
+        the LAST position [b, vocab]."""
+        cos, sin = self._get_rope(input_ids.device)
+        x = self.tok(input_ids)
+        for block, cache in zip(self.blocks, kv_cache):
+            x = block(x, cos, sin, pos, cache)
+        x = self.norm(x[:, -1:])
+        return self.lm_head(x)[:, 0]
+
+    @torch.inference_mode()
+    def generate(self, input_ids, max_new_tokens):
+        """Greedy decode; yields one token id tensor [b] per step."""
+        device = next(self.parameters()).device
+        dtype = next(self.parameters()).dtype
+        input_ids = input_ids.to(device)
+        b, s = input_ids.shape
+        kv_cache = self.make_kv_cache(b, device, dtype)
+        logits = self.forward_step(input_ids, 0, kv_cache)
+        pos = s
+        for _ in range(max_new_tokens):
+            next_tok = logits.argmax(-1)
+            yield next_tok
+            logits = self.forward_step(next_tok[:, None], pos, kv_cache)
+            pos += 1

@@ -12,6 +12,7 @@ from .core import InferenceCore, InferenceError, ShmRegistry
 from .http_server import HttpServer
 from .models import (
     AddSubModel,
+    GenerateModel,
     IdentityModel,
     Model,
     RepeatModel,
@@ -25,6 +26,7 @@ __all__ = [
     "ShmRegistry",
     "HttpServer",
     "Model",
+    "GenerateModel",
     "IdentityModel",
     "AddSubModel",
     "SequenceModel",

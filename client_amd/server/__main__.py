@@ -57,6 +57,50 @@ def build_core(model_names, device="cuda:0", dtype="bf16"):
                     dtype=tdt,
                 )
             )
+        elif name in ("bert_large", "bert_tiny"):
+            import torch
+
+            from ..models.bert import bert_large, bert_tiny
+
+            tdt = {"bf16": torch.bfloat16, "fp16": torch.float16,
+                   "fp32": torch.float32}[dtype]
+            io_dt = {"bf16": "BF16", "fp16": "FP16", "fp32": "FP32"}[dtype]
+            module = bert_large() if name == "bert_large" else bert_tiny()
+            hidden = 1024 if name == "bert_large" else 32
+            core.add_model(
+                TorchModel(
+                    name,
+                    module,
+                    inputs=[("input_ids", "INT64", [-1, -1])],
+                    outputs=[("pooled", io_dt, [-1, hidden])],
+                    device=device,
+                    dtype=tdt if device.startswith("cuda") else None,
+                    use_graph=False,  # dynamic seq lengths
+                )
+            )
+        elif name in ("llama3_8b", "llama_tiny"):
+            import torch
+
+            from . import GenerateModel
+            from ..models.llama import (
+                LlamaModel,
+                llama3_8b_config,
+                llama_tiny_config,
+            )
+
+            cfg = (llama3_8b_config() if name == "llama3_8b"
+                   else llama_tiny_config())
+            tdt = {"bf16": torch.bfloat16, "fp16": torch.float16,
+                   "fp32": torch.float32}[dtype]
+            use_gpu = device.startswith("cuda")
+            # construct directly on the target device: random-init of 8B
+            # params on the host would take minutes
+            with torch.device(device if use_gpu else "cpu"):
+                module = LlamaModel(cfg)
+            core.add_model(
+                GenerateModel(name, module, device=device,
+                              dtype=tdt if use_gpu else None)
+            )
         elif name == "identity_gpu":
             # GPU identity via TorchModel (device fast path test target)
             import torch
@@ -91,11 +135,21 @@ def main(argv=None):
     parser.add_argument("--dtype", default="bf16",
                         choices=["bf16", "fp16", "fp32"])
     parser.add_argument("--grpc-workers", type=int, default=8)
+    parser.add_argument("--dynamic-batching", action="store_true",
+                        help="enable dynamic batching on torch models")
+    parser.add_argument("--preferred-batch-size", type=int, default=32)
+    parser.add_argument("--max-queue-delay-us", type=int, default=500)
     args = parser.parse_args(argv)
 
     core = build_core(
         [m for m in args.models.split(",") if m], args.device, args.dtype
     )
+    if args.dynamic_batching:
+        for model in core.models.values():
+            if hasattr(model, "enable_dynamic_batching"):
+                model.enable_dynamic_batching(
+                    args.preferred_batch_size, args.max_queue_delay_us
+                )
 
     stoppers = []
     if args.grpc_port != 0:

@@ -1,0 +1,107 @@
+"""Dynamic batching for served torch models.
+
+Triton-style request coalescing (the reference's ModelDynamicBatching
+config, model_config.proto / SURVEY.md §2.7): concurrent requests are
+merged along the batch dimension up to ``preferred_batch_size`` (or
+until ``max_queue_delay_us`` expires), executed as one forward, and the
+outputs are split back per request. On the MI355X path both the gather
+(torch.cat) and the scatter (slice copy_) are device-side — requests
+arriving via HIP-shm never touch the host.
+"""
+
+import threading
+import time
+
+
+class DynamicBatcher:
+    def __init__(self, model, preferred_batch_size=32, max_queue_delay_us=500,
+                 max_batch_size=64):
+        self._model = model
+        self._preferred = preferred_batch_size
+        self._delay_s = max_queue_delay_us / 1e6
+        self._max = max_batch_size
+        self._lock = threading.Lock()
+        self._queue = []
+        self._cv = threading.Condition(self._lock)
+        self._worker = threading.Thread(target=self._run, daemon=True)
+        self._alive = True
+        self._worker.start()
+
+    class _Item:
+        __slots__ = ("tensors", "batch", "event", "outputs", "error")
+
+        def __init__(self, tensors, batch):
+            self.tensors = tensors
+            self.batch = batch
+            self.event = threading.Event()
+            self.outputs = None
+            self.error = None
+
+    def infer(self, tensors):
+        """Blocking: returns the list of output tensors for this request's
+        slice. ``tensors`` batch dims must match across inputs."""
+        batch = tensors[0].shape[0]
+        item = self._Item(tensors, batch)
+        with self._cv:
+            self._queue.append(item)
+            self._cv.notify()
+        if not item.event.wait(timeout=300):
+            raise RuntimeError("dynamic batcher timed out")
+        if item.error is not None:
+            raise item.error
+        return item.outputs
+
+    def shutdown(self):
+        with self._cv:
+            self._alive = False
+            self._cv.notify()
+
+    def _run(self):
+        import torch
+
+        while True:
+            with self._cv:
+                while self._alive and not self._queue:
+                    self._cv.wait()
+                if not self._alive:
+                    return
+                items = [self._queue.pop(0)]
+            # linger for more requests up to preferred/max or delay
+            total = items[0].batch
+            deadline = time.monotonic() + self._delay_s
+            while total < self._preferred:
+                remaining = deadline - time.monotonic()
+                with self._cv:
+                    if self._queue:
+                        nxt = self._queue[0]
+                        if total + nxt.batch > self._max:
+                            break
+                        self._queue.pop(0)
+                        items.append(nxt)
+                        total += nxt.batch
+                        continue
+                if remaining <= 0:
+                    break
+                time.sleep(min(remaining, 0.0002))
+
+            try:
+                if len(items) == 1:
+                    outputs = self._model._execute_direct(items[0].tensors)
+                    items[0].outputs = outputs
+                else:
+                    n_inputs = len(items[0].tensors)
+                    merged = [
+                        torch.cat([it.tensors[i] for it in items], dim=0)
+                        for i in range(n_inputs)
+                    ]
+                    outputs = self._model._execute_direct(merged)
+                    off = 0
+                    for it in items:
+                        it.outputs = [o[off : off + it.batch] for o in outputs]
+                        off += it.batch
+                for it in items:
+                    it.event.set()
+            except Exception as e:  # pragma: no cover
+                for it in items:
+                    it.error = e
+                    it.event.set()

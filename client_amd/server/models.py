@@ -171,6 +171,50 @@ class RepeatModel(Model):
         return {"OUT": np.zeros(1, np.int32), "IDX": np.zeros(1, np.uint32)}
 
 
+class GenerateModel(Model):
+    """Decoupled LLM token streaming: one response per generated token
+    (the decoupled protocol is the reference's
+    repeat_int32 shape, simple_grpc_custom_repeat.cc:135-176, applied to
+    a real decode loop with KV cache)."""
+
+    def __init__(self, name, llama_module, device="cuda:0", dtype=None):
+        super().__init__(
+            name,
+            [("input_ids", "INT64", [-1]), ("max_tokens", "INT32", [1])],
+            [("token_id", "INT64", [1]), ("index", "INT32", [1])],
+            platform="pytorch",
+            decoupled=True,
+        )
+        import torch
+
+        self._torch = torch
+        self.device = device
+        self.module = llama_module.to(device)
+        if dtype is not None:
+            self.module = self.module.to(dtype)
+        self.module.eval()
+
+    def execute_decoupled(self, inputs, parameters):
+        torch = self._torch
+        input_ids = torch.from_numpy(
+            np.ascontiguousarray(inputs["input_ids"].astype(np.int64))
+        )[None]
+        max_tokens = int(inputs.get("max_tokens", np.array([16]))[0])
+        for idx, tok in enumerate(self.module.generate(input_ids, max_tokens)):
+            yield {
+                "token_id": tok.detach().cpu().numpy().astype(np.int64),
+                "index": np.array([idx], dtype=np.int32),
+            }
+
+    def execute(self, inputs, parameters):
+        for out in self.execute_decoupled(inputs, parameters):
+            return out
+        return {
+            "token_id": np.zeros(1, np.int64),
+            "index": np.zeros(1, np.int32),
+        }
+
+
 class TorchModel(Model):
     """Executes a torch.nn.Module on the configured device.
 
@@ -199,6 +243,19 @@ class TorchModel(Model):
         self.use_graph = use_graph and device.startswith("cuda")
         self._graphs = {}
         self._graph_lock = threading.Lock()
+        self._batcher = None
+
+    def enable_dynamic_batching(self, preferred_batch_size=32,
+                                max_queue_delay_us=500, max_batch_size=64):
+        """Triton-style dynamic batching in front of the forward."""
+        from .batcher import DynamicBatcher
+
+        self._batcher = DynamicBatcher(
+            self, preferred_batch_size, max_queue_delay_us, max_batch_size
+        )
+        if self.max_batch_size == 0:
+            self.max_batch_size = max_batch_size
+        return self
 
     def execute(self, inputs, parameters):
         torch = self._torch
@@ -242,9 +299,15 @@ class TorchModel(Model):
         return static_ins, graph, list(out)
 
     def execute_torch(self, device_tensors):
-        """Device-resident fast path: takes a list of torch tensors already
-        on self.device, returns the module outputs as device tensors.
-        Used by the HIP-shm data plane (no numpy, no host copies).
+        """Device-resident entry point for the HIP-shm data plane. Routes
+        through the dynamic batcher when enabled (the batcher worker
+        calls _execute_direct on the merged batch)."""
+        if self._batcher is not None:
+            return self._batcher.infer(device_tensors)
+        return self._execute_direct(device_tensors)
+
+    def _execute_direct(self, device_tensors):
+        """Run the forward on device tensors (no numpy, no host copies).
 
         With use_graph, the forward is hipGraph-captured per input
         signature and replayed; returned tensors are fresh clones so a
