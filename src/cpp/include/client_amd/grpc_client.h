@@ -1,0 +1,141 @@
+// KServe-v2 gRPC client over the from-scratch h2 transport.
+//
+// API-compatible with the reference's InferenceServerGrpcClient
+// (src/c++/library/grpc_client.h:100-) but with no grpc++/protobuf
+// dependency: messages are hand-encoded (kserve_pb.h) and the wire is
+// the local HTTP/2 implementation (h2.h) + gRPC framing (1-byte flag +
+// 4-byte BE length per message). Supports sync Infer, callback
+// AsyncInfer, and the bi-di ModelStreamInfer stream with decoupled
+// final-response semantics (reference grpc_client.cc:1323-1416).
+#pragma once
+
+#include <memory>
+#include <mutex>
+
+#include "client_amd/common.h"
+#include "client_amd/h2.h"
+#include "client_amd/kserve_pb.h"
+
+namespace client_amd {
+
+class InferResultGrpc;
+
+class InferenceServerGrpcClient : public InferenceServerClient {
+ public:
+  ~InferenceServerGrpcClient() override;
+
+  static Error Create(
+      std::unique_ptr<InferenceServerGrpcClient>* client,
+      const std::string& server_url, bool verbose = false);
+
+  Error IsServerLive(bool* live);
+  Error IsServerReady(bool* ready);
+  Error IsModelReady(
+      bool* ready, const std::string& model_name,
+      const std::string& model_version = "");
+  Error ServerMetadata(kserve::ServerMetadataPb* metadata);
+  Error ModelMetadata(
+      kserve::ModelMetadataPb* metadata, const std::string& model_name,
+      const std::string& model_version = "");
+  Error ModelRepositoryIndex(
+      std::vector<kserve::RepositoryIndexEntryPb>* index);
+  Error LoadModel(const std::string& model_name);
+  Error UnloadModel(const std::string& model_name);
+  Error ModelInferenceStatistics(
+      std::vector<kserve::ModelStatisticsPb>* stats,
+      const std::string& model_name = "", const std::string& version = "");
+
+  Error RegisterSystemSharedMemory(
+      const std::string& name, const std::string& key, size_t byte_size,
+      size_t offset = 0);
+  Error UnregisterSystemSharedMemory(const std::string& name = "");
+  // raw 64-byte hipIpcMemHandle_t bytes on the wire
+  // (grpc_service.proto:1610-1643)
+  Error RegisterCudaSharedMemory(
+      const std::string& name, const std::string& raw_handle,
+      size_t device_id, size_t byte_size);
+  Error UnregisterCudaSharedMemory(const std::string& name = "");
+  Error RegisterHipSharedMemory(
+      const std::string& name, const std::string& raw_handle,
+      size_t device_id, size_t byte_size) {
+    return RegisterCudaSharedMemory(name, raw_handle, device_id, byte_size);
+  }
+
+  Error Infer(
+      InferResult** result, const InferOptions& options,
+      const std::vector<InferInput*>& inputs,
+      const std::vector<const InferRequestedOutput*>& outputs = {});
+
+  Error AsyncInfer(
+      OnCompleteFn callback, const InferOptions& options,
+      const std::vector<InferInput*>& inputs,
+      const std::vector<const InferRequestedOutput*>& outputs = {});
+
+  // Bi-di streaming: one active stream per client (reference rule).
+  Error StartStream(OnCompleteFn stream_callback);
+  Error AsyncStreamInfer(
+      const InferOptions& options, const std::vector<InferInput*>& inputs,
+      const std::vector<const InferRequestedOutput*>& outputs = {});
+  Error StopStream();
+
+ private:
+  InferenceServerGrpcClient(const std::string& url, bool verbose);
+
+  // Unary gRPC call: returns grpc-status + response bytes.
+  Error UnaryCall(
+      const std::string& method, const std::string& request,
+      std::string* response, uint64_t timeout_us = 0);
+  Error AsyncUnaryCall(
+      const std::string& method, const std::string& request,
+      std::function<void(Error, std::string)> on_done,
+      uint64_t timeout_us = 0);
+  kserve::ModelInferRequestPb BuildRequest(
+      const InferOptions& options, const std::vector<InferInput*>& inputs,
+      const std::vector<const InferRequestedOutput*>& outputs);
+  Error EnsureConnected();
+
+  std::string host_;
+  int port_;
+  std::unique_ptr<H2Connection> conn_;
+  std::mutex conn_mu_;
+
+  // active bidi stream state
+  struct BidiState;
+  std::shared_ptr<BidiState> bidi_;
+};
+
+//==============================================================================
+// gRPC inference result: zero-copy views into raw_output_contents
+// (reference InferResultGrpc grpc_client.cc:399-446).
+class InferResultGrpc : public InferResult {
+ public:
+  static void Create(
+      InferResult** result, std::shared_ptr<kserve::ModelInferResponsePb>
+      response, Error status = Error::Success);
+
+  Error ModelName(std::string* name) const override;
+  Error ModelVersion(std::string* version) const override;
+  Error Id(std::string* id) const override;
+  Error Shape(const std::string& output_name,
+              std::vector<int64_t>* shape) const override;
+  Error Datatype(const std::string& output_name,
+                 std::string* datatype) const override;
+  Error RawData(const std::string& output_name, const uint8_t** buf,
+                size_t* byte_size) const override;
+  std::string DebugString() const override;
+  Error RequestStatus() const override { return status_; }
+
+  // decoupled-stream params (reference common.h:534-540)
+  bool IsFinalResponse() const;
+  bool IsNullResponse() const;
+
+ private:
+  InferResultGrpc(std::shared_ptr<kserve::ModelInferResponsePb> response,
+                  Error status);
+  const kserve::InferOutputTensorPb* Find(const std::string& name,
+                                          size_t* raw_index) const;
+  std::shared_ptr<kserve::ModelInferResponsePb> response_;
+  Error status_;
+};
+
+}  // namespace client_amd

@@ -13,18 +13,22 @@ CPP = REPO / "src" / "cpp"
 BUILD = CPP / "build_test"
 
 
-def _compile(name, main_src):
+def _compile(name, main_src, grpc=False):
     BUILD.mkdir(exist_ok=True)
     out = BUILD / name
-    srcs = [CPP / s for s in
-            ("common.cc", "json.cc", "shm_utils.cc", "http_client.cc",
-             "hip_shm.cc")] + [main_src]
+    base = ["common.cc", "json.cc", "shm_utils.cc"]
+    base += (["h2.cc", "kserve_pb.cc", "grpc_client.cc"] if grpc
+             else ["http_client.cc"])
+    base += ["hip_shm.cc"]
+    srcs = [CPP / s for s in base] + [main_src]
     newest = max(p.stat().st_mtime for p in srcs + [CPP / "include" /
                                                     "client_amd" / "common.h"])
     if out.exists() and out.stat().st_mtime > newest:
         return out
     cmd = ["g++", "-std=c++17", "-O1", f"-I{CPP}/include", "-Wall",
            *map(str, srcs), "-o", str(out), "-lpthread", "-lrt"]
+    if grpc:
+        cmd.append("-l:libnghttp2.so.14")
     subprocess.run(cmd, check=True, capture_output=True, text=True)
     return out
 
@@ -56,6 +60,43 @@ def test_cc_client(cc_binaries, http_fixture_server):
 def test_cc_example(cc_binaries, http_fixture_server):
     host, port, _ = http_fixture_server
     _, example_bin, _ = cc_binaries
+    proc = subprocess.run(
+        [str(example_bin), "-u", f"{host}:{port}"], capture_output=True,
+        text=True, timeout=60,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "PASS" in proc.stdout
+
+
+@pytest.fixture(scope="module")
+def cc_grpc_binaries():
+    try:
+        test_bin = _compile("cc_grpc_test", CPP / "tests" / "cc_grpc_test.cc",
+                            grpc=True)
+        example_bin = _compile(
+            "simple_grpc_infer_client",
+            CPP / "examples" / "simple_grpc_infer_client.cc", grpc=True)
+    except subprocess.CalledProcessError as e:
+        pytest.fail(f"C++ gRPC compile failed:\n{e.stderr}")
+    return test_bin, example_bin
+
+
+def test_cc_grpc_client(cc_grpc_binaries, grpc_fixture_server):
+    """The from-scratch HTTP/2 + hand-encoded protobuf gRPC client
+    against the grpcio fixture server."""
+    host, port, _ = grpc_fixture_server
+    test_bin, _ = cc_grpc_binaries
+    proc = subprocess.run(
+        [str(test_bin), f"{host}:{port}"], capture_output=True, text=True,
+        timeout=120,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "ALL PASSED" in proc.stdout
+
+
+def test_cc_grpc_example(cc_grpc_binaries, grpc_fixture_server):
+    host, port, _ = grpc_fixture_server
+    _, example_bin = cc_grpc_binaries
     proc = subprocess.run(
         [str(example_bin), "-u", f"{host}:{port}"], capture_output=True,
         text=True, timeout=60,
