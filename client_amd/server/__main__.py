@@ -57,6 +57,24 @@ def build_core(model_names, device="cuda:0", dtype="bf16"):
                     dtype=tdt,
                 )
             )
+        elif name == "densenet121":
+            import torch
+
+            from ..models import densenet121
+
+            tdt = {"bf16": torch.bfloat16, "fp16": torch.float16,
+                   "fp32": torch.float32}[dtype]
+            io_dt = {"bf16": "BF16", "fp16": "FP16", "fp32": "FP32"}[dtype]
+            core.add_model(
+                TorchModel(
+                    "densenet121",
+                    densenet121(),
+                    inputs=[("INPUT0", io_dt, [-1, 3, 224, 224])],
+                    outputs=[("OUTPUT0", io_dt, [-1, 1000])],
+                    device=device,
+                    dtype=tdt,
+                )
+            )
         elif name in ("bert_large", "bert_tiny"):
             import torch
 

@@ -259,3 +259,43 @@ def test_ipc_cross_process(hipshm):
         np.testing.assert_allclose(out, x * 2, rtol=1e-6)
     finally:
         hipshm.destroy_shared_memory_region(h)
+
+
+def test_rccl_broadcast_region_tensor(hipshm):
+    """The bench.py fan-out path: a HIP-shm region wrapped via DLPack is
+    a valid RCCL collective operand. world_size=1 nccl group (the 8-GPU
+    case is driver-run; this pins the API path)."""
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no torch GPU")
+    import os
+
+    import torch.distributed as dist
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29611")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        n = 1 << 16
+        h = hipshm.create_shared_memory_region("bcast0", n * 2, 0)
+        try:
+            x = np.random.rand(n).astype(np.float32)
+            hipshm.set_shared_memory_region_cast(h, x, "BF16")
+            smt = hipshm.as_shared_memory_tensor(h, "BF16", [n])
+            t = torch.from_dlpack(smt)
+            assert t.dtype == torch.bfloat16 and t.is_cuda
+            dist.broadcast(t, src=0)
+            torch.cuda.synchronize()
+            back = hipshm.get_contents_cast(h, "BF16", [n])
+            from client_amd.utils import (
+                deserialize_bf16_tensor,
+                serialize_bf16_tensor,
+            )
+
+            np.testing.assert_array_equal(
+                back, deserialize_bf16_tensor(serialize_bf16_tensor(x).tobytes())
+            )
+        finally:
+            hipshm.destroy_shared_memory_region(h)
+    finally:
+        dist.destroy_process_group()
