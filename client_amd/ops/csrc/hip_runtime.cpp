@@ -163,6 +163,36 @@ extern "C" __global__ void cast_fp8e4m3_fp32_kernel(const uint8_t* __restrict__ 
   }
 }
 
+// A/B variant: 16 elems/lane (64 B loads, 32 B stores per lane per
+// iteration) — measured against the 8-elem kernel on hardware.
+extern "C" __global__ void cast_fp32_bf16_v2_kernel(
+    const uint32_t* __restrict__ src, uint16_t* __restrict__ dst, long n) {
+  long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 16;
+  long stride = (long)gridDim.x * blockDim.x * 16;
+  for (long i = i0; i + 16 <= n; i += stride) {
+    uint4 a[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      a[j] = *reinterpret_cast<const uint4*>(src + i + 4 * j);
+    uint16_t out[16];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      out[4 * j + 0] = (uint16_t)(a[j].x >> 16);
+      out[4 * j + 1] = (uint16_t)(a[j].y >> 16);
+      out[4 * j + 2] = (uint16_t)(a[j].z >> 16);
+      out[4 * j + 3] = (uint16_t)(a[j].w >> 16);
+    }
+    *reinterpret_cast<uint4*>(dst + i) = *reinterpret_cast<uint4*>(out);
+    *reinterpret_cast<uint4*>(dst + i + 8) =
+        *reinterpret_cast<uint4*>(out + 8);
+  }
+  long tail_start = (n / 16) * 16;
+  long ti = tail_start + (blockIdx.x * blockDim.x + threadIdx.x);
+  if (ti < n && (blockIdx.x * blockDim.x + threadIdx.x) < 16) {
+    dst[ti] = (uint16_t)(src[ti] >> 16);
+  }
+}
+
 // Scalar fallbacks for pointers not 16-byte aligned (region offsets are
 // caller-controlled; hipMalloc bases are 256-B aligned so the vector
 // path is the common case).
@@ -441,6 +471,20 @@ static void cast_fp32_bf16(uintptr_t src, uintptr_t dst, long n, int device,
   }
 }
 
+static void cast_fp32_bf16_v2(uintptr_t src, uintptr_t dst, long n,
+                              int device, bool sync, int stream_idx) {
+  hipStream_t s = get_stream(device, stream_idx);
+  int grid = grid_for((n + 15) / 16);
+  hipLaunchKernelGGL(cast_fp32_bf16_v2_kernel, dim3(grid), dim3(256), 0, s,
+                     reinterpret_cast<const uint32_t*>(src),
+                     reinterpret_cast<uint16_t*>(dst), n);
+  HIP_CHECK(hipGetLastError());
+  if (sync) {
+    py::gil_scoped_release release;
+    HIP_CHECK(hipStreamSynchronize(s));
+  }
+}
+
 static void cast_bf16_fp32(uintptr_t src, uintptr_t dst, long n, int device,
                            bool sync, int stream_idx) {
   hipStream_t s = get_stream(device, stream_idx);
@@ -592,6 +636,9 @@ PYBIND11_MODULE(_hip_c, m) {
   m.def("cast_fp32_bf16", &cast_fp32_bf16, py::arg("src"), py::arg("dst"),
         py::arg("n"), py::arg("device") = 0, py::arg("sync") = true,
         py::arg("stream_idx") = 0);
+  m.def("cast_fp32_bf16_v2", &cast_fp32_bf16_v2, py::arg("src"),
+        py::arg("dst"), py::arg("n"), py::arg("device") = 0,
+        py::arg("sync") = true, py::arg("stream_idx") = 0);
   m.def("cast_bf16_fp32", &cast_bf16_fp32, py::arg("src"), py::arg("dst"),
         py::arg("n"), py::arg("device") = 0, py::arg("sync") = true,
         py::arg("stream_idx") = 0);
