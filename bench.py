@@ -26,6 +26,7 @@ import time
 
 import numpy as np
 
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 BATCH = 8

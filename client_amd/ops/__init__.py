@@ -6,6 +6,13 @@ HIP path is the product); on CPU-only machines importing this package
 is fine and only using a GPU function raises.
 """
 
+import os as _os
+
+# must be in the environment before the HIP runtime initializes:
+# the host driver only supports dmabuf IPC (hipIpc* fails EINVAL
+# otherwise)
+_os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
 from . import build as _build
 
 

@@ -454,14 +454,15 @@ static void device_sync() {
 static void cast_fp32_bf16(uintptr_t src, uintptr_t dst, long n, int device,
                            bool sync, int stream_idx) {
   hipStream_t s = get_stream(device, stream_idx);
-  int grid = grid_for((n + 7) / 8);
   if ((src | dst) & 15) {
     hipLaunchKernelGGL(cast_fp32_bf16_scalar_kernel, dim3(grid_for(n)),
                        dim3(256), 0, s, reinterpret_cast<const uint32_t*>(src),
                        reinterpret_cast<uint16_t*>(dst), n);
   } else {
-    hipLaunchKernelGGL(cast_fp32_bf16_kernel, dim3(grid), dim3(256), 0, s,
-                       reinterpret_cast<const uint32_t*>(src),
+    // 16-elem/lane variant: measured 5222 vs 4941 GB/s for 8-elem
+    // (profiles/kernels_rocprof_r01.txt A/B)
+    hipLaunchKernelGGL(cast_fp32_bf16_v2_kernel, dim3(grid_for((n + 15) / 16)),
+                       dim3(256), 0, s, reinterpret_cast<const uint32_t*>(src),
                        reinterpret_cast<uint16_t*>(dst), n);
   }
   HIP_CHECK(hipGetLastError());

@@ -142,6 +142,9 @@ def main(argv=None):
 
     # fast MIOpen kernel selection: skip exhaustive conv tuning at start
     os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+    # dmabuf IPC is the only mode the host driver supports; must be set
+    # before the HIP runtime initializes or hipIpc* fails EINVAL
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     parser = argparse.ArgumentParser("client_amd.server")
     parser.add_argument("--http-port", type=int, default=0,
                         help="0 disables HTTP")

@@ -226,10 +226,23 @@ def test_image_preprocess_kernel(hipshm, mode):
 def _ipc_child(handle_bytes, n, conn):
     """Child process: open the IPC handle, double the data in-place."""
     try:
+        import os
+
+        os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+        import time
+
         from client_amd.ops import hip_runtime as hr
         import numpy as np
 
-        ptr = hr.ipc_open_mem_handle(handle_bytes)
+        ptr = None
+        for attempt in range(3):
+            try:
+                ptr = hr.ipc_open_mem_handle(handle_bytes)
+                break
+            except RuntimeError:
+                if attempt == 2:
+                    raise
+                time.sleep(1.0)
         data = np.frombuffer(hr.memcpy_d2h(ptr, n * 4, 0), dtype=np.float32)
         hr.memcpy_h2d(ptr, (data * 2).view(np.uint8), n * 4, 0, True)
         hr.ipc_close_mem_handle(ptr)
