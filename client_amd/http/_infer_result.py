@@ -25,6 +25,10 @@ class InferResult:
     def __init__(self, response, verbose):
         """``response`` is an object with ``.headers`` (case-insensitive
         mapping) and ``.read()`` returning the raw body bytes."""
+        # keep response headers reachable: the ORCA per-response load
+        # metrics ride in `endpoint-load-metrics[-format]`
+        # (reference README.md:352-366)
+        self._headers = response.headers
         header_length_str = response.headers.get(
             "Inference-Header-Content-Length", None
         )
@@ -134,3 +138,11 @@ class InferResult:
     def get_response(self):
         """The full parsed JSON response."""
         return self._result
+
+    def get_response_header(self, name, default=None):
+        """A raw response header (e.g. the ORCA
+        ``endpoint-load-metrics`` reported per response)."""
+        try:
+            return self._headers.get(name, default)
+        except AttributeError:
+            return default

@@ -577,6 +577,76 @@ Error InferenceServerGrpcClient::Infer(
   return res->RequestStatus();
 }
 
+Error InferenceServerGrpcClient::InferMulti(
+    std::vector<InferResult*>* results,
+    const std::vector<InferOptions>& options,
+    const std::vector<std::vector<InferInput*>>& inputs,
+    const std::vector<std::vector<const InferRequestedOutput*>>& outputs) {
+  if (options.size() != 1 && options.size() != inputs.size()) {
+    return Error("'options' must be of size 1 or match 'inputs'");
+  }
+  if (!outputs.empty() && outputs.size() != 1 &&
+      outputs.size() != inputs.size()) {
+    return Error("'outputs' must be empty, size 1, or match 'inputs'");
+  }
+  results->clear();
+  for (size_t i = 0; i < inputs.size(); ++i) {
+    const InferOptions& opt = options.size() == 1 ? options[0] : options[i];
+    std::vector<const InferRequestedOutput*> outs;
+    if (!outputs.empty())
+      outs = outputs.size() == 1 ? outputs[0] : outputs[i];
+    InferResult* result = nullptr;
+    Error err = Infer(&result, opt, inputs[i], outs);
+    results->push_back(result);
+    if (!err.IsOk()) return err;
+  }
+  return Error::Success;
+}
+
+Error InferenceServerGrpcClient::AsyncInferMulti(
+    OnMultiCompleteFn callback, const std::vector<InferOptions>& options,
+    const std::vector<std::vector<InferInput*>>& inputs,
+    const std::vector<std::vector<const InferRequestedOutput*>>& outputs) {
+  if (options.size() != 1 && options.size() != inputs.size()) {
+    return Error("'options' must be of size 1 or match 'inputs'");
+  }
+  if (!outputs.empty() && outputs.size() != 1 &&
+      outputs.size() != inputs.size()) {
+    return Error("'outputs' must be empty, size 1, or match 'inputs'");
+  }
+  // atomic countdown join -> single callback
+  // (reference grpc_client.cc:1283-1302)
+  struct MultiState {
+    std::mutex mu;
+    std::vector<InferResult*> results;
+    size_t remaining;
+    OnMultiCompleteFn callback;
+  };
+  auto state = std::make_shared<MultiState>();
+  state->results.resize(inputs.size(), nullptr);
+  state->remaining = inputs.size();
+  state->callback = std::move(callback);
+  for (size_t i = 0; i < inputs.size(); ++i) {
+    const InferOptions& opt = options.size() == 1 ? options[0] : options[i];
+    std::vector<const InferRequestedOutput*> outs;
+    if (!outputs.empty())
+      outs = outputs.size() == 1 ? outputs[0] : outputs[i];
+    Error err = AsyncInfer(
+        [state, i](InferResult* result) {
+          bool fire = false;
+          {
+            std::lock_guard<std::mutex> lock(state->mu);
+            state->results[i] = result;
+            fire = (--state->remaining == 0);
+          }
+          if (fire) state->callback(state->results);
+        },
+        opt, inputs[i], outs);
+    if (!err.IsOk()) return err;
+  }
+  return Error::Success;
+}
+
 // ---- bi-di streaming ----
 
 Error InferenceServerGrpcClient::StartStream(OnCompleteFn stream_callback) {

@@ -215,6 +215,45 @@ int main(int argc, char** argv) {
   CHECK(stats.size() == 1);
   CHECK(stats[0].inference_count >= 10);
 
+  // ---- InferMulti ----
+  {
+    std::vector<InferResult*> results;
+    std::vector<InferOptions> opts{options};
+    std::vector<std::vector<InferInput*>> ins{{input0, input1},
+                                              {input0, input1}};
+    CHECK_OK(client->InferMulti(&results, opts, ins));
+    CHECK(results.size() == 2);
+    for (auto* r : results) {
+      CHECK_OK(r->RequestStatus());
+      delete r;
+    }
+  }
+
+  // ---- client timeout (reference client_timeout_test.cc pattern) ----
+  {
+    InferInput* din;
+    InferInput* ddel;
+    InferInput* dwait;
+    CHECK_OK(InferInput::Create(&din, "IN", {1}, "INT32"));
+    CHECK_OK(InferInput::Create(&ddel, "DELAY", {1}, "UINT32"));
+    CHECK_OK(InferInput::Create(&dwait, "WAIT", {1}, "UINT32"));
+    std::unique_ptr<InferInput> dp0(din), dp1(ddel), dp2(dwait);
+    int32_t one = 1;
+    uint32_t delay_ms = 500, wait0 = 0;
+    CHECK_OK(din->AppendRaw((uint8_t*)&one, 4));
+    CHECK_OK(ddel->AppendRaw((uint8_t*)&delay_ms, 4));
+    CHECK_OK(dwait->AppendRaw((uint8_t*)&wait0, 4));
+    InferOptions topt("repeat_int32");
+    topt.client_timeout_ = 50000;  // 50 ms in usec
+    InferResult* tres = nullptr;
+    Error terr = client->Infer(&tres, topt, {din, ddel, dwait});
+    CHECK(!terr.IsOk());
+    CHECK(terr.Message().find("Deadline") != std::string::npos ||
+          terr.Message().find("DEADLINE") != std::string::npos ||
+          terr.Message().find("deadline") != std::string::npos);
+    delete tres;
+  }
+
   // ---- error mapping ----
   InferOptions bad("no_such_model");
   InferResult* bad_result = nullptr;

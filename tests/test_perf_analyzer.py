@@ -47,3 +47,26 @@ def test_perf_cli(grpc_fixture_server, tmp_path, capsys):
 
     data = json.loads(out.read_text())
     assert data[0]["concurrency"] == 1
+
+
+def test_genai_perf_llm_stream():
+    """genai-perf-class LLM metrics against the tiny llama served on CPU."""
+    from client_amd.perf.genai import GenAiPerf
+    from client_amd.server.__main__ import build_core
+    from client_amd.server.grpc_server import GrpcServer
+
+    core = build_core(["llama_tiny"], device="cpu", dtype="fp32")
+    server = GrpcServer(core, host="127.0.0.1", port=0)
+    server.start()
+    try:
+        ga = GenAiPerf(
+            url=f"127.0.0.1:{server.port}", model_name="llama_tiny",
+            prompt_tokens=8, output_tokens=4, vocab_size=256,
+        )
+        result = ga.run(concurrency=2, requests_per_stream=2)
+        assert result["errors"] == 0
+        assert result["total_output_tokens"] == 2 * 2 * 4
+        assert result["ttft_ms"]["p50"] > 0
+        assert result["output_tokens_per_sec"] > 0
+    finally:
+        server.stop(grace=1)
