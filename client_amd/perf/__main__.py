@@ -26,6 +26,8 @@ def main(argv=None):
     p.add_argument("--warmup", type=float, default=1.0)
     p.add_argument("--max-windows", type=int, default=6)
     p.add_argument("--stability-percentage", type=float, default=10.0)
+    p.add_argument("--shape", action="append", default=[],
+                   help="NAME:d1,d2 override for dynamic input dims")
     p.add_argument("--json", default=None, help="write results to file")
     p.add_argument("-v", "--verbose", action="store_true")
     args = p.parse_args(argv)
@@ -38,6 +40,10 @@ def main(argv=None):
 
     from .analyzer import PerfAnalyzer
 
+    shapes = {}
+    for spec in args.shape:
+        name, _, dims = spec.partition(":")
+        shapes[name] = [int(d) for d in dims.split(",") if d]
     pa = PerfAnalyzer(
         url=args.url,
         protocol=args.protocol,
@@ -46,6 +52,7 @@ def main(argv=None):
         shared_memory=args.shared_memory,
         repack=args.repack,
         verbose=args.verbose,
+        shapes=shapes,
     )
     results = pa.run(
         concurrency_list,

@@ -120,7 +120,8 @@ class ConcurrencyDriver:
 class PerfAnalyzer:
     def __init__(self, url, protocol="grpc", model_name="identity_fp32",
                  batch_size=1, shared_memory="none", input_dtype=None,
-                 repack=False, verbose=False, device_id=0):
+                 repack=False, verbose=False, device_id=0, shapes=None,
+                 int_range=(0, 127)):
         self.url = url
         self.protocol = protocol
         self.model_name = model_name
@@ -130,6 +131,9 @@ class PerfAnalyzer:
         self.verbose = verbose
         self.device_id = device_id
         self.input_dtype = input_dtype
+        # perf_analyzer-style --shape NAME:d1,d2 overrides for dynamic dims
+        self.shapes = shapes or {}
+        self.int_range = int_range
         self._client = None
         self._slots = []
 
@@ -161,13 +165,16 @@ class PerfAnalyzer:
             outputs = [(o["name"], o["datatype"], list(o["shape"]))
                        for o in meta["outputs"]]
 
-        def concrete(shape):
-            s = [d if d > 0 else (self.batch_size if i == 0 else 16)
-                 for i, d in enumerate(shape)]
-            return s
+        def concrete(name, shape):
+            if name in self.shapes:
+                return [self.batch_size] + list(self.shapes[name]) \
+                    if len(self.shapes[name]) == len(shape) - 1 \
+                    else list(self.shapes[name])
+            return [d if d > 0 else (self.batch_size if i == 0 else 16)
+                    for i, d in enumerate(shape)]
 
-        inputs = [(n, d, concrete(s)) for n, d, s in inputs]
-        outputs = [(n, d, concrete(s)) for n, d, s in outputs]
+        inputs = [(n, d, concrete(n, s)) for n, d, s in inputs]
+        outputs = [(n, d, concrete(n, s)) for n, d, s in outputs]
         return inputs, outputs
 
     def _synth_array(self, datatype, shape):
@@ -178,7 +185,8 @@ class PerfAnalyzer:
             ).reshape(shape)
         if np_dt in (np.float16, np.float32, np.float64):
             return np.random.rand(*shape).astype(np_dt)
-        return np.random.randint(0, 127, size=shape).astype(np_dt)
+        lo, hi = self.int_range
+        return np.random.randint(lo, hi, size=shape).astype(np_dt)
 
     def _setup_wire_slots(self, mod, inputs, outputs, concurrency):
         """Pre-serialize request objects, one set per slot (reused —
