@@ -157,6 +157,67 @@ class LlamaModel(nn.Module):
         x = self.norm(x[:, -1:])
         return self.lm_head(x)[:, 0]
 
+    def forward_decode_batch(self, tokens, pos_rows, kv_cache):
+        """One decode step for a batch whose rows are at DIFFERENT
+        positions (continuous batching): tokens [b,1], pos_rows int64
+        [b]; kv_cache rows hold each row's history. Returns logits
+        [b, vocab].
+
+        Row independence: RoPE uses per-row offsets and attention is
+        masked to each row's own length, so a row's logits are identical
+        to what a batch-1 decode at its position would produce (verified
+        in tests against sequential generate)."""
+        cos, sin = self._get_rope(tokens.device)
+        b = tokens.shape[0]
+        max_len = int(pos_rows.max().item()) + 1
+        # additive mask: key j visible to row i iff j <= pos_rows[i]
+        key_idx = torch.arange(max_len, device=tokens.device)[None]
+        mask = torch.where(
+            key_idx <= pos_rows[:, None],
+            torch.zeros((), device=tokens.device, dtype=torch.float32),
+            torch.full((), float("-inf"), device=tokens.device,
+                       dtype=torch.float32),
+        )[:, None, None, :]  # [b,1,1,max_len]
+        c_rows = cos[pos_rows][:, None, None, :]  # [b,1,1,hd/2]
+        s_rows = sin[pos_rows][:, None, None, :]
+        x = self.tok(tokens)  # [b,1,dim]
+        ar = torch.arange(b, device=tokens.device)
+        for block, (ck, cv) in zip(self.blocks, kv_cache):
+            h = block.attn_norm(x)
+            q = block.wq(h).view(b, 1, self.cfg.n_heads, block.head_dim
+                                 ).transpose(1, 2)
+            k = block.wk(h).view(b, 1, self.cfg.n_kv_heads, block.head_dim
+                                 ).transpose(1, 2)
+            v = block.wv(h).view(b, 1, self.cfg.n_kv_heads, block.head_dim
+                                 ).transpose(1, 2)
+
+            def rope_rows(t):
+                t1, t2 = t[..., 0::2], t[..., 1::2]
+                out = torch.empty_like(t)
+                out[..., 0::2] = t1 * c_rows - t2 * s_rows
+                out[..., 1::2] = t1 * s_rows + t2 * c_rows
+                return out
+
+            q = rope_rows(q)
+            k = rope_rows(k)
+            # scatter this step's k/v at each row's own position
+            ck[ar, :, pos_rows] = k[:, :, 0]
+            cv[ar, :, pos_rows] = v[:, :, 0]
+            k_all = ck[:, :, :max_len]
+            v_all = cv[:, :, :max_len]
+            rep = self.cfg.n_heads // self.cfg.n_kv_heads
+            k_all = k_all.repeat_interleave(rep, dim=1)
+            v_all = v_all.repeat_interleave(rep, dim=1)
+            attn = F.scaled_dot_product_attention(
+                q, k_all, v_all, attn_mask=mask.to(q.dtype)
+            )
+            attn = attn.transpose(1, 2).reshape(b, 1, -1)
+            x = x + block.wo(attn)
+            h = block.ffn_norm(x)
+            x = x + block.w2(F.silu(block.w1(h)) * block.w3(h))
+        x = self.norm(x)
+        return self.lm_head(x)[:, 0]
+
     @torch.inference_mode()
     def generate(self, input_ids, max_new_tokens):
         """Greedy decode; yields one token id tensor [b] per step."""

@@ -1,0 +1,138 @@
+"""Continuous batched decoding for decoupled LLM serving.
+
+Serialized per-stream decode loops waste the GPU: a decode step's cost
+is dominated by reading the weights out of HBM3E, which is independent
+of batch size until the batch is large. This scheduler keeps ONE decode
+loop over a preallocated batched KV cache; concurrent streams claim a
+slot, are prefilled individually into their cache rows, and then every
+loop iteration advances ALL active rows one token (per-row positions +
+masked attention — LlamaModel.forward_decode_batch). Streams join and
+leave at token boundaries, so N concurrent streams cost ~1 stream of
+wall-clock per token instead of N.
+
+(Measured motivation, r01: serialized streams degrade inter-token
+latency 14 ms -> 99 ms from concurrency 1 -> 4; profiles/genai_*.)
+"""
+
+import queue
+import threading
+
+import torch
+
+
+class _Slot:
+    __slots__ = ("active", "pos", "last_token", "remaining", "out_queue")
+
+    def __init__(self):
+        self.active = False
+        self.pos = 0
+        self.last_token = 0
+        self.remaining = 0
+        self.out_queue = None
+
+
+class DecodeScheduler:
+    END = object()
+
+    def __init__(self, model, max_batch=8, device="cuda:0", dtype=None):
+        self.model = model
+        self.device = device
+        self.dtype = dtype if dtype is not None else next(
+            model.parameters()
+        ).dtype
+        self.max_batch = max_batch
+        self.kv_cache = model.make_kv_cache(max_batch, device, self.dtype)
+        self.slots = [_Slot() for _ in range(max_batch)]
+        self._pending = queue.Queue()
+        self._cv = threading.Condition()
+        self._alive = True
+        self._worker = threading.Thread(target=self._run, daemon=True)
+        self._worker.start()
+
+    def shutdown(self):
+        with self._cv:
+            self._alive = False
+            self._cv.notify()
+
+    def submit(self, input_ids, max_new_tokens):
+        """Returns a queue yielding token ids (ints), then END."""
+        out = queue.Queue()
+        with self._cv:
+            self._pending.put((input_ids, max_new_tokens, out))
+            self._cv.notify()
+        return out
+
+    # ---- worker ----
+
+    def _admit(self):
+        """Prefill pending requests into free slots (one at a time)."""
+        while True:
+            free = [i for i, s in enumerate(self.slots) if not s.active]
+            if not free:
+                return
+            try:
+                input_ids, max_new, out = self._pending.get_nowait()
+            except queue.Empty:
+                return
+            idx = free[0]
+            slot = self.slots[idx]
+            ids = torch.as_tensor(input_ids, dtype=torch.int64,
+                                  device=self.device)[None]
+            s = ids.shape[1]
+            # per-slot cache row views: prefill writes rows [idx:idx+1]
+            row_cache = [(ck[idx : idx + 1], cv[idx : idx + 1])
+                         for ck, cv in self.kv_cache]
+            with torch.inference_mode():
+                logits = self.model.forward_step(ids, 0, row_cache)
+                first = int(logits.argmax(-1)[0])
+            out.put(first)
+            slot.active = True
+            slot.pos = s  # position the NEXT token will be written at
+            slot.last_token = first
+            slot.remaining = max_new - 1
+            slot.out_queue = out
+            if slot.remaining <= 0:
+                slot.active = False
+                out.put(self.END)
+
+    def _decode_step(self):
+        active = [i for i, s in enumerate(self.slots) if s.active]
+        if not active:
+            return False
+        # run the FULL preallocated batch: decode cost is weight-read
+        # bound, so inactive rows are free; masks keep rows independent
+        tokens = torch.tensor(
+            [s.last_token for s in self.slots], dtype=torch.int64,
+            device=self.device,
+        )[:, None]
+        pos_rows = torch.tensor(
+            [max(s.pos, 1) if s.active else 1 for s in self.slots],
+            dtype=torch.int64, device=self.device,
+        )
+        with torch.inference_mode():
+            logits = self.model.forward_decode_batch(
+                tokens, pos_rows, self.kv_cache
+            )
+            next_tokens = logits.argmax(-1).tolist()
+        for i in active:
+            slot = self.slots[i]
+            tok = int(next_tokens[i])
+            slot.out_queue.put(tok)
+            slot.last_token = tok
+            slot.pos += 1
+            slot.remaining -= 1
+            if slot.remaining <= 0 or slot.pos >= self.model.cfg.max_seq - 1:
+                slot.active = False
+                slot.out_queue.put(self.END)
+        return True
+
+    def _run(self):
+        while True:
+            with self._cv:
+                while (self._alive and self._pending.empty()
+                       and not any(s.active for s in self.slots)):
+                    self._cv.wait()
+                if not self._alive:
+                    return
+            self._admit()
+            self._decode_step()

@@ -177,7 +177,8 @@ class GenerateModel(Model):
     repeat_int32 shape, simple_grpc_custom_repeat.cc:135-176, applied to
     a real decode loop with KV cache)."""
 
-    def __init__(self, name, llama_module, device="cuda:0", dtype=None):
+    def __init__(self, name, llama_module, device="cuda:0", dtype=None,
+                 use_scheduler=True, max_batch=8):
         super().__init__(
             name,
             [("input_ids", "INT64", [-1]), ("max_tokens", "INT32", [1])],
@@ -193,13 +194,36 @@ class GenerateModel(Model):
         if dtype is not None:
             self.module = self.module.to(dtype)
         self.module.eval()
+        self._scheduler = None
+        if use_scheduler:
+            from .decode_scheduler import DecodeScheduler
+
+            self._scheduler = DecodeScheduler(
+                self.module, max_batch=max_batch, device=device
+            )
 
     def execute_decoupled(self, inputs, parameters):
         torch = self._torch
+        max_tokens = int(inputs.get("max_tokens", np.array([16]))[0])
+        if self._scheduler is not None:
+            # continuous batching: concurrent streams share one decode
+            # loop (see decode_scheduler.py)
+            ids = np.ascontiguousarray(inputs["input_ids"].astype(np.int64))
+            out = self._scheduler.submit(ids, max_tokens)
+            idx = 0
+            while True:
+                tok = out.get(timeout=600)
+                if tok is self._scheduler.END:
+                    break
+                yield {
+                    "token_id": np.array([tok], dtype=np.int64),
+                    "index": np.array([idx], dtype=np.int32),
+                }
+                idx += 1
+            return
         input_ids = torch.from_numpy(
             np.ascontiguousarray(inputs["input_ids"].astype(np.int64))
         )[None]
-        max_tokens = int(inputs.get("max_tokens", np.array([16]))[0])
         for idx, tok in enumerate(self.module.generate(input_ids, max_tokens)):
             yield {
                 "token_id": tok.detach().cpu().numpy().astype(np.int64),

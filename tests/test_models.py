@@ -132,3 +132,82 @@ def test_bert_tiny_served(llama_grpc_server):
         client.close()
     finally:
         server.stop(grace=1)
+
+
+def test_decode_scheduler_matches_sequential():
+    """Continuous-batched decode must produce exactly the tokens that
+    per-stream sequential generate produces (row independence of
+    forward_decode_batch)."""
+    from client_amd.models.llama import LlamaModel, llama_tiny_config
+    from client_amd.server.decode_scheduler import DecodeScheduler
+
+    torch.manual_seed(3)
+    cfg = llama_tiny_config()
+    m = LlamaModel(cfg).eval()
+    prompts = [
+        torch.randint(0, cfg.vocab_size, (1, n)) for n in (5, 9, 3)
+    ]
+    expected = [
+        [int(t[0]) for t in m.generate(p, 6)] for p in prompts
+    ]
+
+    sched = DecodeScheduler(m, max_batch=4, device="cpu",
+                            dtype=torch.float32)
+    try:
+        queues = [
+            sched.submit(p[0].numpy(), 6) for p in prompts
+        ]
+        got = []
+        for q in queues:
+            toks = []
+            while True:
+                t = q.get(timeout=60)
+                if t is sched.END:
+                    break
+                toks.append(t)
+            got.append(toks)
+        assert got == expected
+    finally:
+        sched.shutdown()
+
+
+def test_decode_scheduler_concurrent_submit():
+    """Streams submitted while others are mid-decode join the batch and
+    still produce the sequential-equivalent tokens."""
+    import threading
+    import time as _time
+
+    from client_amd.models.llama import LlamaModel, llama_tiny_config
+    from client_amd.server.decode_scheduler import DecodeScheduler
+
+    torch.manual_seed(4)
+    cfg = llama_tiny_config()
+    m = LlamaModel(cfg).eval()
+    prompts = [torch.randint(0, cfg.vocab_size, (1, 4 + i)) for i in range(3)]
+    expected = [[int(t[0]) for t in m.generate(p, 8)] for p in prompts]
+
+    sched = DecodeScheduler(m, max_batch=2, device="cpu",  # forces queuing
+                            dtype=torch.float32)
+    results = [None] * 3
+
+    def run(i):
+        if i:
+            _time.sleep(0.05 * i)
+        q = sched.submit(prompts[i][0].numpy(), 8)
+        toks = []
+        while True:
+            t = q.get(timeout=60)
+            if t is sched.END:
+                break
+            toks.append(t)
+        results[i] = toks
+
+    try:
+        threads = [threading.Thread(target=run, args=(i,)) for i in range(3)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(60)
+        assert results == expected
+    finally:
+        sched.shutdown()
