@@ -12,10 +12,12 @@ from .core import InferenceCore, InferenceError, ShmRegistry
 from .http_server import HttpServer
 from .models import (
     AddSubModel,
+    EnsembleModel,
     GenerateModel,
     IdentityModel,
     Model,
     RepeatModel,
+    PreprocessModel,
     SequenceModel,
     TorchModel,
 )
@@ -27,6 +29,8 @@ __all__ = [
     "HttpServer",
     "Model",
     "GenerateModel",
+    "EnsembleModel",
+    "PreprocessModel",
     "IdentityModel",
     "AddSubModel",
     "SequenceModel",

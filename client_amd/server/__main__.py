@@ -119,6 +119,30 @@ def build_core(model_names, device="cuda:0", dtype="bf16"):
                 GenerateModel(name, module, device=device,
                               dtype=tdt if use_gpu else None)
             )
+        elif name == "ensemble_image":
+            # preprocess (HIP kernel on GPU) -> resnet50 pipeline; the
+            # reference's ensemble_image_client analog. Requires
+            # resnet50 earlier in --models.
+            from . import EnsembleModel, PreprocessModel
+
+            if "resnet50" not in core.models:
+                raise SystemExit("ensemble_image requires resnet50 in --models")
+            classifier = core.models["resnet50"]
+            pre = PreprocessModel("preprocess_inception", device=device
+                                  if device.startswith("cuda") else "cpu")
+            core.add_model(pre)
+            io_dt = {"bf16": "BF16", "fp16": "FP16", "fp32": "FP32"}[dtype]
+            core.add_model(EnsembleModel(
+                "ensemble_image",
+                inputs=[("IMAGE", "UINT8", [-1, -1, 3])],
+                outputs=[("OUTPUT0", io_dt if device.startswith("cuda")
+                          else "FP32", [-1, 1000])],
+                steps=[
+                    (pre, {"IMAGE": "IMAGE"}, {"TENSOR": "preprocessed"}),
+                    (classifier, {"INPUT0": "preprocessed"},
+                     {"OUTPUT0": "OUTPUT0"}),
+                ],
+            ))
         elif name == "identity_gpu":
             # GPU identity via TorchModel (device fast path test target)
             import torch

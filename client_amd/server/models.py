@@ -239,6 +239,101 @@ class GenerateModel(Model):
         }
 
 
+class PreprocessModel(Model):
+    """Server-side image preprocessing: u8 HWC image -> fp32 CHW tensor
+    (bilinear resize + normalize). On a GPU this runs the CDNA4
+    image_preprocess kernel (client_amd.ops); CPU fixture runs numpy.
+    Used standalone or as the first step of an ensemble (the reference's
+    ensemble_image_client sends raw images to a preprocess+classify
+    ensemble)."""
+
+    def __init__(self, name="preprocess", size=224, mode=1,
+                 mean=(104.0, 117.0, 123.0), std=(1.0, 1.0, 1.0),
+                 device="cpu"):
+        super().__init__(
+            name,
+            [("IMAGE", "UINT8", [-1, -1, 3])],
+            [("TENSOR", "FP32", [1, 3, size, size])],
+        )
+        self.size = size
+        self.mode = mode
+        self.mean = list(mean)
+        self.std = list(std)
+        self.device = device
+
+    def execute(self, inputs, parameters):
+        img = inputs["IMAGE"].astype(np.uint8)
+        ih, iw, _ = img.shape
+        oh = ow = self.size
+        if self.device.startswith("cuda"):
+            from ..ops import hip_runtime as hr
+
+            dev = int(self.device.split(":")[1]) if ":" in self.device else 0
+            src = hr.malloc(dev, img.nbytes)
+            dst = hr.malloc(dev, 3 * oh * ow * 4)
+            try:
+                hr.memcpy_h2d(src, img.reshape(-1), img.nbytes, dev, False)
+                hr.image_preprocess(src, dst, ih, iw, oh, ow, self.mode,
+                                    False, self.mean, self.std, dev, True)
+                out = np.empty(3 * oh * ow, dtype=np.float32)
+                hr.memcpy_d2h_into(dst, out.view(np.uint8), out.nbytes, dev)
+                return {"TENSOR": out.reshape(1, 3, oh, ow)}
+            finally:
+                hr.free(src)
+                hr.free(dst)
+        # CPU reference path (same pixel-center bilinear convention)
+        sy, sx = ih / oh, iw / ow
+        fy = (np.arange(oh) + 0.5) * sy - 0.5
+        fx = (np.arange(ow) + 0.5) * sx - 0.5
+        y0 = np.clip(np.floor(fy).astype(int), 0, ih - 1)
+        x0 = np.clip(np.floor(fx).astype(int), 0, iw - 1)
+        y1 = np.minimum(ih - 1, y0 + 1)
+        x1 = np.minimum(iw - 1, x0 + 1)
+        wy = np.where(fy < 0, 0.0, fy - np.floor(fy))[:, None]
+        wx = np.where(fx < 0, 0.0, fx - np.floor(fx))[None, :]
+        out = np.empty((3, oh, ow), dtype=np.float32)
+        for c in range(3):
+            p = img[:, :, c].astype(np.float32)
+            v = ((1 - wy) * ((1 - wx) * p[y0][:, x0] + wx * p[y0][:, x1])
+                 + wy * ((1 - wx) * p[y1][:, x0] + wx * p[y1][:, x1]))
+            if self.mode == 1:
+                v = v / 127.5 - 1.0
+            elif self.mode == 2:
+                v = v - self.mean[c]
+            else:
+                v = (v - self.mean[c]) * self.std[c]
+            out[c] = v
+        return {"TENSOR": out[None]}
+
+
+class EnsembleModel(Model):
+    """Ensemble scheduling: a linear pipeline of member models with
+    tensor-name maps between steps (the core of Triton's ensemble
+    extension; reference example ensemble_image_client feeds raw images
+    to preprocess -> classify).
+
+    steps: list of (model, input_map, output_map) where input_map maps
+    the member's input names to ensemble-scope tensor names and
+    output_map maps member output names to ensemble-scope names.
+    """
+
+    def __init__(self, name, inputs, outputs, steps):
+        super().__init__(name, inputs, outputs, platform="ensemble")
+        self.steps = steps
+
+    def execute(self, inputs, parameters):
+        pool = dict(inputs)
+        for model, input_map, output_map in self.steps:
+            member_inputs = {
+                member_name: pool[ens_name]
+                for member_name, ens_name in input_map.items()
+            }
+            result = model.execute(member_inputs, parameters)
+            for member_name, ens_name in output_map.items():
+                pool[ens_name] = result[member_name]
+        return {name: pool[name] for name, _, _ in self.outputs}
+
+
 class TorchModel(Model):
     """Executes a torch.nn.Module on the configured device.
 

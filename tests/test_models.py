@@ -211,3 +211,45 @@ def test_decode_scheduler_concurrent_submit():
         assert results == expected
     finally:
         sched.shutdown()
+
+
+def test_ensemble_preprocess_classify():
+    """Ensemble pipeline: raw u8 image -> preprocess -> classifier, one
+    request (reference ensemble_image_client shape), CPU."""
+    from client_amd.server import (
+        EnsembleModel,
+        PreprocessModel,
+        InferenceCore,
+        TorchModel,
+    )
+
+    pre = PreprocessModel("pre", size=32)
+    tiny_classifier = TorchModel(
+        "cls", torch.nn.Sequential(torch.nn.Flatten(), torch.nn.Linear(3 * 32 * 32, 10)),
+        inputs=[("INPUT0", "FP32", [-1, 3, 32, 32])],
+        outputs=[("OUTPUT0", "FP32", [-1, 10])],
+        device="cpu", use_graph=False,
+    )
+    ens = EnsembleModel(
+        "ens",
+        inputs=[("IMAGE", "UINT8", [-1, -1, 3])],
+        outputs=[("OUTPUT0", "FP32", [-1, 10])],
+        steps=[
+            (pre, {"IMAGE": "IMAGE"}, {"TENSOR": "t"}),
+            (tiny_classifier, {"INPUT0": "t"}, {"OUTPUT0": "OUTPUT0"}),
+        ],
+    )
+    core = InferenceCore()
+    core.add_model(ens)
+    img = np.random.randint(0, 256, (48, 64, 3), dtype=np.uint8)
+    request = {
+        "inputs": [{
+            "name": "IMAGE", "datatype": "UINT8", "shape": list(img.shape),
+            "parameters": {"binary_data_size": img.nbytes},
+        }],
+        "parameters": {"binary_data_output": True},
+    }
+    response, parts = core.infer("ens", request, img.tobytes())
+    assert response["outputs"][0]["shape"] == [1, 10]
+    out = np.frombuffer(parts[0], dtype=np.float32)
+    assert np.isfinite(out).all()
