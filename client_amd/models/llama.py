@@ -157,7 +157,7 @@ class LlamaModel(nn.Module):
         x = self.norm(x[:, -1:])
         return self.lm_head(x)[:, 0]
 
-    def forward_decode_batch(self, tokens, pos_rows, kv_cache):
+    def forward_decode_batch(self, tokens, pos_rows, kv_cache, max_len=None):
         """One decode step for a batch whose rows are at DIFFERENT
         positions (continuous batching): tokens [b,1], pos_rows int64
         [b]; kv_cache rows hold each row's history. Returns logits
@@ -166,10 +166,15 @@ class LlamaModel(nn.Module):
         Row independence: RoPE uses per-row offsets and attention is
         masked to each row's own length, so a row's logits are identical
         to what a batch-1 decode at its position would produce (verified
-        in tests against sequential generate)."""
+        in tests against sequential generate).
+
+        ``max_len`` may be padded past the true max position (the mask
+        blanks the excess) — fixed buckets make the step's shapes static
+        so it can be hipGraph-captured and replayed."""
         cos, sin = self._get_rope(tokens.device)
         b = tokens.shape[0]
-        max_len = int(pos_rows.max().item()) + 1
+        if max_len is None:
+            max_len = int(pos_rows.max().item()) + 1
         # additive mask: key j visible to row i iff j <= pos_rows[i]
         key_idx = torch.arange(max_len, device=tokens.device)[None]
         mask = torch.where(

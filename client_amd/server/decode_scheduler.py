@@ -34,7 +34,8 @@ class _Slot:
 class DecodeScheduler:
     END = object()
 
-    def __init__(self, model, max_batch=8, device="cuda:0", dtype=None):
+    def __init__(self, model, max_batch=8, device="cuda:0", dtype=None,
+                 use_graph=None, len_bucket=512):
         self.model = model
         self.device = device
         self.dtype = dtype if dtype is not None else next(
@@ -43,6 +44,18 @@ class DecodeScheduler:
         self.max_batch = max_batch
         self.kv_cache = model.make_kv_cache(max_batch, device, self.dtype)
         self.slots = [_Slot() for _ in range(max_batch)]
+        # hipGraph capture of the decode step: pad max_len to fixed
+        # buckets so every shape is static; the mask makes the padding
+        # inert. Inputs/outputs are persistent device tensors that
+        # replays read/write in place.
+        self.use_graph = (device.startswith("cuda")
+                          if use_graph is None else use_graph)
+        self.len_bucket = len_bucket
+        self._tokens_dev = torch.zeros(max_batch, 1, dtype=torch.int64,
+                                       device=device)
+        self._pos_dev = torch.ones(max_batch, dtype=torch.int64,
+                                   device=device)
+        self._graphs = {}  # bucket -> (graph, next_tokens_out)
         self._pending = queue.Queue()
         self._cv = threading.Condition()
         self._alive = True
@@ -95,6 +108,39 @@ class DecodeScheduler:
                 slot.active = False
                 out.put(self.END)
 
+    def _bucket(self, max_len):
+        b = ((max_len + self.len_bucket - 1) // self.len_bucket
+             ) * self.len_bucket
+        return min(b, self.model.cfg.max_seq)
+
+    def _get_graph(self, bucket):
+        entry = self._graphs.get(bucket)
+        if entry is not None:
+            return entry
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            with torch.inference_mode():
+                for _ in range(2):  # warmup (idempotent cache writes)
+                    logits = self.model.forward_decode_batch(
+                        self._tokens_dev, self._pos_dev, self.kv_cache,
+                        max_len=bucket,
+                    )
+                    warm = logits.argmax(-1)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
+        with torch.inference_mode():
+            with torch.cuda.graph(graph):
+                logits = self.model.forward_decode_batch(
+                    self._tokens_dev, self._pos_dev, self.kv_cache,
+                    max_len=bucket,
+                )
+                next_out = logits.argmax(-1)
+        entry = (graph, next_out)
+        self._graphs[bucket] = entry
+        return entry
+
     def _decode_step(self):
         active = [i for i, s in enumerate(self.slots) if s.active]
         if not active:
@@ -109,11 +155,19 @@ class DecodeScheduler:
             [max(s.pos, 1) if s.active else 1 for s in self.slots],
             dtype=torch.int64, device=self.device,
         )
-        with torch.inference_mode():
-            logits = self.model.forward_decode_batch(
-                tokens, pos_rows, self.kv_cache
-            )
-            next_tokens = logits.argmax(-1).tolist()
+        if self.use_graph:
+            self._tokens_dev.copy_(tokens)
+            self._pos_dev.copy_(pos_rows)
+            max_pos = max(s.pos for s in self.slots if s.active)
+            graph, next_out = self._get_graph(self._bucket(max_pos + 1))
+            graph.replay()
+            next_tokens = next_out.tolist()
+        else:
+            with torch.inference_mode():
+                logits = self.model.forward_decode_batch(
+                    tokens, pos_rows, self.kv_cache
+                )
+                next_tokens = logits.argmax(-1).tolist()
         for i in active:
             slot = self.slots[i]
             tok = int(next_tokens[i])

@@ -253,3 +253,29 @@ def test_ensemble_preprocess_classify():
     assert response["outputs"][0]["shape"] == [1, 10]
     out = np.frombuffer(parts[0], dtype=np.float32)
     assert np.isfinite(out).all()
+
+
+def test_decode_batch_padded_maxlen_equivalent():
+    """forward_decode_batch with a padded (bucketed) max_len must equal
+    the exact-length result — the property hipGraph bucketing relies on."""
+    from client_amd.models.llama import LlamaModel, llama_tiny_config
+
+    torch.manual_seed(5)
+    cfg = llama_tiny_config()
+    m = LlamaModel(cfg).eval()
+    b = 3
+    kv1 = m.make_kv_cache(b, "cpu", torch.float32)
+    kv2 = m.make_kv_cache(b, "cpu", torch.float32)
+    with torch.inference_mode():
+        # seed some history at different lengths per row
+        for row, plen in enumerate((4, 7, 2)):
+            ids = torch.randint(0, cfg.vocab_size, (1, plen))
+            row_kv1 = [(ck[row:row+1], cv[row:row+1]) for ck, cv in kv1]
+            row_kv2 = [(ck[row:row+1], cv[row:row+1]) for ck, cv in kv2]
+            m.forward_step(ids, 0, row_kv1)
+            m.forward_step(ids, 0, row_kv2)
+        tokens = torch.randint(0, cfg.vocab_size, (b, 1))
+        pos = torch.tensor([4, 7, 2])
+        exact = m.forward_decode_batch(tokens, pos, kv1)
+        padded = m.forward_decode_batch(tokens, pos, kv2, max_len=64)
+    torch.testing.assert_close(exact, padded, rtol=1e-5, atol=1e-5)
