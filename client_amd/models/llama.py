@@ -102,11 +102,9 @@ class LlamaBlock(nn.Module):
         cv[:, :, pos : pos + s] = v
         k_all = ck[:, :, : pos + s]
         v_all = cv[:, :, : pos + s]
-        rep = self.cfg.n_heads // self.cfg.n_kv_heads
-        k_all = k_all.repeat_interleave(rep, dim=1)
-        v_all = v_all.repeat_interleave(rep, dim=1)
+        # GQA handled inside sdpa: no K/V head materialization
         attn = F.scaled_dot_product_attention(
-            q, k_all, v_all, is_causal=(s > 1)
+            q, k_all, v_all, is_causal=(s > 1), enable_gqa=True
         )
         attn = attn.transpose(1, 2).reshape(b, s, -1)
         x = x + self.wo(attn)
@@ -210,11 +208,10 @@ class LlamaModel(nn.Module):
             cv[ar, :, pos_rows] = v[:, :, 0]
             k_all = ck[:, :, :max_len]
             v_all = cv[:, :, :max_len]
-            rep = self.cfg.n_heads // self.cfg.n_kv_heads
-            k_all = k_all.repeat_interleave(rep, dim=1)
-            v_all = v_all.repeat_interleave(rep, dim=1)
+            # GQA handled inside sdpa: no K/V head materialization
             attn = F.scaled_dot_product_attention(
-                q, k_all, v_all, attn_mask=mask.to(q.dtype)
+                q, k_all, v_all, attn_mask=mask.to(q.dtype),
+                enable_gqa=True,
             )
             attn = attn.transpose(1, 2).reshape(b, 1, -1)
             x = x + block.wo(attn)

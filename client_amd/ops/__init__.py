@@ -26,6 +26,15 @@ class _HipRuntimeProxy:
     def _load(self):
         if self._mod is not None:
             return self._mod
+        # One process, ONE HIP runtime: torch bundles its own
+        # libamdhip64 with the same SONAME as /opt/rocm's. Importing
+        # torch first makes the dynamic linker bind _hip_c to torch's
+        # already-loaded copy; loading ours first breaks torch.cuda in
+        # this process and cross-process IPC against torch processes.
+        try:
+            import torch  # noqa: F401
+        except ImportError:
+            pass
         try:
             from . import _hip_c  # noqa
 
