@@ -193,6 +193,89 @@ extern "C" __global__ void cast_fp32_bf16_v2_kernel(
   }
 }
 
+// Fused RMSNorm for bf16 rows: out = x * rsqrt(mean(x^2)+eps) * w,
+// computed in fp32 (byte-compatible with the torch reference sequence
+// float() -> pow/mean/rsqrt -> mul -> to(bf16), which launches ~7
+// kernels; this is one). One workgroup per row; dim must be a multiple
+// of 8 for the vectorized loads (4096/1024/... in practice).
+extern "C" __global__ void rmsnorm_bf16_kernel(
+    const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
+    uint16_t* __restrict__ out, int dim, float eps) {
+  const int row = blockIdx.x;
+  const uint16_t* xr = x + (long)row * dim;
+  uint16_t* outr = out + (long)row * dim;
+  float acc = 0.f;
+  for (int i = threadIdx.x * 8; i < dim; i += blockDim.x * 8) {
+    uint4 v = *reinterpret_cast<const uint4*>(xr + i);
+    const uint16_t* e = reinterpret_cast<const uint16_t*>(&v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = __uint_as_float((uint32_t)e[j] << 16);
+      acc += f * f;
+    }
+  }
+  // wave + LDS reduction (64-wide waves)
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  __shared__ float warp_sums[16];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  if (lane == 0) warp_sums[wave] = acc;
+  __syncthreads();
+  const int n_waves = (blockDim.x + 63) >> 6;
+  if (threadIdx.x == 0) {
+    float total = 0.f;
+    for (int i = 0; i < n_waves; ++i) total += warp_sums[i];
+    warp_sums[0] = rsqrtf(total / dim + eps);
+  }
+  __syncthreads();
+  const float scale = warp_sums[0];
+  for (int i = threadIdx.x * 8; i < dim; i += blockDim.x * 8) {
+    uint4 v = *reinterpret_cast<const uint4*>(xr + i);
+    uint4 wv = *reinterpret_cast<const uint4*>(w + i);
+    const uint16_t* e = reinterpret_cast<const uint16_t*>(&v);
+    const uint16_t* we = reinterpret_cast<const uint16_t*>(&wv);
+    uint16_t o[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = __uint_as_float((uint32_t)e[j] << 16);
+      float wf = __uint_as_float((uint32_t)we[j] << 16);
+      o[j] = __bfloat16_as_ushort(__float2bfloat16(f * scale * wf));
+    }
+    *reinterpret_cast<uint4*>(outr + i) = *reinterpret_cast<uint4*>(o);
+  }
+}
+
+// Fused decode-step RoPE for bf16 q AND k in one launch, with per-row
+// positions (continuous batching): q [b, hq, d], k [b, hk, d]
+// contiguous (s=1), cos/sin tables [max_seq, d/2] fp32, pos [b] int64.
+// Replaces ~8 slicing/elementwise launches per projection.
+extern "C" __global__ void rope_decode_bf16_kernel(
+    uint16_t* __restrict__ q, uint16_t* __restrict__ k,
+    const float* __restrict__ cos_tab, const float* __restrict__ sin_tab,
+    const long* __restrict__ pos, int b, int hq, int hk, int d) {
+  const int half = d / 2;
+  const long total = (long)b * (hq + hk) * half;
+  long i = (long)(blockIdx.x * blockDim.x + threadIdx.x);
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    long rem = i;
+    const int j = (int)(rem % half);  // rotation pair index
+    rem /= half;
+    const int head = (int)(rem % (hq + hk));
+    const int row = (int)(rem / (hq + hk));
+    uint16_t* base = (head < hq)
+                         ? q + ((long)row * hq + head) * d
+                         : k + ((long)row * hk + (head - hq)) * d;
+    const float c = cos_tab[pos[row] * half + j];
+    const float s = sin_tab[pos[row] * half + j];
+    const float x1 = __uint_as_float((uint32_t)base[2 * j] << 16);
+    const float x2 = __uint_as_float((uint32_t)base[2 * j + 1] << 16);
+    base[2 * j] = __bfloat16_as_ushort(__float2bfloat16(x1 * c - x2 * s));
+    base[2 * j + 1] = __bfloat16_as_ushort(__float2bfloat16(x1 * s + x2 * c));
+  }
+}
+
 // Scalar fallbacks for pointers not 16-byte aligned (region offsets are
 // caller-controlled; hipMalloc bases are 256-B aligned so the vector
 // path is the common case).
@@ -595,6 +678,34 @@ static void gather_pack(uintptr_t src, uintptr_t dst, int elem_size,
   }
 }
 
+static void rmsnorm_bf16(uintptr_t x, uintptr_t w, uintptr_t out,
+                         long rows, int dim, double eps,
+                         uintptr_t stream_handle) {
+  // launches on the CALLER's stream (torch's current stream) so it
+  // composes with torch ops and hipGraph capture
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream_handle);
+  if (dim % 8 != 0) throw std::runtime_error("rmsnorm: dim % 8 != 0");
+  hipLaunchKernelGGL(rmsnorm_bf16_kernel, dim3((uint32_t)rows), dim3(256), 0,
+                     s, reinterpret_cast<const uint16_t*>(x),
+                     reinterpret_cast<const uint16_t*>(w),
+                     reinterpret_cast<uint16_t*>(out), dim, (float)eps);
+  HIP_CHECK(hipGetLastError());
+}
+
+static void rope_decode_bf16(uintptr_t q, uintptr_t k, uintptr_t cos_tab,
+                             uintptr_t sin_tab, uintptr_t pos, int b, int hq,
+                             int hk, int d, uintptr_t stream_handle) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream_handle);
+  long total = (long)b * (hq + hk) * (d / 2);
+  hipLaunchKernelGGL(rope_decode_bf16_kernel, dim3(grid_for(total)),
+                     dim3(256), 0, s, reinterpret_cast<uint16_t*>(q),
+                     reinterpret_cast<uint16_t*>(k),
+                     reinterpret_cast<const float*>(cos_tab),
+                     reinterpret_cast<const float*>(sin_tab),
+                     reinterpret_cast<const long*>(pos), b, hq, hk, d);
+  HIP_CHECK(hipGetLastError());
+}
+
 static void image_preprocess(uintptr_t src, uintptr_t dst, int ih, int iw,
                              int oh, int ow, int mode, bool out_bf16,
                              std::vector<float> mean, std::vector<float> stdev,
@@ -649,6 +760,13 @@ PYBIND11_MODULE(_hip_c, m) {
   m.def("cast_fp8e4m3_fp32", &cast_fp8e4m3_fp32, py::arg("src"), py::arg("dst"),
         py::arg("n"), py::arg("device") = 0, py::arg("sync") = true,
         py::arg("stream_idx") = 0);
+  m.def("rmsnorm_bf16", &rmsnorm_bf16, py::arg("x"), py::arg("w"),
+        py::arg("out"), py::arg("rows"), py::arg("dim"), py::arg("eps"),
+        py::arg("stream_handle"));
+  m.def("rope_decode_bf16", &rope_decode_bf16, py::arg("q"), py::arg("k"),
+        py::arg("cos_tab"), py::arg("sin_tab"), py::arg("pos"), py::arg("b"),
+        py::arg("hq"), py::arg("hk"), py::arg("d"),
+        py::arg("stream_handle"));
   m.def("gather_pack", &gather_pack, py::arg("src"), py::arg("dst"),
         py::arg("elem_size"), py::arg("shape"), py::arg("strides"),
         py::arg("device") = 0, py::arg("sync") = true);

@@ -350,3 +350,59 @@ def test_decode_scheduler_graph_matches_eager(hipshm):
     graph = run(True)
     assert eager == graph
     assert all(len(t) == 8 for t in eager)
+
+
+def test_fused_rmsnorm_and_rope_numerics(hipshm):
+    """Fused decode kernels vs the plain torch reference (bf16, fp32
+    internal math)."""
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no torch GPU")
+    from client_amd.ops import hip_runtime as hr
+
+    torch.manual_seed(11)
+    stream = torch.cuda.current_stream().cuda_stream
+
+    # rmsnorm
+    rows, dim = 8, 4096
+    x = torch.randn(rows, dim, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(dim, device="cuda", dtype=torch.bfloat16)
+    out = torch.empty_like(x)
+    hr.rmsnorm_bf16(x.data_ptr(), w.data_ptr(), out.data_ptr(), rows, dim,
+                    1e-5, stream)
+    torch.cuda.synchronize()
+    xf = x.float()
+    ref = (xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5)
+           * w.float()).to(torch.bfloat16)
+    # reduction order differs (tree vs torch mean): allow 2 ulp of bf16
+    torch.testing.assert_close(out.float(), ref.float(), rtol=2e-2,
+                               atol=2e-2)
+
+    # rope (q+k fused, per-row positions)
+    b, hq, hk, d = 4, 8, 2, 128
+    max_seq = 64
+    inv = 1.0 / (10000 ** (torch.arange(0, d, 2, device="cuda").float() / d))
+    t = torch.arange(max_seq, device="cuda").float()
+    freqs = torch.outer(t, inv)
+    cos, sin = torch.cos(freqs).contiguous(), torch.sin(freqs).contiguous()
+    pos = torch.tensor([3, 17, 0, 42], dtype=torch.int64, device="cuda")
+    q = torch.randn(b, hq, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, hk, d, device="cuda", dtype=torch.bfloat16)
+    q2, k2 = q.clone(), k.clone()
+    hr.rope_decode_bf16(q2.data_ptr(), k2.data_ptr(), cos.data_ptr(),
+                        sin.data_ptr(), pos.data_ptr(), b, hq, hk, d, stream)
+    torch.cuda.synchronize()
+
+    def ref_rope(x, heads):
+        c = cos[pos][:, None, :]  # [b,1,d/2]
+        s = sin[pos][:, None, :]
+        x1, x2 = x[..., 0::2].float(), x[..., 1::2].float()
+        out = torch.empty_like(x)
+        out[..., 0::2] = (x1 * c - x2 * s).to(torch.bfloat16)
+        out[..., 1::2] = (x1 * s + x2 * c).to(torch.bfloat16)
+        return out
+
+    torch.testing.assert_close(q2.float(), ref_rope(q, hq).float(),
+                               rtol=1e-2, atol=1e-2)
+    torch.testing.assert_close(k2.float(), ref_rope(k, hk).float(),
+                               rtol=1e-2, atol=1e-2)
