@@ -35,7 +35,7 @@ class DecodeScheduler:
     END = object()
 
     def __init__(self, model, max_batch=8, device="cuda:0", dtype=None,
-                 use_graph=None, len_bucket=512):
+                 use_graph=None, len_bucket=256):
         self.model = model
         self.device = device
         self.dtype = dtype if dtype is not None else next(
