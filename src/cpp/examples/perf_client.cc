@@ -75,7 +75,27 @@ int main(int argc, char** argv) {
       ca::Json meta = ca::Json::Parse(meta_str);
       std::vector<std::unique_ptr<ca::InferInput>> inputs;
       std::vector<ca::InferInput*> input_ptrs;
+      std::vector<std::unique_ptr<ca::InferRequestedOutput>> outputs;
+      std::vector<const ca::InferRequestedOutput*> output_ptrs;
       std::vector<std::vector<uint8_t>> buffers;
+      // optional system-shm input/output mode: one region per worker,
+      // inputs packed once, requests carry only region references
+      int shm_fd = -1;
+      void* shm_base = nullptr;
+      std::string shm_key = "/perf_client_" + std::to_string(slot);
+      std::string shm_name = "pc_slot_" + std::to_string(slot);
+      size_t shm_cursor = 0;
+      const size_t kShmBytes = 1 << 22;
+      if (args.shm) {
+        if (!ca::CreateSharedMemoryRegion(shm_key, kShmBytes, &shm_fd)
+                 .IsOk() ||
+            !ca::MapSharedMemory(shm_fd, 0, kShmBytes, &shm_base).IsOk() ||
+            !client->RegisterSystemSharedMemory(shm_name, shm_key, kShmBytes)
+                 .IsOk()) {
+          errors++;
+          return;
+        }
+      }
       for (const auto& in : meta["inputs"].AsArray()) {
         std::vector<int64_t> shape;
         int64_t elems = 1;
@@ -89,16 +109,44 @@ int main(int argc, char** argv) {
         ca::InferInput::Create(&input, in["name"].AsString(), shape,
                                in["datatype"].AsString());
         size_t elem_size = 4;
-        buffers.emplace_back(elems * elem_size, 1);
-        input->AppendRaw(buffers.back().data(), buffers.back().size());
+        size_t nbytes = elems * elem_size;
+        if (args.shm) {
+          memset((char*)shm_base + shm_cursor, 1, nbytes);
+          input->SetSharedMemory(shm_name, nbytes, shm_cursor);
+          shm_cursor += nbytes;
+        } else {
+          buffers.emplace_back(nbytes, 1);
+          input->AppendRaw(buffers.back().data(), buffers.back().size());
+        }
         inputs.emplace_back(input);
         input_ptrs.push_back(input);
+      }
+      if (args.shm) {
+        for (const auto& out : meta["outputs"].AsArray()) {
+          std::vector<int64_t> shape;
+          int64_t elems = 1;
+          for (const auto& d : out["shape"].AsArray()) {
+            int64_t v = d.AsInt() > 0 ? d.AsInt()
+                                      : (shape.empty() ? args.batch : 16);
+            elems *= v;
+            shape.push_back(v);
+          }
+          size_t nbytes = elems * 4;
+          if (shm_cursor + nbytes > kShmBytes) break;
+          ca::InferRequestedOutput* o;
+          ca::InferRequestedOutput::Create(&o, out["name"].AsString());
+          o->SetSharedMemory(shm_name, nbytes, shm_cursor);
+          shm_cursor += nbytes;
+          outputs.emplace_back(o);
+          output_ptrs.push_back(o);
+        }
       }
       ca::InferOptions options(args.model);
       while (!stop.load(std::memory_order_relaxed)) {
         auto t0 = Clock::now();
         ca::InferResult* result = nullptr;
-        ca::Error err = client->Infer(&result, options, input_ptrs);
+        ca::Error err = client->Infer(&result, options, input_ptrs,
+                                      output_ptrs);
         auto t1 = Clock::now();
         if (err.IsOk() && result != nullptr &&
             result->RequestStatus().IsOk()) {
@@ -110,6 +158,12 @@ int main(int argc, char** argv) {
           errors++;
         }
         delete result;
+      }
+      if (args.shm) {
+        client->UnregisterSystemSharedMemory(shm_name);
+        ca::UnmapSharedMemory(shm_base, kShmBytes);
+        ca::CloseSharedMemory(shm_fd);
+        ca::UnlinkSharedMemoryRegion(shm_key);
       }
     };
 

@@ -117,3 +117,17 @@ def test_cc_perf_client(cc_binaries, http_fixture_server):
     assert proc.returncode == 0, proc.stdout + proc.stderr
     assert "infer/sec" in proc.stdout
     assert "errors: 0" in proc.stdout
+
+
+def test_cc_perf_client_shm(cc_binaries, http_fixture_server):
+    """C++ load generator in system-shm I/O mode."""
+    host, port, _ = http_fixture_server
+    _, _, perf_bin = cc_binaries
+    proc = subprocess.run(
+        [str(perf_bin), "-u", f"{host}:{port}", "-m", "simple", "--shm",
+         "--concurrency-range", "2:2:1", "--measurement-interval", "0.3",
+         "--max-windows", "1"],
+        capture_output=True, text=True, timeout=60,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "errors: 0" in proc.stdout
