@@ -14,6 +14,8 @@
 #include <cstring>
 #include <sstream>
 
+#include <zlib.h>
+
 #include "client_amd/base64.h"
 
 namespace client_amd {
@@ -161,6 +163,55 @@ void AppendDataJson(
     }
   }
   // FP16/BF16 rejected earlier (must be binary)
+}
+
+}  // namespace
+
+namespace {
+
+// zlib helpers for request/response bodies (reference
+// http_client.cc:145-221 compresses the scatter-list the same way).
+bool ZlibCompress(const std::string& in, std::string* out, bool gzip) {
+  z_stream zs;
+  memset(&zs, 0, sizeof(zs));
+  int window = gzip ? 15 + 16 : 15;
+  if (deflateInit2(&zs, Z_DEFAULT_COMPRESSION, Z_DEFLATED, window, 8,
+                   Z_DEFAULT_STRATEGY) != Z_OK) {
+    return false;
+  }
+  zs.next_in = (Bytef*)in.data();
+  zs.avail_in = (uInt)in.size();
+  out->resize(deflateBound(&zs, in.size()));
+  zs.next_out = (Bytef*)out->data();
+  zs.avail_out = (uInt)out->size();
+  int rc = deflate(&zs, Z_FINISH);
+  deflateEnd(&zs);
+  if (rc != Z_STREAM_END) return false;
+  out->resize(zs.total_out);
+  return true;
+}
+
+bool ZlibDecompress(const std::string& in, std::string* out) {
+  z_stream zs;
+  memset(&zs, 0, sizeof(zs));
+  if (inflateInit2(&zs, 15 + 32) != Z_OK) return false;  // auto gzip/zlib
+  zs.next_in = (Bytef*)in.data();
+  zs.avail_in = (uInt)in.size();
+  out->clear();
+  char buf[1 << 16];
+  int rc;
+  do {
+    zs.next_out = (Bytef*)buf;
+    zs.avail_out = sizeof(buf);
+    rc = inflate(&zs, Z_NO_FLUSH);
+    if (rc != Z_OK && rc != Z_STREAM_END) {
+      inflateEnd(&zs);
+      return false;
+    }
+    out->append(buf, sizeof(buf) - zs.avail_out);
+  } while (rc != Z_STREAM_END && zs.avail_in > 0);
+  inflateEnd(&zs);
+  return rc == Z_STREAM_END;
 }
 
 }  // namespace
@@ -779,7 +830,9 @@ Error InferenceServerHttpClient::Infer(
     InferResult** result, const InferOptions& options,
     const std::vector<InferInput*>& inputs,
     const std::vector<const InferRequestedOutput*>& outputs,
-    const Headers& headers, const Parameters& query_params) {
+    const Headers& headers, const Parameters& query_params,
+    const CompressionType request_compression_algorithm,
+    const CompressionType response_compression_algorithm) {
   RequestTimers timer;
   timer.CaptureTimestamp(RequestTimers::Kind::REQUEST_START);
 
@@ -793,6 +846,21 @@ Error InferenceServerHttpClient::Infer(
     hdrs["Inference-Header-Content-Length"] = std::to_string(header_length);
   }
   hdrs["Content-Type"] = "application/octet-stream";
+  std::string compressed;
+  if (request_compression_algorithm != CompressionType::NONE) {
+    bool gzip = request_compression_algorithm == CompressionType::GZIP;
+    std::string raw(body.data(), body.size());
+    if (!ZlibCompress(raw, &compressed, gzip)) {
+      return Error("failed to compress request body");
+    }
+    body.assign(compressed.begin(), compressed.end());
+    hdrs["Content-Encoding"] = gzip ? "gzip" : "deflate";
+  }
+  if (response_compression_algorithm == CompressionType::GZIP) {
+    hdrs["Accept-Encoding"] = "gzip";
+  } else if (response_compression_algorithm == CompressionType::DEFLATE) {
+    hdrs["Accept-Encoding"] = "deflate";
+  }
 
   std::string path = "/v2/models/" + UrlEncode(options.model_name_);
   if (!options.model_version_.empty())
@@ -824,6 +892,14 @@ Error InferenceServerHttpClient::Infer(
   size_t json_size = 0;
   auto it = response_headers.find("inference-header-content-length");
   if (it != response_headers.end()) json_size = (size_t)atoll(it->second.c_str());
+  auto enc = response_headers.find("content-encoding");
+  if (enc != response_headers.end()) {
+    std::string decompressed;
+    if (!ZlibDecompress(response, &decompressed)) {
+      return Error("failed to decompress response body");
+    }
+    response = std::move(decompressed);
+  }
 
   auto body_ptr = std::make_shared<std::string>(std::move(response));
   InferResultHttp::Create(result, body_ptr, json_size, http_code);

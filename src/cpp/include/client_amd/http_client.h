@@ -33,6 +33,8 @@ class InferResultHttp;
 
 class InferenceServerHttpClient : public InferenceServerClient {
  public:
+  enum class CompressionType { NONE, DEFLATE, GZIP };
+
   ~InferenceServerHttpClient() override;
 
   // url is host:port (no scheme), like the reference.
@@ -112,7 +114,11 @@ class InferenceServerHttpClient : public InferenceServerClient {
       InferResult** result, const InferOptions& options,
       const std::vector<InferInput*>& inputs,
       const std::vector<const InferRequestedOutput*>& outputs = {},
-      const Headers& headers = {}, const Parameters& query_params = {});
+      const Headers& headers = {}, const Parameters& query_params = {},
+      const CompressionType request_compression_algorithm =
+          CompressionType::NONE,
+      const CompressionType response_compression_algorithm =
+          CompressionType::NONE);
 
   Error AsyncInfer(
       OnCompleteFn callback, const InferOptions& options,

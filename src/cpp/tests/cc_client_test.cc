@@ -261,6 +261,28 @@ int main(int argc, char** argv) {
   CHECK(!bad.IsOk());
   delete bad_result;
 
+  // ---- request/response compression ----
+  {
+    InferResult* cres = nullptr;
+    CHECK_OK(client->Infer(
+        &cres, options, {input0, input1}, {output0, output1}, {}, {},
+        InferenceServerHttpClient::CompressionType::GZIP,
+        InferenceServerHttpClient::CompressionType::NONE));
+    std::unique_ptr<InferResult> crp(cres);
+    CHECK_OK(cres->RequestStatus());
+    const uint8_t* cb;
+    size_t cn;
+    CHECK_OK(cres->RawData("OUTPUT0", &cb, &cn));
+    CHECK(cn == 64);
+    InferResult* dres = nullptr;
+    CHECK_OK(client->Infer(
+        &dres, options, {input0, input1}, {output0, output1}, {}, {},
+        InferenceServerHttpClient::CompressionType::DEFLATE,
+        InferenceServerHttpClient::CompressionType::NONE));
+    std::unique_ptr<InferResult> drp(dres);
+    CHECK_OK(dres->RequestStatus());
+  }
+
   // ---- base64 roundtrip (cencode parity) ----
   std::string raw64(64, '\0');
   for (int i = 0; i < 64; ++i) raw64[i] = (char)i;

@@ -26,7 +26,7 @@ def _compile(name, main_src, grpc=False):
     if out.exists() and out.stat().st_mtime > newest:
         return out
     cmd = ["g++", "-std=c++17", "-O1", f"-I{CPP}/include", "-Wall",
-           *map(str, srcs), "-o", str(out), "-lpthread", "-lrt"]
+           *map(str, srcs), "-o", str(out), "-lpthread", "-lrt", "-lz"]
     if grpc:
         cmd.append("-l:libnghttp2.so.14")
     subprocess.run(cmd, check=True, capture_output=True, text=True)
