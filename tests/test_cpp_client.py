@@ -13,22 +13,41 @@ CPP = REPO / "src" / "cpp"
 BUILD = CPP / "build_test"
 
 
+_LIB_SRCS = ["common.cc", "json.cc", "shm_utils.cc", "http_client.cc",
+             "h2.cc", "kserve_pb.cc", "grpc_client.cc", "hip_shm.cc"]
+
+
+def _headers_mtime():
+    return max(p.stat().st_mtime
+               for p in (CPP / "include" / "client_amd").glob("*.h"))
+
+
+def _object(src_name):
+    """Compile one library source to a cached .o."""
+    BUILD.mkdir(exist_ok=True)
+    src = CPP / src_name
+    obj = BUILD / (src_name + ".o")
+    if obj.exists() and obj.stat().st_mtime > max(src.stat().st_mtime,
+                                                 _headers_mtime()):
+        return obj
+    subprocess.run(
+        ["g++", "-std=c++17", "-O1", f"-I{CPP}/include", "-Wall", "-c",
+         str(src), "-o", str(obj)],
+        check=True, capture_output=True, text=True)
+    return obj
+
+
 def _compile(name, main_src, grpc=False):
     BUILD.mkdir(exist_ok=True)
     out = BUILD / name
-    base = ["common.cc", "json.cc", "shm_utils.cc"]
-    base += (["h2.cc", "kserve_pb.cc", "grpc_client.cc"] if grpc
-             else ["http_client.cc"])
-    base += ["hip_shm.cc"]
-    srcs = [CPP / s for s in base] + [main_src]
-    newest = max(p.stat().st_mtime for p in srcs + [CPP / "include" /
-                                                    "client_amd" / "common.h"])
+    objs = [_object(s) for s in _LIB_SRCS]
+    newest = max([main_src.stat().st_mtime, _headers_mtime()]
+                 + [o.stat().st_mtime for o in objs])
     if out.exists() and out.stat().st_mtime > newest:
         return out
     cmd = ["g++", "-std=c++17", "-O1", f"-I{CPP}/include", "-Wall",
-           *map(str, srcs), "-o", str(out), "-lpthread", "-lrt", "-lz"]
-    if grpc:
-        cmd.append("-l:libnghttp2.so.14")
+           str(main_src), *map(str, objs), "-o", str(out), "-lpthread",
+           "-lrt", "-lz", "-l:libnghttp2.so.14"]
     subprocess.run(cmd, check=True, capture_output=True, text=True)
     return out
 
@@ -131,3 +150,83 @@ def test_cc_perf_client_shm(cc_binaries, http_fixture_server):
     )
     assert proc.returncode == 0, proc.stdout + proc.stderr
     assert "errors: 0" in proc.stdout
+
+
+HTTP_CC_EXAMPLES = [
+    "simple_http_async_infer_client.cc",
+    "simple_http_string_infer_client.cc",
+    "simple_http_shm_client.cc",
+    "reuse_infer_objects_client.cc",
+]
+
+
+@pytest.mark.parametrize("src", HTTP_CC_EXAMPLES)
+def test_cc_http_examples(src, http_fixture_server):
+    host, port, _ = http_fixture_server
+    binary = _compile(src[:-3], CPP / "examples" / src)
+    proc = subprocess.run(
+        [str(binary), "-u", f"{host}:{port}"], capture_output=True, text=True,
+        timeout=60,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "PASS" in proc.stdout
+
+
+def test_cc_grpc_sequence_stream_example(grpc_fixture_server):
+    host, port, _ = grpc_fixture_server
+    binary = _compile("simple_grpc_sequence_stream_client",
+                      CPP / "examples" / "simple_grpc_sequence_stream_client.cc")
+    proc = subprocess.run(
+        [str(binary), "-u", f"{host}:{port}"], capture_output=True, text=True,
+        timeout=60,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "PASS" in proc.stdout
+
+
+@pytest.mark.gpu
+def test_cc_hipshm_example_gpu():
+    """Compile the C++ HIP-IPC example with hipcc and run it against an
+    out-of-process server — the C++ analog of the cudashm example
+    round trip (GPU)."""
+    import os
+    import time
+
+    from client_amd.ops import gpu_available
+
+    if not gpu_available():
+        pytest.skip("no HIP device")
+    BUILD.mkdir(exist_ok=True)
+    binary = BUILD / "simple_http_hipshm_client"
+    srcs = [CPP / s for s in ("common.cc", "json.cc", "shm_utils.cc",
+                              "http_client.cc", "hip_shm.cc")]
+    srcs.append(CPP / "examples" / "simple_http_hipshm_client.cc")
+    subprocess.run(
+        ["/opt/rocm/bin/hipcc", "--offload-arch=gfx950", "-std=c++17", "-O1",
+         f"-I{CPP}/include", "-DTRITON_ENABLE_HIP", *map(str, srcs),
+         "-o", str(binary), "-lpthread", "-lrt", "-lz"],
+        check=True, capture_output=True, text=True)
+    server = subprocess.Popen(
+        [sys.executable, "-m", "client_amd.server", "--http-port", "18511",
+         "--models", "simple"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        cwd=REPO,
+    )
+    try:
+        deadline = time.time() + 60
+        ready = False
+        while time.time() < deadline:
+            line = server.stdout.readline()
+            if line.startswith("HTTP_READY"):
+                ready = True
+                break
+        assert ready, "server not ready"
+        proc = subprocess.run(
+            [str(binary), "-u", "127.0.0.1:18511"], capture_output=True,
+            text=True, timeout=120,
+        )
+        assert proc.returncode == 0, proc.stdout + proc.stderr
+        assert "PASS" in proc.stdout
+    finally:
+        server.terminate()
+        server.wait(timeout=10)
