@@ -93,14 +93,18 @@ def main():
         return 1
 
     torch.cuda.set_device(local_rank)
+    from client_amd.parallel import (
+        RegionBroadcaster,
+        aggregate_max,
+        init_distributed,
+    )
+
+    init_distributed("nccl" if distributed else None)
     if distributed:
         import torch.distributed as dist
 
-        dist.init_process_group("nccl")
-
     import client_amd.grpc as grpcclient
     import client_amd.utils.hip_shared_memory as hipshm
-    import torch.utils.dlpack
 
     server_args = [] if args.no_dynamic_batching else [
         "--dynamic-batching", "--preferred-batch-size", "32",
@@ -148,13 +152,12 @@ def main():
         for slot in slots:
             hipshm.set_shared_memory_region_cast(slot["in"], host_x, "BF16")
 
-        # torch views of slot-0 input region for the RCCL broadcast path
-        bcast_tensor = None
-        if distributed:
-            smt = hipshm.as_shared_memory_tensor(
-                slots[0]["in"], "BF16", list(IN_SHAPE)
-            )
-            bcast_tensor = torch.from_dlpack(smt)
+        # zero-copy view of the slot-0 input region for the RCCL
+        # broadcast fan-out (client_amd.parallel)
+        broadcaster = (
+            RegionBroadcaster(slots[0]["in"], IN_SHAPE, "BF16", src=0)
+            if distributed else None
+        )
 
         latencies = []
         lat_lock = threading.Lock()
@@ -167,10 +170,8 @@ def main():
             )
             if distributed:
                 # fan the staged input out to every replica over xGMI
-                import torch.distributed as dist
-
                 torch.cuda.synchronize()
-                dist.broadcast(bcast_tensor, src=0)
+                broadcaster.broadcast()
                 torch.cuda.synchronize()
             # dispatcher model: the main thread issues every request,
             # bounded by a concurrency semaphore; gRPC completion
@@ -231,10 +232,7 @@ def main():
         elapsed = t_end - t_start
 
         # max elapsed over ranks -> whole-job throughput
-        if distributed:
-            e = torch.tensor([elapsed], device="cuda")
-            dist.all_reduce(e, op=dist.ReduceOp.MAX)
-            elapsed = float(e.item())
+        elapsed = aggregate_max(elapsed, device="cuda")
 
         total_requests = args.steps * args.reqs_per_step * world
         total_inferences = total_requests * BATCH
