@@ -5,7 +5,6 @@ Single-slot plugin registry with a hook invoked before every network op
 """
 
 from ._plugin import InferenceServerClientPlugin
-from ._request import Request
 
 
 class InferenceServerClientBase:

@@ -8,13 +8,11 @@ bi-di streaming with decoupled final-response semantics, cancellation,
 keepalive, compression, and the full management surface.
 """
 
-import struct
-
 import grpc
 
 from .._client import InferenceServerClientBase
 from .._request import Request
-from ..utils import InferenceServerException, raise_error
+from ..utils import raise_error
 from ._infer_result import InferResult
 from ._infer_stream import _InferStream, _RequestIterator
 from ._proto import RPCS, SERVICE_NAME, service_pb2

@@ -4,7 +4,6 @@
 import grpc
 
 from ..utils import InferenceServerException, _reserved_params, raise_error
-from ._proto import service_pb2
 
 
 def get_error_grpc(rpc_error):

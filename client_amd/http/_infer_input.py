@@ -11,8 +11,6 @@ from ..utils import (
     raise_error,
     serialize_bf16_tensor,
     serialize_byte_tensor,
-    serialized_byte_size,
-    triton_to_np_dtype,
 )
 
 

@@ -17,7 +17,6 @@ from ..utils import (
     deserialize_bytes_tensor,
     raise_error,
     triton_to_np_dtype,
-    TRITON_DTYPE_SIZES,
 )
 
 

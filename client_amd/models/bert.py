@@ -5,8 +5,6 @@ batching and HIP-shm I/O. Standard architecture (hidden 1024, 24
 layers, 16 heads), random-init weights, synthetic token inputs.
 """
 
-import math
-
 import torch
 import torch.nn as nn
 

@@ -9,7 +9,6 @@ inputs synthetic, which is exactly what the decode-path benchmark
 needs — the compute per token is the real thing.
 """
 
-import math
 from dataclasses import dataclass
 
 import torch

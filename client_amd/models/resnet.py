@@ -6,7 +6,6 @@ this image, so the standard architecture is defined here directly
 (random-init weights; the benchmark uses synthetic inputs).
 """
 
-import torch
 import torch.nn as nn
 
 

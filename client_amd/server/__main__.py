@@ -9,7 +9,6 @@ by a *different* process, so the server always runs out-of-process.
 
 import argparse
 import signal
-import sys
 import threading
 
 

@@ -18,7 +18,6 @@ from ..utils import (
     serialize_bf16_tensor,
     triton_to_np_dtype,
     np_to_triton_dtype,
-    TRITON_DTYPE_SIZES,
 )
 
 

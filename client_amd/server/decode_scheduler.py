@@ -92,6 +92,10 @@ class DecodeScheduler:
             ids = torch.as_tensor(input_ids, dtype=torch.int64,
                                   device=self.device)[None]
             s = ids.shape[1]
+            if s >= self.model.cfg.max_seq:
+                # prompt longer than the KV cache: reject the stream
+                out.put(self.END)
+                continue
             # per-slot cache row views: prefill writes rows [idx:idx+1]
             row_cache = [(ck[idx : idx + 1], cv[idx : idx + 1])
                          for ck, cv in self.kv_cache]

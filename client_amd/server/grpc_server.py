@@ -12,7 +12,6 @@ from concurrent import futures
 import grpc
 
 from ..grpc._proto import RPCS, SERVICE_NAME, service_pb2
-from ..utils import np_to_triton_dtype
 from .core import InferenceCore, InferenceError
 
 _DT_ENUM = {

@@ -18,7 +18,7 @@ import time
 
 import numpy as np
 
-from ..utils import triton_to_np_dtype, np_to_triton_dtype
+from ..utils import triton_to_np_dtype
 
 
 class Model:

@@ -6,7 +6,6 @@ import ctypes
 
 from ._dlpack import (
     DLDevice,
-    DLDeviceType,
     DLManagedTensor,
     DLManagedTensorDeleter,
     DLTensor,

@@ -18,7 +18,7 @@ import base64
 
 import numpy as np
 
-from .. import serialize_byte_tensor, triton_to_np_dtype
+from .. import serialize_byte_tensor
 from .._dlpack import (
     DLDeviceType,
     get_byte_size,

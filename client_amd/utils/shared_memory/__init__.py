@@ -9,11 +9,10 @@ repeated create/destroy of the same key safe (reference :36, :78-112,
 
 import mmap
 import os
-import struct
 
 import numpy as np
 
-from .. import serialize_byte_tensor, serialize_bf16_tensor
+from .. import serialize_byte_tensor
 
 _key_mapping = {}
 
@@ -99,7 +98,7 @@ def set_shared_memory_region(shm_handle, input_values, offset=0):
 
 def get_contents_as_numpy(shm_handle, datatype, shape, offset=0):
     """View region contents as a numpy tensor (reference :166-210)."""
-    from .. import deserialize_bytes_tensor, deserialize_bf16_tensor
+    from .. import deserialize_bytes_tensor
 
     mem = shm_handle._mpsm_handle
     if datatype == np.object_ or datatype == bytes:
