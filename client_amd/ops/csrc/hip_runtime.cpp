@@ -193,6 +193,39 @@ extern "C" __global__ void cast_fp32_bf16_v2_kernel(
   }
 }
 
+// A/B variant 3: 16 elems/lane with nontemporal loads/stores (the
+// packed stream is written once and read by a different consumer —
+// bypassing L2 may help at HBM-bound sizes).
+extern "C" __global__ void cast_fp32_bf16_v3_kernel(
+    const uint32_t* __restrict__ src, uint16_t* __restrict__ dst, long n) {
+  long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 16;
+  long stride = (long)gridDim.x * blockDim.x * 16;
+  for (long i = i0; i + 16 <= n; i += stride) {
+    uint4 a[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      a[j] = __builtin_nontemporal_load(
+          reinterpret_cast<const uint4*>(src + i + 4 * j));
+    uint16_t out[16];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      out[4 * j + 0] = (uint16_t)(a[j].x >> 16);
+      out[4 * j + 1] = (uint16_t)(a[j].y >> 16);
+      out[4 * j + 2] = (uint16_t)(a[j].z >> 16);
+      out[4 * j + 3] = (uint16_t)(a[j].w >> 16);
+    }
+    __builtin_nontemporal_store(*reinterpret_cast<uint4*>(out),
+                                reinterpret_cast<uint4*>(dst + i));
+    __builtin_nontemporal_store(*reinterpret_cast<uint4*>(out + 8),
+                                reinterpret_cast<uint4*>(dst + i + 8));
+  }
+  long tail_start = (n / 16) * 16;
+  long ti = tail_start + (blockIdx.x * blockDim.x + threadIdx.x);
+  if (ti < n && (blockIdx.x * blockDim.x + threadIdx.x) < 16) {
+    dst[ti] = (uint16_t)(src[ti] >> 16);
+  }
+}
+
 // Fused RMSNorm for bf16 rows: out = x * rsqrt(mean(x^2)+eps) * w,
 // computed in fp32 (byte-compatible with the torch reference sequence
 // float() -> pow/mean/rsqrt -> mul -> to(bf16), which launches ~7
@@ -569,6 +602,19 @@ static void cast_fp32_bf16_v2(uintptr_t src, uintptr_t dst, long n,
   }
 }
 
+static void cast_fp32_bf16_v3(uintptr_t src, uintptr_t dst, long n,
+                              int device, bool sync, int stream_idx) {
+  hipStream_t s = get_stream(device, stream_idx);
+  hipLaunchKernelGGL(cast_fp32_bf16_v3_kernel, dim3(grid_for((n + 15) / 16)),
+                     dim3(256), 0, s, reinterpret_cast<const uint32_t*>(src),
+                     reinterpret_cast<uint16_t*>(dst), n);
+  HIP_CHECK(hipGetLastError());
+  if (sync) {
+    py::gil_scoped_release release;
+    HIP_CHECK(hipStreamSynchronize(s));
+  }
+}
+
 static void cast_bf16_fp32(uintptr_t src, uintptr_t dst, long n, int device,
                            bool sync, int stream_idx) {
   hipStream_t s = get_stream(device, stream_idx);
@@ -749,6 +795,9 @@ PYBIND11_MODULE(_hip_c, m) {
         py::arg("n"), py::arg("device") = 0, py::arg("sync") = true,
         py::arg("stream_idx") = 0);
   m.def("cast_fp32_bf16_v2", &cast_fp32_bf16_v2, py::arg("src"),
+        py::arg("dst"), py::arg("n"), py::arg("device") = 0,
+        py::arg("sync") = true, py::arg("stream_idx") = 0);
+  m.def("cast_fp32_bf16_v3", &cast_fp32_bf16_v3, py::arg("src"),
         py::arg("dst"), py::arg("n"), py::arg("device") = 0,
         py::arg("sync") = true, py::arg("stream_idx") = 0);
   m.def("cast_bf16_fp32", &cast_bf16_fp32, py::arg("src"), py::arg("dst"),
