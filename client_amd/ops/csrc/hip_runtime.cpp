@@ -198,14 +198,17 @@ extern "C" __global__ void cast_fp32_bf16_v2_kernel(
 // bypassing L2 may help at HBM-bound sizes).
 extern "C" __global__ void cast_fp32_bf16_v3_kernel(
     const uint32_t* __restrict__ src, uint16_t* __restrict__ dst, long n) {
+  // clang's nontemporal builtins need ext_vector types, not the HIP
+  // vector structs
+  typedef unsigned int u32x4 __attribute__((ext_vector_type(4)));
   long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 16;
   long stride = (long)gridDim.x * blockDim.x * 16;
   for (long i = i0; i + 16 <= n; i += stride) {
-    uint4 a[4];
+    u32x4 a[4];
 #pragma unroll
     for (int j = 0; j < 4; ++j)
       a[j] = __builtin_nontemporal_load(
-          reinterpret_cast<const uint4*>(src + i + 4 * j));
+          reinterpret_cast<const u32x4*>(src + i + 4 * j));
     uint16_t out[16];
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
@@ -214,10 +217,10 @@ extern "C" __global__ void cast_fp32_bf16_v3_kernel(
       out[4 * j + 2] = (uint16_t)(a[j].z >> 16);
       out[4 * j + 3] = (uint16_t)(a[j].w >> 16);
     }
-    __builtin_nontemporal_store(*reinterpret_cast<uint4*>(out),
-                                reinterpret_cast<uint4*>(dst + i));
-    __builtin_nontemporal_store(*reinterpret_cast<uint4*>(out + 8),
-                                reinterpret_cast<uint4*>(dst + i + 8));
+    __builtin_nontemporal_store(*reinterpret_cast<u32x4*>(out),
+                                reinterpret_cast<u32x4*>(dst + i));
+    __builtin_nontemporal_store(*reinterpret_cast<u32x4*>(out + 8),
+                                reinterpret_cast<u32x4*>(dst + i + 8));
   }
   long tail_start = (n / 16) * 16;
   long ti = tail_start + (blockIdx.x * blockDim.x + threadIdx.x);
