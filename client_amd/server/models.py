@@ -355,6 +355,14 @@ class TorchModel(Model):
         if dtype is not None:
             self.module = self.module.to(dtype)
         self.dtype = dtype
+        import os
+
+        if os.environ.get("CLIENT_AMD_CHANNELS_LAST") == "1":
+            # MIOpen's NHWC path: often the fast conv layout on CDNA
+            self.module = self.module.to(memory_format=torch.channels_last)
+            self._channels_last = True
+        else:
+            self._channels_last = False
         self.module.eval()
         # hipGraph capture-and-replay per input signature: the serving
         # loop is launch-bound in eager mode; replay submits the whole
@@ -439,6 +447,12 @@ class TorchModel(Model):
                 if not isinstance(result, (tuple, list)):
                     result = (result,)
                 return list(result)
+        if self._channels_last:
+            device_tensors = [
+                t.to(memory_format=self._torch.channels_last)
+                if t.dim() == 4 else t
+                for t in device_tensors
+            ]
         key = tuple((tuple(t.shape), t.dtype) for t in device_tensors)
         with self._graph_lock:
             entry = self._graphs.get(key)
