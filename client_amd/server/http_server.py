@@ -13,6 +13,7 @@ import gzip
 import json
 import re
 import zlib
+from concurrent.futures import ThreadPoolExecutor
 
 from .core import InferenceCore, InferenceError
 
@@ -38,6 +39,10 @@ class _HttpProtocol(asyncio.Protocol):
         self._core = server.core
         self._buf = bytearray()
         self._transport = None
+        # per-connection FIFO: requests are handled on the executor (GPU
+        # models must not block the event loop) but responses must go
+        # out in request order on a keep-alive connection
+        self._chain = None
 
     def connection_made(self, transport):
         self._transport = transport
@@ -75,7 +80,26 @@ class _HttpProtocol(asyncio.Protocol):
                 return
             body = bytes(self._buf[end + 4 : total])
             del self._buf[:total]
-            self._handle(method.decode(), path.decode(), headers, body)
+            self._schedule(method.decode(), path.decode(), headers, body)
+
+    def _schedule(self, method, path, headers, body):
+        loop = asyncio.get_event_loop()
+
+        async def run(prev):
+            reply = await loop.run_in_executor(
+                self._server.executor, self._process, method, path, headers,
+                body,
+            )
+            if prev is not None:
+                await prev
+            if self._transport is not None and not self._transport.is_closing():
+                self._transport.write(reply)
+
+        self._chain = loop.create_task(run(self._chain))
+
+    def _process(self, method, path, headers, body):
+        """Runs on the executor; returns the raw response bytes."""
+        return self._handle(method, path, headers, body)
 
     def _handle(self, method, path, headers, body):
         enc = headers.get("content-encoding")
@@ -106,16 +130,17 @@ class _HttpProtocol(asyncio.Protocol):
             out.append(f"{k}: {v}\r\n".encode())
         out.append(b"\r\n")
         out.append(resp_body)
-        self._transport.write(b"".join(out))
+        return b"".join(out)
 
 
 class HttpServer:
     """asyncio KServe-v2 HTTP server around an InferenceCore."""
 
-    def __init__(self, core=None, host="127.0.0.1", port=8000):
+    def __init__(self, core=None, host="127.0.0.1", port=8000, workers=8):
         self.core = core if core is not None else InferenceCore()
         self.host = host
         self.port = port
+        self.executor = ThreadPoolExecutor(max_workers=workers)
         self._server = None
 
     # ---- routing ----
