@@ -48,16 +48,23 @@ class _PooledConnection:
     """One persistent HTTP/1.1 connection with its own lock-free usage
     (pool hands a connection to exactly one request at a time)."""
 
-    def __init__(self, host, port, timeout):
+    def __init__(self, host, port, timeout, ssl_context=None):
         self._host = host
         self._port = port
         self._timeout = timeout
+        self._ssl_context = ssl_context
         self._conn = None
 
     def _connect(self):
-        self._conn = http.client.HTTPConnection(
-            self._host, self._port, timeout=self._timeout
-        )
+        if self._ssl_context is not None:
+            self._conn = http.client.HTTPSConnection(
+                self._host, self._port, timeout=self._timeout,
+                context=self._ssl_context,
+            )
+        else:
+            self._conn = http.client.HTTPConnection(
+                self._host, self._port, timeout=self._timeout
+            )
         self._conn.connect()
         # Disable Nagle: small JSON requests must not wait for ACKs
         # (the reference sets TCP_NODELAY via curl, http_client.cc:2172-2174).
@@ -128,17 +135,30 @@ class InferenceServerClient(InferenceServerClientBase):
             raise_error("url should not include the scheme")
         parsed = urllib.parse.urlparse("http://" + url)
         self._host = parsed.hostname
-        self._port = parsed.port if parsed.port is not None else 80
+        self._port = parsed.port if parsed.port is not None else (
+            443 if ssl else 80)
         self._base_path = parsed.path.rstrip("/")
         self._verbose = verbose
         self._concurrency = concurrency
         self._timeout = network_timeout
+        ssl_context = None
+        if ssl:
+            import ssl as ssl_mod
+
+            if ssl_context_factory is not None:
+                ssl_context = ssl_context_factory()
+            else:
+                ssl_context = ssl_mod.create_default_context()
+                if insecure:
+                    ssl_context.check_hostname = False
+                    ssl_context.verify_mode = ssl_mod.CERT_NONE
         self._pool = []
         self._pool_lock = threading.Lock()
         self._pool_sem = threading.Semaphore(concurrency)
         for _ in range(concurrency):
             self._pool.append(
-                _PooledConnection(self._host, self._port, network_timeout)
+                _PooledConnection(self._host, self._port, network_timeout,
+                                  ssl_context)
             )
         self._executor = ThreadPoolExecutor(max_workers=concurrency)
         self._closed = False
