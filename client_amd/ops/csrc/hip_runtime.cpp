@@ -105,6 +105,33 @@ extern "C" __global__ void cast_bf16_fp32_kernel(const uint16_t* __restrict__ sr
   }
 }
 
+// 16-elem/lane unpack variant (mirrors the pack-kernel A/B win)
+extern "C" __global__ void cast_bf16_fp32_v2_kernel(
+    const uint16_t* __restrict__ src, uint32_t* __restrict__ dst, long n) {
+  long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 16;
+  long stride = (long)gridDim.x * blockDim.x * 16;
+  for (long i = i0; i + 16 <= n; i += stride) {
+    uint4 a[2];
+    a[0] = *reinterpret_cast<const uint4*>(src + i);
+    a[1] = *reinterpret_cast<const uint4*>(src + i + 8);
+    const uint16_t* e = reinterpret_cast<const uint16_t*>(a);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      uint4 o;
+      o.x = (uint32_t)e[4 * j + 0] << 16;
+      o.y = (uint32_t)e[4 * j + 1] << 16;
+      o.z = (uint32_t)e[4 * j + 2] << 16;
+      o.w = (uint32_t)e[4 * j + 3] << 16;
+      *reinterpret_cast<uint4*>(dst + i + 4 * j) = o;
+    }
+  }
+  long tail_start = (n / 16) * 16;
+  long ti = tail_start + (blockIdx.x * blockDim.x + threadIdx.x);
+  if (ti < n && (blockIdx.x * blockDim.x + threadIdx.x) < 16) {
+    dst[ti] = (uint32_t)src[ti] << 16;
+  }
+}
+
 // fp32 -> fp8 e4m3 (OCP fn, the CDNA4-native format — NOT MI300X fnuz;
 // cdna_hip_programming.md §4). RNE via the __hip_fp8_e4m3 HW convert.
 extern "C" __global__ void cast_fp32_fp8e4m3_kernel(const float* __restrict__ src,
@@ -621,14 +648,13 @@ static void cast_fp32_bf16_v3(uintptr_t src, uintptr_t dst, long n,
 static void cast_bf16_fp32(uintptr_t src, uintptr_t dst, long n, int device,
                            bool sync, int stream_idx) {
   hipStream_t s = get_stream(device, stream_idx);
-  int grid = grid_for((n + 7) / 8);
   if ((src | dst) & 15) {
     hipLaunchKernelGGL(cast_bf16_fp32_scalar_kernel, dim3(grid_for(n)),
                        dim3(256), 0, s, reinterpret_cast<const uint16_t*>(src),
                        reinterpret_cast<uint32_t*>(dst), n);
   } else {
-    hipLaunchKernelGGL(cast_bf16_fp32_kernel, dim3(grid), dim3(256), 0, s,
-                       reinterpret_cast<const uint16_t*>(src),
+    hipLaunchKernelGGL(cast_bf16_fp32_v2_kernel, dim3(grid_for((n + 15) / 16)),
+                       dim3(256), 0, s, reinterpret_cast<const uint16_t*>(src),
                        reinterpret_cast<uint32_t*>(dst), n);
   }
   HIP_CHECK(hipGetLastError());
