@@ -209,6 +209,28 @@ int main(int argc, char** argv) {
     CHECK_OK(client->StopStream());
   }
 
+  // ---- large-message flow control (4 MB each way) ----
+  {
+    const size_t kBig = 1 << 20;  // 1M floats = 4 MB
+    std::vector<float> big(kBig);
+    for (size_t i = 0; i < kBig; ++i) big[i] = (float)(i % 997);
+    InferInput* bin;
+    CHECK_OK(InferInput::Create(&bin, "INPUT0", {(int64_t)kBig}, "FP32"));
+    std::unique_ptr<InferInput> bp(bin);
+    CHECK_OK(bin->AppendRaw((uint8_t*)big.data(), kBig * 4));
+    InferOptions bopt("identity_fp32");
+    InferResult* bres = nullptr;
+    CHECK_OK(client->Infer(&bres, bopt, {bin}));
+    std::unique_ptr<InferResult> brp(bres);
+    CHECK_OK(bres->RequestStatus());
+    const uint8_t* bb;
+    size_t bn;
+    CHECK_OK(bres->RawData("OUTPUT0", &bb, &bn));
+    CHECK(bn == kBig * 4);
+    const float* bf = (const float*)bb;
+    for (size_t i = 0; i < kBig; i += 4097) CHECK(bf[i] == big[i]);
+  }
+
   // ---- statistics ----
   std::vector<kserve::ModelStatisticsPb> stats;
   CHECK_OK(client->ModelInferenceStatistics(&stats, "simple"));
