@@ -3,12 +3,67 @@
 #include <condition_variable>
 #include <cstring>
 #include <deque>
+#include <map>
 
 namespace client_amd {
 
 namespace {
 
 constexpr const char* kService = "/inference.GRPCInferenceService/";
+
+// Global per-url connection cache (reference grpc_client.cc:80-152).
+struct SharedConn {
+  std::shared_ptr<H2Connection> conn;
+  int share_count = 0;
+};
+std::mutex g_conn_mu;
+std::map<std::string, std::vector<SharedConn>> g_conns;
+
+int MaxShareCount() {
+  const char* env = getenv("TRITON_CLIENT_GRPC_CHANNEL_MAX_SHARE_COUNT");
+  if (env != nullptr) {
+    int v = atoi(env);
+    if (v > 0) return v;
+  }
+  return 6;
+}
+
+std::shared_ptr<H2Connection> AcquireConnection(
+    const std::string& host, int port, Error* err) {
+  std::string key = host + ":" + std::to_string(port);
+  std::lock_guard<std::mutex> lock(g_conn_mu);
+  auto& vec = g_conns[key];
+  int max_share = MaxShareCount();
+  for (auto& entry : vec) {
+    if (entry.conn->IsAlive() && entry.share_count < max_share) {
+      entry.share_count++;
+      return entry.conn;
+    }
+  }
+  auto conn = std::make_shared<H2Connection>();
+  *err = conn->Connect(host, port);
+  if (!err->IsOk()) return nullptr;
+  vec.push_back({conn, 1});
+  return conn;
+}
+
+void ReleaseConnection(const std::string& host, int port,
+                       const std::shared_ptr<H2Connection>& conn) {
+  std::string key = host + ":" + std::to_string(port);
+  std::lock_guard<std::mutex> lock(g_conn_mu);
+  auto it = g_conns.find(key);
+  if (it == g_conns.end()) return;
+  auto& vec = it->second;
+  for (size_t i = 0; i < vec.size(); ++i) {
+    if (vec[i].conn == conn) {
+      if (--vec[i].share_count <= 0) {
+        vec[i].conn->Close();
+        vec.erase(vec.begin() + i);
+      }
+      break;
+    }
+  }
+}
 
 // Accumulates DATA bytes and splits gRPC length-prefixed messages.
 struct GrpcMessageBuffer {
@@ -189,14 +244,22 @@ InferenceServerGrpcClient::InferenceServerGrpcClient(
 
 InferenceServerGrpcClient::~InferenceServerGrpcClient() {
   StopStream();
-  if (conn_ != nullptr) conn_->Close();
+  if (conn_ != nullptr) {
+    ReleaseConnection(host_, port_, conn_);
+    conn_.reset();
+  }
 }
 
 Error InferenceServerGrpcClient::EnsureConnected() {
   std::lock_guard<std::mutex> lock(conn_mu_);
   if (conn_ != nullptr && conn_->IsAlive()) return Error::Success;
-  conn_ = std::make_unique<H2Connection>();
-  return conn_->Connect(host_, port_);
+  if (conn_ != nullptr) {
+    ReleaseConnection(host_, port_, conn_);
+    conn_.reset();
+  }
+  Error err = Error::Success;
+  conn_ = AcquireConnection(host_, port_, &err);
+  return err;
 }
 
 Error InferenceServerGrpcClient::AsyncUnaryCall(

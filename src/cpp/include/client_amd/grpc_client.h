@@ -108,7 +108,11 @@ class InferenceServerGrpcClient : public InferenceServerClient {
 
   std::string host_;
   int port_;
-  std::unique_ptr<H2Connection> conn_;
+  // Shared h2 connection from the global per-url cache (reference
+  // grpc_client.cc:80-152: up to TRITON_CLIENT_GRPC_CHANNEL_MAX_SHARE_COUNT
+  // clients share one channel before a new one is created; HTTP/2
+  // stream multiplexing makes sharing natural).
+  std::shared_ptr<H2Connection> conn_;
   std::mutex conn_mu_;
 
   // active bidi stream state
