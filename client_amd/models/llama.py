@@ -122,9 +122,21 @@ class LlamaBlock(nn.Module):
         # ROCm for decode shapes (11.8 vs 14.5 ms ITL @8 streams)
         k_all = k_all.repeat_interleave(rep, dim=1)
         v_all = v_all.repeat_interleave(rep, dim=1)
-        attn = F.scaled_dot_product_attention(
-            q, k_all, v_all, is_causal=(s > 1)
-        )
+        if s > 1 and pos > 0:
+            # continuation chunk: causal mask offset by the cached prefix
+            q_pos = torch.arange(pos, pos + s, device=x.device)[:, None]
+            k_pos = torch.arange(pos + s, device=x.device)[None]
+            mask = torch.where(
+                k_pos <= q_pos,
+                torch.zeros((), device=x.device, dtype=q.dtype),
+                torch.full((), float("-inf"), device=x.device, dtype=q.dtype),
+            )[None, None]
+            attn = F.scaled_dot_product_attention(q, k_all, v_all,
+                                                  attn_mask=mask)
+        else:
+            attn = F.scaled_dot_product_attention(
+                q, k_all, v_all, is_causal=(s > 1)
+            )
         attn = attn.transpose(1, 2).reshape(b, s, -1)
         x = x + self.wo(attn)
         h = self.ffn_norm(x)
@@ -166,7 +178,13 @@ class LlamaModel(nn.Module):
 
     def forward_step(self, input_ids, pos, kv_cache):
         """input_ids [b, s] starting at position pos; returns logits of
-        the LAST position [b, vocab]."""
+        the LAST position [b, vocab].
+
+        For pos > 0 with s > 1 (a prefill CONTINUATION chunk) a causal
+        mask offset by ``pos`` is required: torch sdpa's ``is_causal``
+        aligns top-left (assumes q and k start together), which would
+        hide the cached prefix. Block.forward builds the offset mask
+        whenever pos > 0 and s > 1."""
         cos, sin = self._get_rope(input_ids.device)
         x = self.tok(input_ids)
         for block, cache in zip(self.blocks, kv_cache):

@@ -21,21 +21,29 @@ import torch
 
 
 class _Slot:
-    __slots__ = ("active", "pos", "last_token", "remaining", "out_queue")
+    __slots__ = ("state", "pos", "last_token", "remaining", "out_queue",
+                 "prefill_ids", "prefill_pos")
+    FREE, PREFILL, ACTIVE = 0, 1, 2
 
     def __init__(self):
-        self.active = False
+        self.state = _Slot.FREE
         self.pos = 0
         self.last_token = 0
         self.remaining = 0
         self.out_queue = None
+        self.prefill_ids = None
+        self.prefill_pos = 0
+
+    @property
+    def active(self):
+        return self.state == _Slot.ACTIVE
 
 
 class DecodeScheduler:
     END = object()
 
     def __init__(self, model, max_batch=8, device="cuda:0", dtype=None,
-                 use_graph=None, len_bucket=256):
+                 use_graph=None, len_bucket=256, prefill_chunk=512):
         self.model = model
         self.device = device
         self.dtype = dtype if dtype is not None else next(
@@ -51,6 +59,10 @@ class DecodeScheduler:
         self.use_graph = (device.startswith("cuda")
                           if use_graph is None else use_graph)
         self.len_bucket = len_bucket
+        # chunked prefill: a new stream's prompt is consumed in chunks
+        # interleaved with decode steps, so admission never stalls
+        # in-flight streams for more than one chunk's worth of compute
+        self.prefill_chunk = prefill_chunk
         self._tokens_dev = torch.zeros(max_batch, 1, dtype=torch.int64,
                                        device=device)
         self._pos_dev = torch.ones(max_batch, dtype=torch.int64,
@@ -78,9 +90,11 @@ class DecodeScheduler:
     # ---- worker ----
 
     def _admit(self):
-        """Prefill pending requests into free slots (one at a time)."""
+        """Claim free slots for pending requests (prefill happens in
+        chunks from _prefill_step)."""
         while True:
-            free = [i for i, s in enumerate(self.slots) if not s.active]
+            free = [i for i, s in enumerate(self.slots)
+                    if s.state == _Slot.FREE]
             if not free:
                 return
             try:
@@ -91,26 +105,45 @@ class DecodeScheduler:
             slot = self.slots[idx]
             ids = torch.as_tensor(input_ids, dtype=torch.int64,
                                   device=self.device)[None]
-            s = ids.shape[1]
-            if s >= self.model.cfg.max_seq:
+            if ids.shape[1] >= self.model.cfg.max_seq:
                 # prompt longer than the KV cache: reject the stream
                 out.put(self.END)
                 continue
-            # per-slot cache row views: prefill writes rows [idx:idx+1]
+            slot.state = _Slot.PREFILL
+            slot.prefill_ids = ids
+            slot.prefill_pos = 0
+            slot.remaining = max_new
+            slot.out_queue = out
+
+    def _prefill_step(self):
+        """Advance every PREFILL slot by one chunk (bounds the decode
+        stall per iteration to one chunk of compute)."""
+        for idx, slot in enumerate(self.slots):
+            if slot.state != _Slot.PREFILL:
+                continue
+            ids = slot.prefill_ids
+            total = ids.shape[1]
+            start = slot.prefill_pos
+            end = min(start + self.prefill_chunk, total)
+            chunk = ids[:, start:end]
             row_cache = [(ck[idx : idx + 1], cv[idx : idx + 1])
                          for ck, cv in self.kv_cache]
             with torch.inference_mode():
-                logits = self.model.forward_step(ids, 0, row_cache)
-                first = int(logits.argmax(-1)[0])
-            out.put(first)
-            slot.active = True
-            slot.pos = s  # position the NEXT token will be written at
+                logits = self.model.forward_step(chunk, start, row_cache)
+            slot.prefill_pos = end
+            if end < total:
+                continue
+            first = int(logits.argmax(-1)[0])
+            slot.out_queue.put(first)
+            slot.prefill_ids = None
+            slot.pos = total  # position the NEXT token is written at
             slot.last_token = first
-            slot.remaining = max_new - 1
-            slot.out_queue = out
+            slot.remaining -= 1
             if slot.remaining <= 0:
-                slot.active = False
-                out.put(self.END)
+                slot.state = _Slot.FREE
+                slot.out_queue.put(self.END)
+            else:
+                slot.state = _Slot.ACTIVE
 
     def _bucket(self, max_len):
         b = ((max_len + self.len_bucket - 1) // self.len_bucket
@@ -180,7 +213,7 @@ class DecodeScheduler:
             slot.pos += 1
             slot.remaining -= 1
             if slot.remaining <= 0 or slot.pos >= self.model.cfg.max_seq - 1:
-                slot.active = False
+                slot.state = _Slot.FREE
                 slot.out_queue.put(self.END)
         return True
 
@@ -188,9 +221,10 @@ class DecodeScheduler:
         while True:
             with self._cv:
                 while (self._alive and self._pending.empty()
-                       and not any(s.active for s in self.slots)):
+                       and all(s.state == _Slot.FREE for s in self.slots)):
                     self._cv.wait()
                 if not self._alive:
                     return
             self._admit()
+            self._prefill_step()
             self._decode_step()

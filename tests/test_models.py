@@ -279,3 +279,34 @@ def test_decode_batch_padded_maxlen_equivalent():
         exact = m.forward_decode_batch(tokens, pos, kv1)
         padded = m.forward_decode_batch(tokens, pos, kv2, max_len=64)
     torch.testing.assert_close(exact, padded, rtol=1e-5, atol=1e-5)
+
+
+def test_chunked_prefill_matches_full():
+    """Chunked prefill (continuation chunks with offset causal mask)
+    must produce exactly the tokens full-prompt prefill produces."""
+    from client_amd.models.llama import LlamaModel, llama_tiny_config
+    from client_amd.server.decode_scheduler import DecodeScheduler
+
+    torch.manual_seed(9)
+    cfg = llama_tiny_config()
+    m = LlamaModel(cfg).eval()
+    prompts = [torch.randint(0, cfg.vocab_size, (1, n)) for n in (23, 7, 40)]
+    expected = [[int(t[0]) for t in m.generate(p, 6)] for p in prompts]
+
+    # tiny chunk forces multiple continuation chunks per prompt
+    sched = DecodeScheduler(m, max_batch=4, device="cpu",
+                            dtype=torch.float32, prefill_chunk=8)
+    try:
+        queues = [sched.submit(p[0].numpy(), 6) for p in prompts]
+        got = []
+        for q in queues:
+            toks = []
+            while True:
+                t = q.get(timeout=60)
+                if t is sched.END:
+                    break
+                toks.append(t)
+            got.append(toks)
+        assert got == expected
+    finally:
+        sched.shutdown()
