@@ -14,6 +14,7 @@ from pathlib import Path
 
 OPS_DIR = Path(__file__).resolve().parent
 SRC = OPS_DIR / "csrc" / "hip_runtime.cpp"
+KERNELS = OPS_DIR / "csrc" / "kernels.hip"
 OUT = OPS_DIR / "_hip_c.so"
 
 GFX_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950").split(";")[0]
@@ -28,7 +29,8 @@ def _pybind11_includes():
 def needs_build():
     if not OUT.exists():
         return True
-    return SRC.stat().st_mtime > OUT.stat().st_mtime
+    newest = max(SRC.stat().st_mtime, KERNELS.stat().st_mtime)
+    return newest > OUT.stat().st_mtime
 
 
 def build(force=False, verbose=True):
@@ -47,7 +49,9 @@ def build(force=False, verbose=True):
         "-shared",
         "-fvisibility=hidden",
         "-Wno-unused-result",
+        "-x", "hip",
         str(SRC),
+        str(KERNELS),
         "-o",
         str(OUT),
     ] + [f"-I{inc}" for inc in includes]

@@ -44,14 +44,6 @@ def main():
         "cast_fp32_bf16 (4B+2B/elem)",
         lambda: hr.cast_fp32_bf16(a, b, n, 0, True, 0), n * 6,
     )
-    results["cast_fp32_bf16_v2"] = bench(
-        "cast_fp32_bf16_v2 16e/lane",
-        lambda: hr.cast_fp32_bf16_v2(a, b, n, 0, True, 0), n * 6,
-    )
-    results["cast_fp32_bf16_v3"] = bench(
-        "cast_fp32_bf16_v3 NT 16e",
-        lambda: hr.cast_fp32_bf16_v3(a, b, n, 0, True, 0), n * 6,
-    )
     results["cast_bf16_fp32"] = bench(
         "cast_bf16_fp32 (2B+4B/elem)",
         lambda: hr.cast_bf16_fp32(a, b, n, 0, True, 0), n * 6,

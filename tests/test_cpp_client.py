@@ -184,6 +184,62 @@ def test_cc_grpc_sequence_stream_example(grpc_fixture_server):
     assert "PASS" in proc.stdout
 
 
+def _hipcc_example(name, extra_models="simple", with_kernels=False):
+    """Compile a C++ example with hipcc (+kernels.hip when needed) and
+    run it against a spawned server."""
+    import os
+    import time
+
+    BUILD.mkdir(exist_ok=True)
+    binary = BUILD / name
+    srcs = [CPP / s for s in ("common.cc", "json.cc", "shm_utils.cc",
+                              "http_client.cc", "hip_shm.cc")]
+    if with_kernels:
+        srcs.append(REPO / "client_amd" / "ops" / "csrc" / "kernels.hip")
+    srcs.append(CPP / "examples" / f"{name}.cc")
+    subprocess.run(
+        ["/opt/rocm/bin/hipcc", "--offload-arch=gfx950", "-std=c++17", "-O1",
+         f"-I{CPP}/include", "-DTRITON_ENABLE_HIP", "-x", "hip",
+         *map(str, srcs), "-o", str(binary), "-lpthread", "-lrt", "-lz"],
+        check=True, capture_output=True, text=True)
+    server = subprocess.Popen(
+        [sys.executable, "-m", "client_amd.server", "--http-port", "18511",
+         "--models", extra_models],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        cwd=REPO,
+    )
+    try:
+        deadline = time.time() + 60
+        ready = False
+        while time.time() < deadline:
+            line = server.stdout.readline()
+            if line.startswith("HTTP_READY"):
+                ready = True
+                break
+        assert ready, "server not ready"
+        proc = subprocess.run(
+            [str(binary), "-u", "127.0.0.1:18511"], capture_output=True,
+            text=True, timeout=120,
+        )
+        assert proc.returncode == 0, proc.stdout + proc.stderr
+        assert "PASS" in proc.stdout
+    finally:
+        server.terminate()
+        server.wait(timeout=10)
+
+
+@pytest.mark.gpu
+def test_cc_hipshm_pack_kernel_example_gpu():
+    """C++ client + CDNA4 pack/unpack kernels end-to-end (wire-exact
+    bf16 via HIP-shm) — the C++ analog of the Python device data plane."""
+    from client_amd.ops import gpu_available
+
+    if not gpu_available():
+        pytest.skip("no HIP device")
+    _hipcc_example("simple_http_hipshm_pack_client",
+                   extra_models="identity_bf16", with_kernels=True)
+
+
 @pytest.mark.gpu
 def test_cc_hipshm_example_gpu():
     """Compile the C++ HIP-IPC example with hipcc and run it against an
