@@ -51,7 +51,6 @@ def build(force=False, verbose=True):
         "-Wno-unused-result",
         "-x", "hip",
         str(SRC),
-        str(KERNELS),
         "-o",
         str(OUT),
     ] + [f"-I{inc}" for inc in includes]

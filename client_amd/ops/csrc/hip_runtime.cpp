@@ -43,24 +43,12 @@ namespace py = pybind11;
   } while (0)
 
 // ---------------------------------------------------------------------------
-// CDNA4 kernels live in kernels.hip (shared with the C++ client
-// library); C-ABI launchers declared here.
+// CDNA4 kernels + C-ABI launchers. kernels.hip is the single source of
+// truth, shared with the C++ client library; it is INCLUDED here (one
+// translation unit, one device image) — a separate-TU link was observed
+// to abort at first launch on the GPU box.
 // ---------------------------------------------------------------------------
-
-extern "C" {
-hipError_t ca_cast_fp32_bf16(const void*, void*, long, hipStream_t);
-hipError_t ca_cast_bf16_fp32(const void*, void*, long, hipStream_t);
-hipError_t ca_cast_fp32_fp8e4m3(const void*, void*, long, hipStream_t);
-hipError_t ca_cast_fp8e4m3_fp32(const void*, void*, long, hipStream_t);
-hipError_t ca_gather_pack(const void*, void*, int, const long*, const long*,
-                          hipStream_t);
-hipError_t ca_image_preprocess(const void*, void*, int, int, int, int, int,
-                               int, const float*, const float*, hipStream_t);
-hipError_t ca_rmsnorm_bf16(const void*, const void*, void*, long, int, float,
-                           hipStream_t);
-hipError_t ca_rope_decode_bf16(void*, void*, const void*, const void*,
-                               const void*, int, int, int, int, hipStream_t);
-}
+#include "kernels.hip"
 
 // ---------------------------------------------------------------------------
 // Launch helpers
