@@ -29,6 +29,8 @@ def main(argv=None):
     p.add_argument("--shape", action="append", default=[],
                    help="NAME:d1,d2 override for dynamic input dims")
     p.add_argument("--json", default=None, help="write results to file")
+    p.add_argument("-f", "--csv", default=None,
+                   help="write a perf_analyzer-style CSV report")
     p.add_argument("-v", "--verbose", action="store_true")
     args = p.parse_args(argv)
 
@@ -70,6 +72,15 @@ def main(argv=None):
     if args.json:
         with open(args.json, "w") as f:
             json.dump(results, f, indent=2)
+    if args.csv:
+        with open(args.csv, "w") as f:
+            f.write("Concurrency,Inferences/Second,Client Send,"
+                    "p50 latency,p90 latency,p95 latency,p99 latency\n")
+            for r in results:
+                lat = r["latency_us"]
+                f.write(f"{r['concurrency']},{r['inferences_per_sec']},0,"
+                        f"{lat['p50']},{lat['p90']},{lat['p95']},"
+                        f"{lat['p99']}\n")
 
 
 if __name__ == "__main__":

@@ -40,6 +40,7 @@ def test_perf_cli(grpc_fixture_server, tmp_path, capsys):
         "-m", "simple", "-u", f"{host}:{port}", "-i", "grpc",
         "--concurrency-range", "1:1:1", "--measurement-interval", "0.2",
         "--warmup", "0.1", "--max-windows", "2", "--json", str(out),
+        "-f", str(tmp_path / "report.csv"),
     ])
     captured = capsys.readouterr()
     assert "infer/sec" in captured.out
@@ -47,6 +48,9 @@ def test_perf_cli(grpc_fixture_server, tmp_path, capsys):
 
     data = json.loads(out.read_text())
     assert data[0]["concurrency"] == 1
+    csv_text = (tmp_path / "report.csv").read_text()
+    assert csv_text.startswith("Concurrency,Inferences/Second")
+    assert len(csv_text.strip().splitlines()) == 2
 
 
 def test_genai_perf_llm_stream():
