@@ -12,7 +12,8 @@ import signal
 import threading
 
 
-def build_core(model_names, device="cuda:0", dtype="bf16"):
+def build_core(model_names, device="cuda:0", dtype="bf16",
+               decode_max_batch=8):
     from . import (
         AddSubModel,
         IdentityModel,
@@ -116,7 +117,8 @@ def build_core(model_names, device="cuda:0", dtype="bf16"):
                 module = LlamaModel(cfg)
             core.add_model(
                 GenerateModel(name, module, device=device,
-                              dtype=tdt if use_gpu else None)
+                              dtype=tdt if use_gpu else None,
+                              max_batch=decode_max_batch)
             )
         elif name == "ensemble_image":
             # preprocess (HIP kernel on GPU) -> resnet50 pipeline; the
@@ -183,10 +185,13 @@ def main(argv=None):
                         help="enable dynamic batching on torch models")
     parser.add_argument("--preferred-batch-size", type=int, default=32)
     parser.add_argument("--max-queue-delay-us", type=int, default=500)
+    parser.add_argument("--decode-max-batch", type=int, default=8,
+                        help="continuous-batching slots for generate models")
     args = parser.parse_args(argv)
 
     core = build_core(
-        [m for m in args.models.split(",") if m], args.device, args.dtype
+        [m for m in args.models.split(",") if m], args.device, args.dtype,
+        decode_max_batch=args.decode_max_batch,
     )
     if args.dynamic_batching:
         for model in core.models.values():
