@@ -249,3 +249,13 @@ def test_infer_timeout(client, grpc_fixture_server):
         raise AssertionError("expected timeout")
     except InferenceServerException as e:
         assert "DEADLINE" in str(e.status()).upper()
+
+
+def test_large_tensor_roundtrip(client):
+    """16 MB tensor through grpc (message-size limits raised to INT32_MAX
+    as in the reference)."""
+    x = np.random.rand(4, 1024, 1024).astype(np.float32)
+    inp = grpcclient.InferInput("INPUT0", list(x.shape), "FP32")
+    inp.set_data_from_numpy(x)
+    result = client.infer("identity_fp32", [inp])
+    np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)

@@ -245,3 +245,15 @@ def test_basic_auth_plugin(client, http_fixture_server):
 def test_transfer_encoding_header_rejected(client):
     with pytest.raises(InferenceServerException):
         client.is_server_live(headers={"Transfer-Encoding": "chunked"})
+
+
+def test_large_tensor_roundtrip(client):
+    """16 MB tensor through the pooled HTTP transport (framing +
+    keep-alive reuse at the reference's 16 MiB buffer scale)."""
+    x = np.random.rand(4, 1024, 1024).astype(np.float32)  # 16 MiB
+    inp = httpclient.InferInput("INPUT0", list(x.shape), "FP32")
+    inp.set_data_from_numpy(x)
+    result = client.infer("identity_fp32", [inp])
+    np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
+    # connection reuse after the big transfer
+    assert client.is_server_live()
