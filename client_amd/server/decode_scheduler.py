@@ -166,14 +166,18 @@ class DecodeScheduler:
                     warm = logits.argmax(-1)
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
+        from .models import GRAPH_CAPTURE_LOCK
+
         graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
-        with torch.inference_mode():
-            with torch.cuda.graph(graph):
-                logits = self.model.forward_decode_batch(
-                    self._tokens_dev, self._pos_dev, self.kv_cache,
-                    max_len=bucket,
-                )
-                next_out = logits.argmax(-1)
+        with GRAPH_CAPTURE_LOCK:
+            with torch.inference_mode():
+                with torch.cuda.graph(graph,
+                                      capture_error_mode="thread_local"):
+                    logits = self.model.forward_decode_batch(
+                        self._tokens_dev, self._pos_dev, self.kv_cache,
+                        max_len=bucket,
+                    )
+                    next_out = logits.argmax(-1)
         entry = (graph, next_out)
         self._graphs[bucket] = entry
         return entry

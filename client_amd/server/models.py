@@ -14,11 +14,19 @@ fake-server fixture). This module provides:
                         PyTorch-ROCm (bf16), used by the benchmark
 """
 
+import threading
 import time
 
 import numpy as np
 
 from ..utils import triton_to_np_dtype
+
+# Process-wide: only one hipGraph capture at a time, and captures use
+# thread_local error mode so concurrently-serving threads (other models
+# on the same GPU) are not poisoned by capture state. Found by the
+# mixed-load soak: global-mode capture made every other model fail with
+# hipErrorStreamCaptureUnsupported.
+GRAPH_CAPTURE_LOCK = threading.Lock()
 
 
 class Model:
@@ -418,9 +426,11 @@ class TorchModel(Model):
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
         graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
-        with torch.inference_mode():
-            with torch.cuda.graph(graph):
-                out = self.module(*static_ins)
+        with GRAPH_CAPTURE_LOCK:
+            with torch.inference_mode():
+                with torch.cuda.graph(graph,
+                                      capture_error_mode="thread_local"):
+                    out = self.module(*static_ins)
         if not isinstance(out, (tuple, list)):
             out = (out,)
         return static_ins, graph, list(out)
