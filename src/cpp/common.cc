@@ -74,6 +74,7 @@ Error InferenceServerClient::UpdateInferStat(const RequestTimers& timer) {
       RequestTimers::Kind::RECV_START, RequestTimers::Kind::RECV_END);
   if (recv_ns == UINT64_MAX) recv_ns = 0;
 
+  std::lock_guard<std::mutex> lock(stat_mu_);
   infer_stat_.completed_request_count++;
   infer_stat_.cumulative_total_request_time_ns += request_ns;
   infer_stat_.cumulative_send_time_ns += send_ns;

@@ -245,6 +245,14 @@ Error H2Connection::StartStream(const HeaderList& headers,
 
 Error H2Connection::SendData(int32_t stream_id, const uint8_t* data, size_t n,
                              bool end_stream) {
+  {
+    std::lock_guard<std::mutex> lock(mu_);
+    if (streams_.find(stream_id) == streams_.end()) {
+      // stream already finished/reset: sending DATA would be a
+      // protocol violation that could kill the (shared) connection
+      return Error("stream is closed");
+    }
+  }
   size_t pos = 0;
   while (pos < n || (end_stream && n == 0)) {
     size_t want = n - pos;
@@ -279,6 +287,12 @@ Error H2Connection::SendData(int32_t stream_id, const uint8_t* data, size_t n,
 }
 
 Error H2Connection::FinishStream(int32_t stream_id) {
+  {
+    std::lock_guard<std::mutex> lock(mu_);
+    if (streams_.find(stream_id) == streams_.end()) {
+      return Error::Success;  // already finished server-side
+    }
+  }
   if (!WriteFrame(kFrameData, kFlagEndStream, stream_id, "")) {
     return Error("failed to half-close stream");
   }

@@ -12,6 +12,7 @@
 #include <functional>
 #include <map>
 #include <memory>
+#include <mutex>
 #include <string>
 #include <vector>
 
@@ -309,15 +310,18 @@ class InferenceServerClient {
   virtual ~InferenceServerClient() = default;
 
   Error ClientInferStat(InferStat* infer_stat) const {
+    std::lock_guard<std::mutex> lock(stat_mu_);
     *infer_stat = infer_stat_;
     return Error::Success;
   }
 
  protected:
+  // called from worker threads AND caller threads (async completions)
   Error UpdateInferStat(const RequestTimers& timer);
 
   bool verbose_;
   bool exiting_;
+  mutable std::mutex stat_mu_;
   InferStat infer_stat_;
 };
 
