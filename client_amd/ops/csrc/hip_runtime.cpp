@@ -87,6 +87,16 @@ static hipStream_t get_stream(int device, int idx = 0) {
 // Python bindings
 // ---------------------------------------------------------------------------
 
+static std::pair<size_t, size_t> mem_info(int device) {
+  int prev;
+  HIP_CHECK(hipGetDevice(&prev));
+  HIP_CHECK(hipSetDevice(device));
+  size_t free_b = 0, total_b = 0;
+  HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
+  HIP_CHECK(hipSetDevice(prev));
+  return {free_b, total_b};
+}
+
 static int device_count() {
   int n = 0;
   hipError_t e = hipGetDeviceCount(&n);
@@ -306,6 +316,7 @@ static void image_preprocess(uintptr_t src, uintptr_t dst, int ih, int iw,
 PYBIND11_MODULE(_hip_c, m) {
   m.doc() = "client_amd MI355X HIP runtime + CDNA4 kernels";
   m.def("device_count", &device_count);
+  m.def("mem_info", &mem_info, py::arg("device") = 0);
   m.def("malloc", &hip_malloc, py::arg("device"), py::arg("byte_size"));
   m.def("free", &hip_free, py::arg("ptr"));
   m.def("ipc_get_mem_handle", &ipc_get_mem_handle, py::arg("ptr"));

@@ -406,3 +406,26 @@ def test_fused_rmsnorm_and_rope_numerics(hipshm):
                                rtol=1e-2, atol=1e-2)
     torch.testing.assert_close(k2.float(), ref_rope(k, hk).float(),
                                rtol=1e-2, atol=1e-2)
+
+
+def test_region_lifecycle_no_leak(hipshm):
+    """HIP memory growth check (GPU analog of the reference's
+    memory_growth tests): 200 create/pack/destroy cycles must not leak
+    device memory."""
+    from client_amd.ops import hip_runtime as hr
+
+    n = 1 << 18
+    x = np.random.rand(n).astype(np.float32)
+    # warmup (stream/scratch pools allocate lazily)
+    for _ in range(5):
+        h = hipshm.create_shared_memory_region("leak_w", n * 2, 0)
+        hipshm.set_shared_memory_region_cast(h, x, "BF16")
+        hipshm.destroy_shared_memory_region(h)
+    free_before, _ = hr.mem_info(0)
+    for i in range(200):
+        h = hipshm.create_shared_memory_region(f"leak_{i}", n * 2, 0)
+        hipshm.set_shared_memory_region_cast(h, x, "BF16")
+        hipshm.destroy_shared_memory_region(h)
+    free_after, _ = hr.mem_info(0)
+    leaked = free_before - free_after
+    assert leaked < 64 * 2**20, f"leaked {leaked/2**20:.1f} MiB over 200 cycles"
