@@ -275,10 +275,11 @@ class HttpServer:
         raise InferenceError("bad shared memory action", status=400)
 
     # ---- lifecycle ----
-    async def start(self):
+    async def start(self, ssl_context=None):
         loop = asyncio.get_running_loop()
         self._server = await loop.create_server(
-            lambda: _HttpProtocol(self), self.host, self.port
+            lambda: _HttpProtocol(self), self.host, self.port,
+            ssl=ssl_context,
         )
         if self.port == 0:
             self.port = self._server.sockets[0].getsockname()[1]
@@ -289,7 +290,7 @@ class HttpServer:
             self._server.close()
             await self._server.wait_closed()
 
-    def serve_forever_in_thread(self):
+    def serve_forever_in_thread(self, ssl_context=None):
         """Run the server on a dedicated event-loop thread; returns a
         callable that stops it. Used by test fixtures and bench."""
         import threading
@@ -299,7 +300,7 @@ class HttpServer:
 
         def _run():
             asyncio.set_event_loop(loop)
-            loop.run_until_complete(self.start())
+            loop.run_until_complete(self.start(ssl_context))
             started.set()
             loop.run_forever()
 

@@ -12,8 +12,11 @@
 #include <unistd.h>
 
 #include <cstring>
+#include <mutex>
 #include <sstream>
 
+#include <openssl/err.h>
+#include <openssl/ssl.h>
 #include <zlib.h>
 
 #include "client_amd/base64.h"
@@ -329,6 +332,94 @@ Error InferenceServerHttpClient::Create(
   return Error::Success;
 }
 
+Error InferenceServerHttpClient::Create(
+    std::unique_ptr<InferenceServerHttpClient>* client,
+    const std::string& server_url, bool verbose, bool use_ssl,
+    const HttpSslOptions& ssl_options) {
+  client->reset(new InferenceServerHttpClient(server_url, verbose));
+  (*client)->use_ssl_ = use_ssl;
+  (*client)->ssl_options_ = ssl_options;
+  return Error::Success;
+}
+
+//==============================================================================
+// TLS plumbing (sync path). OpenSSL is initialized lazily per client.
+
+Error InferenceServerHttpClient::SslConnect() {
+  if (ssl_ctx_ == nullptr) {
+    static std::once_flag ssl_init;
+    std::call_once(ssl_init, [] { SSL_library_init(); });
+    SSL_CTX* ctx = SSL_CTX_new(TLS_client_method());
+    if (ctx == nullptr) return Error("SSL_CTX_new failed");
+    if (!ssl_options_.ca_info.empty()) {
+      if (SSL_CTX_load_verify_locations(ctx, ssl_options_.ca_info.c_str(),
+                                        nullptr) != 1) {
+        SSL_CTX_free(ctx);
+        return Error("failed to load CA bundle " + ssl_options_.ca_info);
+      }
+    } else {
+      SSL_CTX_set_default_verify_paths(ctx);
+    }
+    if (!ssl_options_.cert.empty()) {
+      if (SSL_CTX_use_certificate_file(ctx, ssl_options_.cert.c_str(),
+                                       SSL_FILETYPE_PEM) != 1 ||
+          SSL_CTX_use_PrivateKey_file(ctx, ssl_options_.key.c_str(),
+                                      SSL_FILETYPE_PEM) != 1) {
+        SSL_CTX_free(ctx);
+        return Error("failed to load client cert/key");
+      }
+    }
+    SSL_CTX_set_verify(
+        ctx, ssl_options_.verify_peer ? SSL_VERIFY_PEER : SSL_VERIFY_NONE,
+        nullptr);
+    ssl_ctx_ = ctx;
+  }
+  SSL* ssl = SSL_new((SSL_CTX*)ssl_ctx_);
+  if (ssl == nullptr) return Error("SSL_new failed");
+  SSL_set_fd(ssl, sync_fd_);
+  SSL_set_tlsext_host_name(ssl, host_.c_str());
+  if (ssl_options_.verify_host) {
+    SSL_set1_host(ssl, host_.c_str());
+  }
+  if (SSL_connect(ssl) != 1) {
+    unsigned long err = ERR_get_error();
+    char buf[256];
+    ERR_error_string_n(err, buf, sizeof(buf));
+    SSL_free(ssl);
+    return Error(std::string("TLS handshake failed: ") + buf);
+  }
+  ssl_ = ssl;
+  return Error::Success;
+}
+
+void InferenceServerHttpClient::SslClose() {
+  if (ssl_ != nullptr) {
+    SSL_shutdown((SSL*)ssl_);
+    SSL_free((SSL*)ssl_);
+    ssl_ = nullptr;
+  }
+}
+
+bool InferenceServerHttpClient::IoSend(const char* data, size_t n) {
+  if (ssl_ != nullptr) {
+    size_t sent = 0;
+    while (sent < n) {
+      int r = SSL_write((SSL*)ssl_, data + sent, (int)(n - sent));
+      if (r <= 0) return false;
+      sent += (size_t)r;
+    }
+    return true;
+  }
+  return SendAll(sync_fd_, data, n);
+}
+
+long InferenceServerHttpClient::IoRecv(char* buf, size_t n) {
+  if (ssl_ != nullptr) {
+    return SSL_read((SSL*)ssl_, buf, (int)n);
+  }
+  return recv(sync_fd_, buf, n, 0);
+}
+
 InferenceServerHttpClient::InferenceServerHttpClient(
     const std::string& url, bool verbose)
     : InferenceServerClient(verbose) {
@@ -343,6 +434,11 @@ InferenceServerHttpClient::InferenceServerHttpClient(
 }
 
 InferenceServerHttpClient::~InferenceServerHttpClient() {
+  SslClose();
+  if (ssl_ctx_ != nullptr) {
+    SSL_CTX_free((SSL_CTX*)ssl_ctx_);
+    ssl_ctx_ = nullptr;
+  }
   exiting_ = true;
   if (worker_running_.load()) {
     // wake the worker so it can observe exiting_
@@ -365,6 +461,14 @@ Error InferenceServerHttpClient::DoRequest(
       if (sync_fd_ < 0)
         return Error("failed to connect to " + host_ + ":" +
                      std::to_string(port_));
+      if (use_ssl_) {
+        Error ssl_err = SslConnect();
+        if (!ssl_err.IsOk()) {
+          close(sync_fd_);
+          sync_fd_ = -1;
+          return ssl_err;
+        }
+      }
     }
     // SO_RCVTIMEO persists on the reused fd: always (re)set it.
     struct timeval tv;
@@ -379,7 +483,8 @@ Error InferenceServerHttpClient::DoRequest(
     for (const auto& kv : headers) req += kv.first + ": " + kv.second + "\r\n";
     req += "\r\n";
     req += body;
-    if (!SendAll(sync_fd_, req.data(), req.size())) {
+    if (!IoSend(req.data(), req.size())) {
+      SslClose();
       close(sync_fd_);
       sync_fd_ = -1;
       if (attempt == 1) return Error("failed to send request");
@@ -391,10 +496,11 @@ Error InferenceServerHttpClient::DoRequest(
     char chunk[kRecvChunk];
     bool conn_err = false;
     while (header_end == std::string::npos) {
-      ssize_t n = recv(sync_fd_, chunk, sizeof(chunk), 0);
+      long n = IoRecv(chunk, sizeof(chunk));
       if (n <= 0) {
         if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK) &&
             timeout_us > 0) {
+          SslClose();
           close(sync_fd_);
           sync_fd_ = -1;
           return Error("Deadline Exceeded");
@@ -406,6 +512,7 @@ Error InferenceServerHttpClient::DoRequest(
       header_end = buf.find("\r\n\r\n");
     }
     if (conn_err) {
+      SslClose();
       close(sync_fd_);
       sync_fd_ = -1;
       if (attempt == 1) return Error("failed to receive response");
@@ -423,10 +530,11 @@ Error InferenceServerHttpClient::DoRequest(
     auto it = resp_headers.find("content-length");
     if (it != resp_headers.end()) content_length = (size_t)atoll(it->second.c_str());
     while (buf.size() - body_start < content_length) {
-      ssize_t n = recv(sync_fd_, chunk, sizeof(chunk), 0);
+      long n = IoRecv(chunk, sizeof(chunk));
       if (n <= 0) {
         bool timed_out = n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK) &&
                          timeout_us > 0;
+        SslClose();
         close(sync_fd_);
         sync_fd_ = -1;
         return Error(

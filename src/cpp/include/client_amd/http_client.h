@@ -29,6 +29,16 @@ namespace client_amd {
 using Headers = std::map<std::string, std::string>;
 using Parameters = std::map<std::string, std::string>;
 
+// HTTPS options (reference HttpSslOptions http_client.h:45-86; TLS via
+// system OpenSSL — sync path only this round, AsyncInfer stays h2c).
+struct HttpSslOptions {
+  bool verify_peer = true;
+  bool verify_host = true;
+  std::string ca_info;     // CA bundle path ("" = system default)
+  std::string cert;        // client cert path (PEM)
+  std::string key;         // client key path (PEM)
+};
+
 class InferResultHttp;
 
 class InferenceServerHttpClient : public InferenceServerClient {
@@ -41,6 +51,11 @@ class InferenceServerHttpClient : public InferenceServerClient {
   static Error Create(
       std::unique_ptr<InferenceServerHttpClient>* client,
       const std::string& server_url, bool verbose = false);
+  // HTTPS variant (reference Create overload with HttpSslOptions)
+  static Error Create(
+      std::unique_ptr<InferenceServerHttpClient>* client,
+      const std::string& server_url, bool verbose, bool use_ssl,
+      const HttpSslOptions& ssl_options);
 
   // ---- health / metadata ----
   Error IsServerLive(bool* live, const Headers& headers = {});
@@ -173,6 +188,15 @@ class InferenceServerHttpClient : public InferenceServerClient {
 
   // sync connection (reused, recreated on failure)
   int sync_fd_ = -1;
+  // TLS state for the sync path (OpenSSL; null when plain HTTP)
+  bool use_ssl_ = false;
+  HttpSslOptions ssl_options_;
+  void* ssl_ctx_ = nullptr;  // SSL_CTX*
+  void* ssl_ = nullptr;      // SSL* bound to sync_fd_
+  Error SslConnect();
+  void SslClose();
+  bool IoSend(const char* data, size_t n);
+  long IoRecv(char* buf, size_t n);
 
   // async machinery
   struct AsyncTransfer;

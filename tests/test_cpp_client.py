@@ -47,7 +47,7 @@ def _compile(name, main_src, grpc=False):
         return out
     cmd = ["g++", "-std=c++17", "-O1", f"-I{CPP}/include", "-Wall",
            str(main_src), *map(str, objs), "-o", str(out), "-lpthread",
-           "-lrt", "-lz", "-l:libnghttp2.so.14"]
+           "-lrt", "-lz", "-lssl", "-lcrypto", "-l:libnghttp2.so.14"]
     subprocess.run(cmd, check=True, capture_output=True, text=True)
     return out
 
