@@ -31,7 +31,10 @@ def _params_to_dict(pb_map):
 
 
 class GrpcServer:
-    def __init__(self, core=None, host="127.0.0.1", port=8001, max_workers=8):
+    def __init__(self, core=None, host="127.0.0.1", port=8001, max_workers=8,
+                 ssl_credentials=None):
+        """ssl_credentials: a grpc.ssl_server_credentials(...) object to
+        serve TLS (ALPN h2) instead of plaintext h2c."""
         self.core = core if core is not None else InferenceCore()
         self.host = host
         self.port = port
@@ -43,7 +46,11 @@ class GrpcServer:
             ],
         )
         self._server.add_generic_rpc_handlers((self._make_handler(),))
-        self.port = self._server.add_insecure_port(f"{host}:{port}")
+        if ssl_credentials is not None:
+            self.port = self._server.add_secure_port(
+                f"{host}:{port}", ssl_credentials)
+        else:
+            self.port = self._server.add_insecure_port(f"{host}:{port}")
 
     # ---- request translation ----
 

@@ -29,8 +29,10 @@ int MaxShareCount() {
 }
 
 std::shared_ptr<H2Connection> AcquireConnection(
-    const std::string& host, int port, Error* err) {
-  std::string key = host + ":" + std::to_string(port);
+    const std::string& host, int port, const H2SslOptions& ssl, Error* err) {
+  // TLS and plaintext channels to the same endpoint must never share.
+  std::string key = host + ":" + std::to_string(port) +
+                    (ssl.use_ssl ? "+tls" : "");
   std::lock_guard<std::mutex> lock(g_conn_mu);
   auto& vec = g_conns[key];
   int max_share = MaxShareCount();
@@ -41,15 +43,16 @@ std::shared_ptr<H2Connection> AcquireConnection(
     }
   }
   auto conn = std::make_shared<H2Connection>();
-  *err = conn->Connect(host, port);
+  *err = conn->Connect(host, port, ssl);
   if (!err->IsOk()) return nullptr;
   vec.push_back({conn, 1});
   return conn;
 }
 
-void ReleaseConnection(const std::string& host, int port,
+void ReleaseConnection(const std::string& host, int port, bool use_ssl,
                        const std::shared_ptr<H2Connection>& conn) {
-  std::string key = host + ":" + std::to_string(port);
+  std::string key = host + ":" + std::to_string(port) +
+                    (use_ssl ? "+tls" : "");
   std::lock_guard<std::mutex> lock(g_conn_mu);
   auto it = g_conns.find(key);
   if (it == g_conns.end()) return;
@@ -229,6 +232,18 @@ Error InferenceServerGrpcClient::Create(
   return Error::Success;
 }
 
+Error InferenceServerGrpcClient::Create(
+    std::unique_ptr<InferenceServerGrpcClient>* client,
+    const std::string& server_url, bool verbose, bool use_ssl,
+    const SslOptions& ssl_options) {
+  client->reset(new InferenceServerGrpcClient(server_url, verbose));
+  (*client)->ssl_.use_ssl = use_ssl;
+  (*client)->ssl_.root_certificates = ssl_options.root_certificates;
+  (*client)->ssl_.private_key = ssl_options.private_key;
+  (*client)->ssl_.certificate_chain = ssl_options.certificate_chain;
+  return Error::Success;
+}
+
 InferenceServerGrpcClient::InferenceServerGrpcClient(
     const std::string& url, bool verbose)
     : InferenceServerClient(verbose) {
@@ -245,7 +260,7 @@ InferenceServerGrpcClient::InferenceServerGrpcClient(
 InferenceServerGrpcClient::~InferenceServerGrpcClient() {
   StopStream();
   if (conn_ != nullptr) {
-    ReleaseConnection(host_, port_, conn_);
+    ReleaseConnection(host_, port_, ssl_.use_ssl, conn_);
     conn_.reset();
   }
 }
@@ -254,11 +269,11 @@ Error InferenceServerGrpcClient::EnsureConnected() {
   std::lock_guard<std::mutex> lock(conn_mu_);
   if (conn_ != nullptr && conn_->IsAlive()) return Error::Success;
   if (conn_ != nullptr) {
-    ReleaseConnection(host_, port_, conn_);
+    ReleaseConnection(host_, port_, ssl_.use_ssl, conn_);
     conn_.reset();
   }
   Error err = Error::Success;
-  conn_ = AcquireConnection(host_, port_, &err);
+  conn_ = AcquireConnection(host_, port_, ssl_, &err);
   return err;
 }
 
@@ -269,7 +284,7 @@ Error InferenceServerGrpcClient::AsyncUnaryCall(
 
   HeaderList headers = {
       {":method", "POST"},
-      {":scheme", "http"},
+      {":scheme", ssl_.use_ssl ? "https" : "http"},
       {":path", std::string(kService) + method},
       {":authority", host_ + ":" + std::to_string(port_)},
       {"te", "trailers"},
@@ -770,7 +785,7 @@ Error InferenceServerGrpcClient::StartStream(OnCompleteFn stream_callback) {
 
   HeaderList headers = {
       {":method", "POST"},
-      {":scheme", "http"},
+      {":scheme", ssl_.use_ssl ? "https" : "http"},
       {":path", std::string(kService) + "ModelStreamInfer"},
       {":authority", host_ + ":" + std::to_string(port_)},
       {"te", "trailers"},

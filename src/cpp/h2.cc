@@ -2,14 +2,23 @@
 
 #include <arpa/inet.h>
 #include <errno.h>
+#include <fcntl.h>
 #include <netdb.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <openssl/err.h>
+#include <openssl/pem.h>
+#include <openssl/ssl.h>
+#include <openssl/x509v3.h>
+#include <poll.h>
 #include <string.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <signal.h>
+
 #include <atomic>
+#include <mutex>
 
 // ---------------------------------------------------------------------------
 // libnghttp2 HPACK inflater ABI (system libnghttp2.so.14; prototypes
@@ -79,6 +88,113 @@ uint32_t rd32(const uint8_t* p) {
 H2Connection::~H2Connection() { Close(); }
 
 Error H2Connection::Connect(const std::string& host, int port) {
+  return Connect(host, port, H2SslOptions());
+}
+
+// Load a PEM bundle (possibly several certs) into the context's trust
+// store. Reference grpc clients take PEM strings, not paths, so we read
+// from a memory BIO rather than SSL_CTX_load_verify_locations.
+static Error AddPemRoots(SSL_CTX* ctx, const std::string& pem) {
+  BIO* bio = BIO_new_mem_buf(pem.data(), (int)pem.size());
+  if (bio == nullptr) return Error("BIO_new_mem_buf failed");
+  X509_STORE* store = SSL_CTX_get_cert_store(ctx);
+  int added = 0;
+  while (X509* cert = PEM_read_bio_X509(bio, nullptr, nullptr, nullptr)) {
+    X509_STORE_add_cert(store, cert);
+    X509_free(cert);
+    added++;
+  }
+  BIO_free(bio);
+  ERR_clear_error();  // trailing PEM_R_NO_START_LINE from the read loop
+  if (added == 0) return Error("no certificates found in root PEM");
+  return Error::Success;
+}
+
+Error H2Connection::TlsHandshake(const std::string& host,
+                                 const H2SslOptions& ssl_opts) {
+  // SSL_write has no MSG_NOSIGNAL: a peer-closed socket would raise
+  // SIGPIPE and kill the process. Same global-ignore libcurl installs.
+  static std::once_flag sigpipe_once;
+  std::call_once(sigpipe_once, [] { signal(SIGPIPE, SIG_IGN); });
+  SSL_CTX* ctx = SSL_CTX_new(TLS_client_method());
+  if (ctx == nullptr) return Error("SSL_CTX_new failed");
+  ssl_ctx_ = ctx;
+  if (!ssl_opts.root_certificates.empty()) {
+    Error err = AddPemRoots(ctx, ssl_opts.root_certificates);
+    if (!err.IsOk()) return err;
+  } else {
+    SSL_CTX_set_default_verify_paths(ctx);
+  }
+  if (!ssl_opts.certificate_chain.empty()) {
+    BIO* cbio = BIO_new_mem_buf(ssl_opts.certificate_chain.data(),
+                                (int)ssl_opts.certificate_chain.size());
+    X509* cert =
+        cbio ? PEM_read_bio_X509(cbio, nullptr, nullptr, nullptr) : nullptr;
+    if (cbio) BIO_free(cbio);
+    BIO* kbio = BIO_new_mem_buf(ssl_opts.private_key.data(),
+                                (int)ssl_opts.private_key.size());
+    EVP_PKEY* key =
+        kbio ? PEM_read_bio_PrivateKey(kbio, nullptr, nullptr, nullptr)
+             : nullptr;
+    if (kbio) BIO_free(kbio);
+    bool ok = cert != nullptr && key != nullptr &&
+              SSL_CTX_use_certificate(ctx, cert) == 1 &&
+              SSL_CTX_use_PrivateKey(ctx, key) == 1;
+    if (cert) X509_free(cert);
+    if (key) EVP_PKEY_free(key);
+    if (!ok) return Error("failed to load client certificate/key PEM");
+  }
+  SSL_CTX_set_verify(ctx, SSL_VERIFY_PEER, nullptr);
+  // gRPC requires ALPN "h2" (RFC 7301); grpcio rejects TLS without it.
+  static const unsigned char kAlpn[] = {2, 'h', '2'};
+  SSL_CTX_set_alpn_protos(ctx, kAlpn, sizeof(kAlpn));
+
+  SSL* ssl = SSL_new(ctx);
+  if (ssl == nullptr) return Error("SSL_new failed");
+  ssl_ = ssl;
+  SSL_set_fd(ssl, fd_);
+  SSL_set_mode(ssl, SSL_MODE_ENABLE_PARTIAL_WRITE |
+                        SSL_MODE_ACCEPT_MOVING_WRITE_BUFFER);
+  // Hostname checks: IP literals match iPAddress SANs, names match
+  // dNSName SANs (+SNI).
+  unsigned char ipbuf[16];
+  if (inet_pton(AF_INET, host.c_str(), ipbuf) == 1) {
+    X509_VERIFY_PARAM_set1_ip(SSL_get0_param(ssl), ipbuf, 4);
+  } else if (inet_pton(AF_INET6, host.c_str(), ipbuf) == 1) {
+    X509_VERIFY_PARAM_set1_ip(SSL_get0_param(ssl), ipbuf, 16);
+  } else {
+    SSL_set_tlsext_host_name(ssl, host.c_str());
+    SSL_set1_host(ssl, host.c_str());
+  }
+  if (SSL_connect(ssl) != 1) {
+    unsigned long e = ERR_get_error();
+    char ebuf[256];
+    ERR_error_string_n(e, ebuf, sizeof(ebuf));
+    long vr = SSL_get_verify_result(ssl);
+    std::string msg = std::string("TLS handshake with ") + host +
+                      " failed: " + ebuf;
+    if (vr != X509_V_OK) {
+      msg += std::string(" (verify: ") +
+             X509_verify_cert_error_string(vr) + ")";
+    }
+    return Error(msg);
+  }
+  const unsigned char* alpn = nullptr;
+  unsigned int alpn_len = 0;
+  SSL_get0_alpn_selected(ssl, &alpn, &alpn_len);
+  if (alpn != nullptr && !(alpn_len == 2 && memcmp(alpn, "h2", 2) == 0)) {
+    return Error("server negotiated ALPN protocol other than h2");
+  }
+  // Steady state runs the socket non-blocking so SSL_read/SSL_write
+  // never block while holding io_mu_ (reader + writer threads share ssl_).
+  int fl = fcntl(fd_, F_GETFL, 0);
+  fcntl(fd_, F_SETFL, fl | O_NONBLOCK);
+  use_tls_ = true;
+  return Error::Success;
+}
+
+Error H2Connection::Connect(const std::string& host, int port,
+                            const H2SslOptions& ssl_opts) {
   struct addrinfo hints;
   memset(&hints, 0, sizeof(hints));
   hints.ai_family = AF_UNSPEC;
@@ -100,6 +216,19 @@ Error H2Connection::Connect(const std::string& host, int port) {
   freeaddrinfo(res);
   if (fd_ < 0) {
     return Error("failed to connect to " + host + ":" + std::to_string(port));
+  }
+
+  if (ssl_opts.use_ssl) {
+    Error err = TlsHandshake(host, ssl_opts);
+    if (!err.IsOk()) {
+      if (ssl_ != nullptr) SSL_free((SSL*)ssl_);
+      if (ssl_ctx_ != nullptr) SSL_CTX_free((SSL_CTX*)ssl_ctx_);
+      ssl_ = nullptr;
+      ssl_ctx_ = nullptr;
+      close(fd_);
+      fd_ = -1;
+      return err;
+    }
   }
 
   nghttp2_hd_inflater* inf = nullptr;
@@ -152,6 +281,14 @@ void H2Connection::Close() {
     shutdown(fd_, SHUT_RDWR);
   }
   if (reader_.joinable()) reader_.join();
+  if (ssl_ != nullptr) {
+    SSL_free((SSL*)ssl_);
+    ssl_ = nullptr;
+  }
+  if (ssl_ctx_ != nullptr) {
+    SSL_CTX_free((SSL_CTX*)ssl_ctx_);
+    ssl_ctx_ = nullptr;
+  }
   if (fd_ >= 0) {
     close(fd_);
     fd_ = -1;
@@ -164,16 +301,60 @@ void H2Connection::Close() {
 }
 
 bool H2Connection::WriteRaw(const uint8_t* data, size_t n) {
+  if (!use_tls_) {
+    size_t sent = 0;
+    while (sent < n) {
+      ssize_t r = send(fd_, data + sent, n - sent, MSG_NOSIGNAL);
+      if (r <= 0) {
+        if (r < 0 && errno == EINTR) continue;
+        return false;
+      }
+      sent += (size_t)r;
+    }
+    return true;
+  }
+  // TLS: non-blocking fd; retry WANT_READ/WANT_WRITE via poll. io_mu_ is
+  // held across the poll — the reader's SSL_read waits, which is the same
+  // backpressure a blocking send() gives the plain path.
+  std::lock_guard<std::mutex> iolock(io_mu_);
+  SSL* ssl = (SSL*)ssl_;
   size_t sent = 0;
   while (sent < n) {
-    ssize_t r = send(fd_, data + sent, n - sent, MSG_NOSIGNAL);
-    if (r <= 0) {
-      if (r < 0 && errno == EINTR) continue;
-      return false;
+    int r = SSL_write(ssl, data + sent, (int)(n - sent));
+    if (r > 0) {
+      sent += (size_t)r;
+      continue;
     }
-    sent += (size_t)r;
+    int e = SSL_get_error(ssl, r);
+    if (e != SSL_ERROR_WANT_READ && e != SSL_ERROR_WANT_WRITE) return false;
+    struct pollfd pfd = {fd_, (short)(e == SSL_ERROR_WANT_READ ? POLLIN
+                                                               : POLLOUT),
+                         0};
+    if (poll(&pfd, 1, 5000) < 0 && errno != EINTR) return false;
+    if (pfd.revents & (POLLERR | POLLNVAL)) return false;
   }
   return true;
+}
+
+ssize_t H2Connection::IoRecv(char* buf, size_t n) {
+  if (!use_tls_) return recv(fd_, buf, n, 0);
+  while (true) {
+    {
+      std::lock_guard<std::mutex> iolock(io_mu_);
+      SSL* ssl = (SSL*)ssl_;
+      int r = SSL_read(ssl, buf, (int)n);
+      if (r > 0) return r;
+      int e = SSL_get_error(ssl, r);
+      if (e != SSL_ERROR_WANT_READ && e != SSL_ERROR_WANT_WRITE) {
+        return 0;  // clean close, fatal TLS error, or fd shut down
+      }
+    }
+    // Wait (lock released so writers can make progress) for the socket;
+    // Close()'s shutdown() wakes this poll and the next SSL_read fails.
+    struct pollfd pfd = {fd_, POLLIN, 0};
+    if (poll(&pfd, 1, 200) < 0 && errno != EINTR) return 0;
+    if (pfd.revents & (POLLERR | POLLNVAL)) return 0;
+  }
 }
 
 bool H2Connection::WriteFrame(uint8_t type, uint8_t flags, int32_t stream_id,
@@ -325,7 +506,7 @@ void H2Connection::ReaderLoop() {
   while (true) {
     // need 9-byte frame header
     while (buf.size() < 9) {
-      ssize_t r = recv(fd_, chunk, sizeof(chunk), 0);
+      ssize_t r = IoRecv(chunk, sizeof(chunk));
       if (r <= 0) {
         alive_ = false;
         FailAllStreams("h2 connection closed by peer");
@@ -340,7 +521,7 @@ void H2Connection::ReaderLoop() {
     int32_t stream_id = (int32_t)(rd32((const uint8_t*)buf.data() + 5) &
                                   0x7FFFFFFF);
     while (buf.size() < 9 + len) {
-      ssize_t r = recv(fd_, chunk, sizeof(chunk), 0);
+      ssize_t r = IoRecv(chunk, sizeof(chunk));
       if (r <= 0) {
         alive_ = false;
         FailAllStreams("h2 connection closed mid-frame");

@@ -6,6 +6,7 @@
 #include <netdb.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <signal.h>
 #include <string.h>
 #include <sys/epoll.h>
 #include <sys/socket.h>
@@ -348,7 +349,12 @@ Error InferenceServerHttpClient::Create(
 Error InferenceServerHttpClient::SslConnect() {
   if (ssl_ctx_ == nullptr) {
     static std::once_flag ssl_init;
-    std::call_once(ssl_init, [] { SSL_library_init(); });
+    // SIGPIPE: SSL_write lacks MSG_NOSIGNAL; a peer reset would kill
+    // the process without this (libcurl installs the same ignore).
+    std::call_once(ssl_init, [] {
+      SSL_library_init();
+      signal(SIGPIPE, SIG_IGN);
+    });
     SSL_CTX* ctx = SSL_CTX_new(TLS_client_method());
     if (ctx == nullptr) return Error("SSL_CTX_new failed");
     if (!ssl_options_.ca_info.empty()) {

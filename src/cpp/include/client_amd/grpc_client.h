@@ -20,6 +20,14 @@ namespace client_amd {
 
 class InferResultGrpc;
 
+// TLS options, reference grpc_client.h SslOptions:112-136 — fields are
+// PEM *contents* (read the files yourself, as the reference examples do).
+struct SslOptions {
+  std::string root_certificates;  // CA bundle PEM; empty = system roots
+  std::string private_key;        // client key PEM (mTLS)
+  std::string certificate_chain;  // client cert PEM (mTLS)
+};
+
 class InferenceServerGrpcClient : public InferenceServerClient {
  public:
   ~InferenceServerGrpcClient() override;
@@ -27,6 +35,11 @@ class InferenceServerGrpcClient : public InferenceServerClient {
   static Error Create(
       std::unique_ptr<InferenceServerGrpcClient>* client,
       const std::string& server_url, bool verbose = false);
+  // TLS variant (ALPN h2 over OpenSSL; hostname/IP-SAN verification on).
+  static Error Create(
+      std::unique_ptr<InferenceServerGrpcClient>* client,
+      const std::string& server_url, bool verbose, bool use_ssl,
+      const SslOptions& ssl_options);
 
   Error IsServerLive(bool* live);
   Error IsServerReady(bool* ready);
@@ -108,6 +121,7 @@ class InferenceServerGrpcClient : public InferenceServerClient {
 
   std::string host_;
   int port_;
+  H2SslOptions ssl_;
   // Shared h2 connection from the global per-url cache (reference
   // grpc_client.cc:80-152: up to TRITON_CLIENT_GRPC_CHANNEL_MAX_SHARE_COUNT
   // clients share one channel before a new one is created; HTTP/2

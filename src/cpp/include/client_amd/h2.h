@@ -26,6 +26,16 @@ namespace client_amd {
 
 using HeaderList = std::vector<std::pair<std::string, std::string>>;
 
+// TLS settings for an h2 connection (gRPC "grpcs"). Fields hold PEM
+// *contents* (not paths), matching the reference grpc_client.h
+// SslOptions:112-136 semantics where callers read files themselves.
+struct H2SslOptions {
+  bool use_ssl = false;
+  std::string root_certificates;  // PEM CA bundle; empty = system default
+  std::string private_key;        // client key PEM (mTLS)
+  std::string certificate_chain;  // client cert PEM (mTLS)
+};
+
 class H2Connection {
  public:
   struct StreamHandler {
@@ -40,6 +50,7 @@ class H2Connection {
   ~H2Connection();
 
   Error Connect(const std::string& host, int port);
+  Error Connect(const std::string& host, int port, const H2SslOptions& ssl);
   void Close();
   bool IsAlive() const { return alive_; }
 
@@ -66,6 +77,11 @@ class H2Connection {
 
   void ReaderLoop();
   bool WriteRaw(const uint8_t* data, size_t n);
+  Error TlsHandshake(const std::string& host, const H2SslOptions& ssl);
+  // Blocking-semantics read: >0 bytes, <=0 on close/error. Plain mode is
+  // a blocking recv(); TLS mode polls a non-blocking fd and serializes
+  // SSL_read/SSL_write through io_mu_ (one SSL object, two threads).
+  ssize_t IoRecv(char* buf, size_t n);
   bool WriteFrame(uint8_t type, uint8_t flags, int32_t stream_id,
                   const std::string& payload);
   void HandleFrame(uint8_t type, uint8_t flags, int32_t stream_id,
@@ -74,6 +90,10 @@ class H2Connection {
   std::string EncodeHeaders(const HeaderList& headers);
 
   int fd_ = -1;
+  bool use_tls_ = false;
+  void* ssl_ctx_ = nullptr;  // SSL_CTX* (avoid OpenSSL headers here)
+  void* ssl_ = nullptr;      // SSL*
+  std::mutex io_mu_;         // guards every SSL_read/SSL_write
   std::thread reader_;
   std::atomic_bool alive_{false};
   bool exiting_ = false;
