@@ -1,0 +1,31 @@
+# Round-end GPU validation: driver mirror (build+smoke, pytest -m gpu,
+# bench N=1) + long-generation decode crossing hipGraph buckets
+# (256/512/768) at concurrency 4.
+set -x
+cd /root/repo
+export HSA_ENABLE_IPC_MODE_LEGACY=0
+mkdir -p gpurun_out
+
+timeout 420 python -c "import __graft_entry__ as g; g.build(); g.smoke(); print('SMOKE-OK')" > gpurun_out/final_entry.log 2>&1
+echo "entry rc=$?"
+tail -2 gpurun_out/final_entry.log
+
+timeout 900 python -m pytest tests -m gpu -q > gpurun_out/final_pytest.log 2>&1
+echo "pytest rc=$?"
+tail -3 gpurun_out/final_pytest.log
+
+timeout 420 python bench.py --steps 20 --warmup 5 > gpurun_out/final_bench.log 2>&1
+echo "bench rc=$?"
+tail -1 gpurun_out/final_bench.log
+
+python -m client_amd.server --models llama3_8b --grpc-port 18001 > gpurun_out/final_llama_server.log 2>&1 &
+SRV=$!
+for i in $(seq 90); do grep -q GRPC_READY gpurun_out/final_llama_server.log && break; sleep 2; done
+timeout 420 python -m client_amd.perf.genai -m llama3_8b -u 127.0.0.1:18001 \
+  --concurrency 4 --requests 8 --prompt-tokens 128 --output-tokens 700 \
+  --json gpurun_out/genai_llama3_8b_longgen700_c4.json > gpurun_out/final_genai.log 2>&1
+echo "genai rc=$?"
+cat gpurun_out/genai_llama3_8b_longgen700_c4.json 2>/dev/null
+kill $SRV 2>/dev/null
+wait $SRV 2>/dev/null
+echo DONE

@@ -29,23 +29,31 @@ int MaxShareCount() {
 }
 
 std::shared_ptr<H2Connection> AcquireConnection(
-    const std::string& host, int port, const H2SslOptions& ssl, Error* err) {
+    const std::string& host, int port, const H2SslOptions& ssl,
+    const H2KeepAlive& ka, bool use_cache, Error* err) {
   // TLS and plaintext channels to the same endpoint must never share.
   std::string key = host + ":" + std::to_string(port) +
                     (ssl.use_ssl ? "+tls" : "");
-  std::lock_guard<std::mutex> lock(g_conn_mu);
-  auto& vec = g_conns[key];
-  int max_share = MaxShareCount();
-  for (auto& entry : vec) {
-    if (entry.conn->IsAlive() && entry.share_count < max_share) {
-      entry.share_count++;
-      return entry.conn;
+  std::unique_lock<std::mutex> lock(g_conn_mu);
+  if (use_cache) {
+    auto& vec = g_conns[key];
+    int max_share = MaxShareCount();
+    for (auto& entry : vec) {
+      if (entry.conn->IsAlive() && entry.share_count < max_share) {
+        entry.share_count++;
+        return entry.conn;
+      }
     }
   }
+  lock.unlock();
   auto conn = std::make_shared<H2Connection>();
   *err = conn->Connect(host, port, ssl);
   if (!err->IsOk()) return nullptr;
-  vec.push_back({conn, 1});
+  conn->StartKeepalive(ka);
+  if (use_cache) {
+    lock.lock();
+    g_conns[key].push_back({conn, 1});
+  }
   return conn;
 }
 
@@ -55,7 +63,10 @@ void ReleaseConnection(const std::string& host, int port, bool use_ssl,
                     (use_ssl ? "+tls" : "");
   std::lock_guard<std::mutex> lock(g_conn_mu);
   auto it = g_conns.find(key);
-  if (it == g_conns.end()) return;
+  if (it == g_conns.end()) {
+    conn->Close();  // private (uncached) channel
+    return;
+  }
   auto& vec = it->second;
   for (size_t i = 0; i < vec.size(); ++i) {
     if (vec[i].conn == conn) {
@@ -227,20 +238,24 @@ struct InferenceServerGrpcClient::BidiState {
 
 Error InferenceServerGrpcClient::Create(
     std::unique_ptr<InferenceServerGrpcClient>* client,
-    const std::string& server_url, bool verbose) {
-  client->reset(new InferenceServerGrpcClient(server_url, verbose));
-  return Error::Success;
-}
-
-Error InferenceServerGrpcClient::Create(
-    std::unique_ptr<InferenceServerGrpcClient>* client,
     const std::string& server_url, bool verbose, bool use_ssl,
-    const SslOptions& ssl_options) {
+    const SslOptions& ssl_options, const KeepAliveOptions& keepalive_options,
+    const bool use_cached_channel) {
   client->reset(new InferenceServerGrpcClient(server_url, verbose));
   (*client)->ssl_.use_ssl = use_ssl;
   (*client)->ssl_.root_certificates = ssl_options.root_certificates;
   (*client)->ssl_.private_key = ssl_options.private_key;
   (*client)->ssl_.certificate_chain = ssl_options.certificate_chain;
+  // INT32_MAX = grpc-core's "keepalive off" sentinel
+  (*client)->keepalive_.enabled =
+      keepalive_options.keepalive_time_ms != 0x7FFFFFFF;
+  (*client)->keepalive_.time_ms = keepalive_options.keepalive_time_ms;
+  (*client)->keepalive_.timeout_ms = keepalive_options.keepalive_timeout_ms;
+  (*client)->keepalive_.permit_without_calls =
+      keepalive_options.keepalive_permit_without_calls;
+  (*client)->keepalive_.max_pings_without_data =
+      keepalive_options.http2_max_pings_without_data;
+  (*client)->use_cached_channel_ = use_cached_channel;
   return Error::Success;
 }
 
@@ -273,7 +288,8 @@ Error InferenceServerGrpcClient::EnsureConnected() {
     conn_.reset();
   }
   Error err = Error::Success;
-  conn_ = AcquireConnection(host_, port_, ssl_, &err);
+  conn_ = AcquireConnection(host_, port_, ssl_, keepalive_,
+                            use_cached_channel_, &err);
   return err;
 }
 

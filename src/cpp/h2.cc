@@ -277,6 +277,8 @@ void H2Connection::Close() {
     exiting_ = true;
   }
   alive_ = false;
+  ka_cv_.notify_all();
+  if (keepalive_.joinable()) keepalive_.join();
   if (fd_ >= 0) {
     shutdown(fd_, SHUT_RDWR);
   }
@@ -298,6 +300,54 @@ void H2Connection::Close() {
     hpack_inflater_ = nullptr;
   }
   window_cv_.notify_all();
+}
+
+void H2Connection::StartKeepalive(const H2KeepAlive& ka) {
+  if (!ka.enabled || ka.time_ms <= 0 || keepalive_.joinable()) return;
+  {
+    std::lock_guard<std::mutex> lock(mu_);
+    ka_ = ka;
+  }
+  keepalive_ = std::thread(&H2Connection::KeepaliveLoop, this);
+}
+
+// gRPC-style transport watchdog (grpc/doc/keepalive.md): PING every
+// time_ms; a missing ACK within timeout_ms kills the connection so
+// blocked callers fail fast instead of hanging on a dead peer.
+void H2Connection::KeepaliveLoop() {
+  std::unique_lock<std::mutex> lock(mu_);
+  while (!exiting_ && alive_) {
+    ka_cv_.wait_for(lock, std::chrono::milliseconds(ka_.time_ms));
+    if (exiting_ || !alive_) return;
+    if (!ka_.permit_without_calls && streams_.empty()) continue;
+    if (data_epoch_ == last_ping_epoch_) {
+      if (ka_.max_pings_without_data > 0 &&
+          pings_without_data_ >= ka_.max_pings_without_data) {
+        continue;  // idle transport: stop pinging (grpc-core rule)
+      }
+      pings_without_data_++;
+    } else {
+      pings_without_data_ = 0;
+    }
+    last_ping_epoch_ = data_epoch_;
+    uint64_t acked_before = pings_acked_;
+    lock.unlock();
+    bool sent = WriteFrame(kFramePing, 0, 0, std::string(8, '\0'));
+    lock.lock();
+    bool acked =
+        sent && ka_cv_.wait_for(
+                    lock, std::chrono::milliseconds(ka_.timeout_ms), [&] {
+                      return exiting_ || pings_acked_ > acked_before;
+                    });
+    if (exiting_) return;
+    if (!acked) {
+      lock.unlock();
+      alive_ = false;
+      if (fd_ >= 0) shutdown(fd_, SHUT_RDWR);
+      FailAllStreams("keepalive watchdog timed out");
+      return;
+    }
+  }
 }
 
 bool H2Connection::WriteRaw(const uint8_t* data, size_t n) {
@@ -410,6 +460,7 @@ Error H2Connection::StartStream(const HeaderList& headers,
     Stream& s = streams_[id];
     s.handler = std::move(handler);
     s.send_window = peer_initial_window_;
+    data_epoch_++;
   }
   std::string buf;
   be24(&buf, (uint32_t)block.size());
@@ -428,6 +479,7 @@ Error H2Connection::SendData(int32_t stream_id, const uint8_t* data, size_t n,
                              bool end_stream) {
   {
     std::lock_guard<std::mutex> lock(mu_);
+    data_epoch_++;
     if (streams_.find(stream_id) == streams_.end()) {
       // stream already finished/reset: sending DATA would be a
       // protocol violation that could kill the (shared) connection
@@ -698,6 +750,10 @@ void H2Connection::HandleFrame(uint8_t type, uint8_t flags, int32_t stream_id,
       if (!(flags & kFlagAck)) {
         WriteFrame(kFramePing, kFlagAck, 0,
                    std::string((const char*)payload, len));
+      } else {
+        std::lock_guard<std::mutex> lock(mu_);
+        pings_acked_++;
+        ka_cv_.notify_all();
       }
       break;
     }

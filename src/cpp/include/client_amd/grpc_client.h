@@ -28,18 +28,29 @@ struct SslOptions {
   std::string certificate_chain;  // client cert PEM (mTLS)
 };
 
+// gRPC keepalive knobs, reference grpc_client.h KeepAliveOptions:62-83
+// (grpc/doc/keepalive.md). Default keepalive_time_ms of INT32_MAX
+// means "never ping" — same as grpc-core.
+struct KeepAliveOptions {
+  int keepalive_time_ms = 0x7FFFFFFF;
+  int keepalive_timeout_ms = 20000;
+  bool keepalive_permit_without_calls = false;
+  int http2_max_pings_without_data = 2;
+};
+
 class InferenceServerGrpcClient : public InferenceServerClient {
  public:
   ~InferenceServerGrpcClient() override;
 
+  // Reference grpc_client.h:120-126 signature. TLS = OpenSSL + ALPN h2
+  // with hostname/IP-SAN verification; keepalive = h2 PING watchdog;
+  // use_cached_channel=false forces a private (unshared) connection.
   static Error Create(
       std::unique_ptr<InferenceServerGrpcClient>* client,
-      const std::string& server_url, bool verbose = false);
-  // TLS variant (ALPN h2 over OpenSSL; hostname/IP-SAN verification on).
-  static Error Create(
-      std::unique_ptr<InferenceServerGrpcClient>* client,
-      const std::string& server_url, bool verbose, bool use_ssl,
-      const SslOptions& ssl_options);
+      const std::string& server_url, bool verbose = false,
+      bool use_ssl = false, const SslOptions& ssl_options = SslOptions(),
+      const KeepAliveOptions& keepalive_options = KeepAliveOptions(),
+      const bool use_cached_channel = true);
 
   Error IsServerLive(bool* live);
   Error IsServerReady(bool* ready);
@@ -122,6 +133,8 @@ class InferenceServerGrpcClient : public InferenceServerClient {
   std::string host_;
   int port_;
   H2SslOptions ssl_;
+  H2KeepAlive keepalive_;
+  bool use_cached_channel_ = true;
   // Shared h2 connection from the global per-url cache (reference
   // grpc_client.cc:80-152: up to TRITON_CLIENT_GRPC_CHANNEL_MAX_SHARE_COUNT
   // clients share one channel before a new one is created; HTTP/2

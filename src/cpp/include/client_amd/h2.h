@@ -36,6 +36,17 @@ struct H2SslOptions {
   std::string certificate_chain;  // client cert PEM (mTLS)
 };
 
+// h2 PING keepalive (gRPC keepalive.md semantics): send a PING every
+// time_ms; if the ACK doesn't arrive within timeout_ms the connection
+// is declared dead and every in-flight stream fails.
+struct H2KeepAlive {
+  bool enabled = false;
+  int time_ms = 0;
+  int timeout_ms = 20000;
+  bool permit_without_calls = false;
+  int max_pings_without_data = 2;  // 0 = unlimited
+};
+
 class H2Connection {
  public:
   struct StreamHandler {
@@ -51,6 +62,8 @@ class H2Connection {
 
   Error Connect(const std::string& host, int port);
   Error Connect(const std::string& host, int port, const H2SslOptions& ssl);
+  // Start the PING watchdog (call once, after Connect succeeds).
+  void StartKeepalive(const H2KeepAlive& ka);
   void Close();
   bool IsAlive() const { return alive_; }
 
@@ -76,6 +89,7 @@ class H2Connection {
   };
 
   void ReaderLoop();
+  void KeepaliveLoop();
   bool WriteRaw(const uint8_t* data, size_t n);
   Error TlsHandshake(const std::string& host, const H2SslOptions& ssl);
   // Blocking-semantics read: >0 bytes, <=0 on close/error. Plain mode is
@@ -108,6 +122,15 @@ class H2Connection {
   int64_t conn_recv_consumed_ = 0;
   uint32_t peer_max_frame_ = 16384;
   int32_t peer_initial_window_ = 65535;
+
+  // keepalive watchdog state (guarded by mu_; thread woken via ka_cv_)
+  H2KeepAlive ka_;
+  std::thread keepalive_;
+  std::condition_variable ka_cv_;
+  uint64_t pings_acked_ = 0;
+  uint64_t data_epoch_ = 0;        // bumped on every HEADERS/DATA we send
+  uint64_t last_ping_epoch_ = 0;   // data_epoch_ when the last ping went out
+  int pings_without_data_ = 0;
 
   // HEADERS/CONTINUATION reassembly
   int32_t pending_headers_stream_ = 0;

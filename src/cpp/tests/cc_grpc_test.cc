@@ -7,6 +7,7 @@
 #include <cstring>
 #include <iostream>
 #include <mutex>
+#include <thread>
 #include <vector>
 
 #include "client_amd/grpc_client.h"
@@ -287,6 +288,31 @@ int main(int argc, char** argv) {
   InferStat stat;
   CHECK_OK(client->ClientInferStat(&stat));
   CHECK(stat.completed_request_count >= 10);
+
+  // ---- keepalive + private channel ----
+  // Aggressive PING period on an unshared channel; grpcio GOAWAYs on a
+  // malformed PING, so surviving several periods + a second infer
+  // proves the watchdog frames are well-formed and ACK-tracked.
+  {
+    KeepAliveOptions ka;
+    ka.keepalive_time_ms = 100;
+    ka.keepalive_timeout_ms = 2000;
+    ka.keepalive_permit_without_calls = true;
+    ka.http2_max_pings_without_data = 0;
+    std::unique_ptr<InferenceServerGrpcClient> ka_client;
+    CHECK_OK(InferenceServerGrpcClient::Create(
+        &ka_client, url, false, false, SslOptions(), ka,
+        /*use_cached_channel=*/false));
+    InferResult* ka_res = nullptr;
+    CHECK_OK(ka_client->Infer(&ka_res, options, {input0, input1}));
+    delete ka_res;
+    std::this_thread::sleep_for(std::chrono::milliseconds(450));
+    ka_res = nullptr;
+    CHECK_OK(ka_client->Infer(&ka_res, options, {input0, input1}));
+    CHECK_OK(ka_res->RawData("OUTPUT0", &buf, &nbytes));
+    CHECK(nbytes == 64);
+    delete ka_res;
+  }
 
   std::cout << "cc_grpc_test: ALL PASSED" << std::endl;
   return 0;
