@@ -200,7 +200,7 @@ def _hipcc_example(name, extra_models="simple", with_kernels=False):
     subprocess.run(
         ["/opt/rocm/bin/hipcc", "--offload-arch=gfx950", "-std=c++17", "-O1",
          f"-I{CPP}/include", "-DTRITON_ENABLE_HIP", "-x", "hip",
-         *map(str, srcs), "-o", str(binary), "-lpthread", "-lrt", "-lz"],
+         *map(str, srcs), "-o", str(binary), "-lpthread", "-lrt", "-lz", "-lssl", "-lcrypto"],
         check=True, capture_output=True, text=True)
     server = subprocess.Popen(
         [sys.executable, "-m", "client_amd.server", "--http-port", "18511",
@@ -260,7 +260,7 @@ def test_cc_hipshm_example_gpu():
     subprocess.run(
         ["/opt/rocm/bin/hipcc", "--offload-arch=gfx950", "-std=c++17", "-O1",
          f"-I{CPP}/include", "-DTRITON_ENABLE_HIP", *map(str, srcs),
-         "-o", str(binary), "-lpthread", "-lrt", "-lz"],
+         "-o", str(binary), "-lpthread", "-lrt", "-lz", "-lssl", "-lcrypto"],
         check=True, capture_output=True, text=True)
     server = subprocess.Popen(
         [sys.executable, "-m", "client_amd.server", "--http-port", "18511",
