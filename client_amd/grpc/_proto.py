@@ -8,7 +8,7 @@ reference vendors it at src/rust/triton-client/proto/grpc_service.proto;
 e.g. ModelInferRequest :575-706 with raw_input_contents = 7,
 ModelStreamInferResponse :821-840, shm messages :1419-1666, statistics
 :881-1235, trace :1673-1741, log :1743-1780), so the bytes on the wire
-are compatible with any KServe-v2 peer. ModelConfig is the commonly-used
+are compatible with any KServe-v2 peer. ModelConfig is the full
 subset of model_config.proto (full 2180-line schema: round 2).
 """
 
@@ -119,7 +119,31 @@ _MESSAGES = {
         ("error_message", 1, "string"),
         ("infer_response", 2, "msg:ModelInferResponse"),
     ],
-    # ---- model config (subset) ----
+    # ---- model config (full reference schema, model_config.proto:86-2180) ----
+    "ModelRateLimiter.Resource": [
+        ("name", 1, "string"),
+        ("global", 2, "bool"),
+        ("count", 3, "uint32"),
+    ],
+    "ModelRateLimiter": [
+        ("resources", 1, "msg:ModelRateLimiter.Resource", "r"),
+        ("priority", 2, "uint32"),
+    ],
+    "ModelInstanceGroup.SecondaryDevice": [
+        ("kind", 1, "enum:ModelInstanceGroup.SecondaryDevice.SecondaryDeviceKind"),
+        ("device_id", 2, "int64"),
+    ],
+    "ModelInstanceGroup": [
+        ("name", 1, "string"),
+        ("kind", 4, "enum:ModelInstanceGroup.Kind"),
+        ("count", 2, "int32"),
+        ("rate_limiter", 6, "msg:ModelRateLimiter"),
+        ("gpus", 3, "int32", "r"),
+        ("secondary_devices", 8, "msg:ModelInstanceGroup.SecondaryDevice", "r"),
+        ("profile", 5, "string", "r"),
+        ("passive", 7, "bool"),
+        ("host_policy", 9, "string"),
+    ],
     "ModelTensorReshape": [("shape", 1, "int64", "r")],
     "ModelInput": [
         ("name", 1, "string"),
@@ -130,43 +154,217 @@ _MESSAGES = {
         ("is_shape_tensor", 6, "bool"),
         ("allow_ragged_batch", 7, "bool"),
         ("optional", 8, "bool"),
+        ("is_non_linear_format_io", 9, "bool"),
     ],
     "ModelOutput": [
         ("name", 1, "string"),
         ("data_type", 2, "enum:DataType"),
         ("dims", 3, "int64", "r"),
-        ("reshape", 4, "msg:ModelTensorReshape"),
-        ("label_filename", 5, "string"),
+        ("label_filename", 4, "string"),
+        ("reshape", 5, "msg:ModelTensorReshape"),
         ("is_shape_tensor", 6, "bool"),
+        ("is_non_linear_format_io", 7, "bool"),
     ],
-    "ModelParameter": [("string_value", 1, "string")],
+    "BatchInput": [
+        ("kind", 1, "enum:BatchInput.Kind"),
+        ("target_name", 2, "string", "r"),
+        ("data_type", 3, "enum:DataType"),
+        ("source_input", 4, "string", "r"),
+    ],
+    "BatchOutput": [
+        ("target_name", 1, "string", "r"),
+        ("kind", 2, "enum:BatchOutput.Kind"),
+        ("source_input", 3, "string", "r"),
+    ],
+    "ModelVersionPolicy.Latest": [("num_versions", 1, "uint32")],
+    "ModelVersionPolicy.All": [],
+    "ModelVersionPolicy.Specific": [("versions", 1, "int64", "r")],
+    "ModelVersionPolicy": [
+        ("latest", 1, "msg:ModelVersionPolicy.Latest", "opolicy_choice"),
+        ("all", 2, "msg:ModelVersionPolicy.All", "opolicy_choice"),
+        ("specific", 3, "msg:ModelVersionPolicy.Specific", "opolicy_choice"),
+    ],
+    "ModelOptimizationPolicy.Graph": [("level", 1, "int32")],
+    "ModelOptimizationPolicy.Cuda.GraphSpec.Shape": [("dim", 1, "int64", "r")],
+    "ModelOptimizationPolicy.Cuda.GraphSpec.LowerBound": [
+        ("batch_size", 1, "int32"),
+        ("input", 2, "map:string:msg:ModelOptimizationPolicy.Cuda.GraphSpec.Shape"),
+    ],
+    "ModelOptimizationPolicy.Cuda.GraphSpec": [
+        ("batch_size", 1, "int32"),
+        ("input", 2, "map:string:msg:ModelOptimizationPolicy.Cuda.GraphSpec.Shape"),
+        ("graph_lower_bound", 3, "msg:ModelOptimizationPolicy.Cuda.GraphSpec.LowerBound"),
+    ],
+    "ModelOptimizationPolicy.Cuda": [
+        ("graphs", 1, "bool"),
+        ("busy_wait_events", 2, "bool"),
+        ("graph_spec", 3, "msg:ModelOptimizationPolicy.Cuda.GraphSpec", "r"),
+        ("output_copy_stream", 4, "bool"),
+    ],
+    "ModelOptimizationPolicy.ExecutionAccelerators.Accelerator": [
+        ("name", 1, "string"),
+        ("parameters", 2, "map:string:string"),
+    ],
+    "ModelOptimizationPolicy.ExecutionAccelerators": [
+        ("gpu_execution_accelerator", 1,
+         "msg:ModelOptimizationPolicy.ExecutionAccelerators.Accelerator", "r"),
+        ("cpu_execution_accelerator", 2,
+         "msg:ModelOptimizationPolicy.ExecutionAccelerators.Accelerator", "r"),
+    ],
+    "ModelOptimizationPolicy.PinnedMemoryBuffer": [("enable", 1, "bool")],
+    "ModelOptimizationPolicy": [
+        ("graph", 1, "msg:ModelOptimizationPolicy.Graph"),
+        ("priority", 2, "enum:ModelOptimizationPolicy.ModelPriority"),
+        ("cuda", 3, "msg:ModelOptimizationPolicy.Cuda"),
+        ("execution_accelerators", 4,
+         "msg:ModelOptimizationPolicy.ExecutionAccelerators"),
+        ("input_pinned_memory", 5,
+         "msg:ModelOptimizationPolicy.PinnedMemoryBuffer"),
+        ("output_pinned_memory", 6,
+         "msg:ModelOptimizationPolicy.PinnedMemoryBuffer"),
+        ("gather_kernel_buffer_threshold", 7, "uint32"),
+        ("eager_batching", 8, "bool"),
+    ],
+    "ModelQueuePolicy": [
+        ("timeout_action", 1, "enum:ModelQueuePolicy.TimeoutAction"),
+        ("default_timeout_microseconds", 2, "uint64"),
+        ("allow_timeout_override", 3, "bool"),
+        ("max_queue_size", 4, "uint32"),
+    ],
     "ModelDynamicBatching": [
         ("preferred_batch_size", 1, "int32", "r"),
         ("max_queue_delay_microseconds", 2, "uint64"),
+        ("preserve_ordering", 3, "bool"),
+        ("priority_levels", 4, "uint64"),
+        ("default_priority_level", 5, "uint64"),
+        ("default_queue_policy", 6, "msg:ModelQueuePolicy"),
+        ("priority_queue_policy", 7, "map:uint64:msg:ModelQueuePolicy"),
+    ],
+    "ModelSequenceBatching.Control": [
+        ("kind", 1, "enum:ModelSequenceBatching.Control.Kind"),
+        ("int32_false_true", 2, "int32", "r"),
+        ("fp32_false_true", 3, "float", "r"),
+        ("bool_false_true", 5, "bool", "r"),
+        ("data_type", 4, "enum:DataType"),
+    ],
+    "ModelSequenceBatching.ControlInput": [
+        ("name", 1, "string"),
+        ("control", 2, "msg:ModelSequenceBatching.Control", "r"),
+    ],
+    "ModelSequenceBatching.InitialState": [
+        ("data_type", 1, "enum:DataType"),
+        ("dims", 2, "int64", "r"),
+        ("zero_data", 3, "bool", "ostate_data"),
+        ("data_file", 4, "string", "ostate_data"),
+        ("name", 5, "string"),
+    ],
+    "ModelSequenceBatching.State": [
+        ("input_name", 1, "string"),
+        ("output_name", 2, "string"),
+        ("data_type", 3, "enum:DataType"),
+        ("dims", 4, "int64", "r"),
+        ("initial_state", 5, "msg:ModelSequenceBatching.InitialState", "r"),
+        ("use_same_buffer_for_input_output", 6, "bool"),
+        ("use_growable_memory", 7, "bool"),
+    ],
+    "ModelSequenceBatching.StrategyDirect": [
+        ("max_queue_delay_microseconds", 1, "uint64"),
+        ("minimum_slot_utilization", 2, "float"),
+    ],
+    "ModelSequenceBatching.StrategyOldest": [
+        ("max_candidate_sequences", 1, "int32"),
+        ("preferred_batch_size", 2, "int32", "r"),
+        ("max_queue_delay_microseconds", 3, "uint64"),
+        ("preserve_ordering", 4, "bool"),
     ],
     "ModelSequenceBatching": [
+        ("direct", 3, "msg:ModelSequenceBatching.StrategyDirect",
+         "ostrategy_choice"),
+        ("oldest", 4, "msg:ModelSequenceBatching.StrategyOldest",
+         "ostrategy_choice"),
         ("max_sequence_idle_microseconds", 1, "uint64"),
+        ("control_input", 2, "msg:ModelSequenceBatching.ControlInput", "r"),
+        ("state", 5, "msg:ModelSequenceBatching.State", "r"),
+        ("iterative_sequence", 6, "bool"),
     ],
-    "ModelTransactionPolicy": [("decoupled", 1, "bool")],
-    "ModelInstanceGroup": [
+    "ModelEnsembling.Step": [
+        ("model_name", 1, "string"),
+        ("model_version", 2, "int64"),
+        ("input_map", 3, "map:string:string"),
+        ("output_map", 4, "map:string:string"),
+        ("model_namespace", 5, "string"),
+    ],
+    "ModelEnsembling": [
+        ("step", 1, "msg:ModelEnsembling.Step", "r"),
+        ("max_inflight_requests", 2, "uint32"),
+    ],
+    "ModelParameter": [("string_value", 1, "string")],
+    "ModelWarmup.Input": [
+        ("data_type", 1, "enum:DataType"),
+        ("dims", 2, "int64", "r"),
+        ("zero_data", 3, "bool", "oinput_data_type"),
+        ("random_data", 4, "bool", "oinput_data_type"),
+        ("input_data_file", 5, "string", "oinput_data_type"),
+    ],
+    "ModelWarmup": [
         ("name", 1, "string"),
-        ("count", 2, "int32"),
-        ("gpus", 3, "int32", "r"),
-        ("kind", 4, "enum:ModelInstanceGroup.Kind"),
+        ("batch_size", 2, "uint32"),
+        ("inputs", 3, "map:string:msg:ModelWarmup.Input"),
+        ("count", 4, "uint32"),
+    ],
+    "ModelOperations": [("op_library_filename", 1, "string", "r")],
+    "ModelTransactionPolicy": [("decoupled", 1, "bool")],
+    "ModelRepositoryAgents.Agent": [
+        ("name", 1, "string"),
+        ("parameters", 2, "map:string:string"),
+    ],
+    "ModelRepositoryAgents": [
+        ("agents", 1, "msg:ModelRepositoryAgents.Agent", "r"),
+    ],
+    "ModelResponseCache": [("enable", 1, "bool")],
+    "ModelMetrics.MetricControl.MetricIdentifier": [("family", 1, "string")],
+    "ModelMetrics.MetricControl.HistogramOptions": [
+        ("buckets", 1, "double", "r"),
+    ],
+    "ModelMetrics.MetricControl": [
+        ("metric_identifier", 1,
+         "msg:ModelMetrics.MetricControl.MetricIdentifier"),
+        ("histogram_options", 2,
+         "msg:ModelMetrics.MetricControl.HistogramOptions",
+         "ometric_options"),
+    ],
+    "ModelMetrics": [
+        ("metric_control", 1, "msg:ModelMetrics.MetricControl", "r"),
     ],
     "ModelConfig": [
         ("name", 1, "string"),
         ("platform", 2, "string"),
+        ("backend", 17, "string"),
+        ("runtime", 25, "string"),
+        ("version_policy", 3, "msg:ModelVersionPolicy"),
         ("max_batch_size", 4, "int32"),
         ("input", 5, "msg:ModelInput", "r"),
         ("output", 6, "msg:ModelOutput", "r"),
+        ("batch_input", 20, "msg:BatchInput", "r"),
+        ("batch_output", 21, "msg:BatchOutput", "r"),
+        ("optimization", 12, "msg:ModelOptimizationPolicy"),
+        ("dynamic_batching", 11, "msg:ModelDynamicBatching",
+         "oscheduling_choice"),
+        ("sequence_batching", 13, "msg:ModelSequenceBatching",
+         "oscheduling_choice"),
+        ("ensemble_scheduling", 15, "msg:ModelEnsembling",
+         "oscheduling_choice"),
         ("instance_group", 7, "msg:ModelInstanceGroup", "r"),
         ("default_model_filename", 8, "string"),
-        ("dynamic_batching", 11, "msg:ModelDynamicBatching", "oscheduling_choice"),
-        ("sequence_batching", 13, "msg:ModelSequenceBatching", "oscheduling_choice"),
+        ("cc_model_filenames", 9, "map:string:string"),
+        ("metric_tags", 10, "map:string:string"),
         ("parameters", 14, "map:string:msg:ModelParameter"),
-        ("backend", 17, "string"),
+        ("model_warmup", 16, "msg:ModelWarmup", "r"),
+        ("model_operations", 18, "msg:ModelOperations"),
         ("model_transaction_policy", 19, "msg:ModelTransactionPolicy"),
+        ("model_repository_agents", 23, "msg:ModelRepositoryAgents"),
+        ("response_cache", 24, "msg:ModelResponseCache"),
+        ("model_metrics", 26, "msg:ModelMetrics"),
     ],
     "ModelConfigRequest": [("name", 1, "string"), ("version", 2, "string")],
     "ModelConfigResponse": [("config", 1, "msg:ModelConfig")],
@@ -315,6 +513,22 @@ _ENUMS = {
     "ModelInput.Format": ["FORMAT_NONE", "FORMAT_NHWC", "FORMAT_NCHW"],
     "ModelInstanceGroup.Kind": [
         "KIND_AUTO", "KIND_GPU", "KIND_CPU", "KIND_MODEL",
+    ],
+    "ModelInstanceGroup.SecondaryDevice.SecondaryDeviceKind": ["KIND_NVDLA"],
+    "BatchInput.Kind": [
+        "BATCH_ELEMENT_COUNT", "BATCH_ACCUMULATED_ELEMENT_COUNT",
+        "BATCH_ACCUMULATED_ELEMENT_COUNT_WITH_ZERO",
+        "BATCH_MAX_ELEMENT_COUNT_AS_SHAPE", "BATCH_ITEM_SHAPE",
+        "BATCH_ITEM_SHAPE_FLATTEN",
+    ],
+    "BatchOutput.Kind": ["BATCH_SCATTER_WITH_INPUT_SHAPE"],
+    "ModelOptimizationPolicy.ModelPriority": [
+        "PRIORITY_DEFAULT", "PRIORITY_MAX", "PRIORITY_MIN",
+    ],
+    "ModelQueuePolicy.TimeoutAction": ["REJECT", "DELAY"],
+    "ModelSequenceBatching.Control.Kind": [
+        "CONTROL_SEQUENCE_START", "CONTROL_SEQUENCE_READY",
+        "CONTROL_SEQUENCE_END", "CONTROL_SEQUENCE_CORRID",
     ],
 }
 

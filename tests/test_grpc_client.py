@@ -259,3 +259,93 @@ def test_large_tensor_roundtrip(client):
     inp.set_data_from_numpy(x)
     result = client.infer("identity_fp32", [inp])
     np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
+
+
+def test_vendored_proto_in_sync():
+    """proto/grpc_service.proto must match the runtime schema render."""
+    from pathlib import Path
+
+    from client_amd.grpc import _proto_gen
+
+    vendored = (Path(__file__).resolve().parent.parent / "proto" /
+                "grpc_service.proto").read_text()
+    assert vendored == _proto_gen.render(), (
+        "regenerate with: python -m client_amd.grpc._proto_gen")
+
+
+def test_model_config_full_schema_roundtrip():
+    """Full reference model_config.proto surface: every top-level section
+    of ModelConfig (model_config.proto:1971-2180) is settable and
+    survives a serialize/parse roundtrip with correct field numbers."""
+    from client_amd.grpc._proto import service_pb2
+
+    cfg = service_pb2.ModelConfig()
+    cfg.name = "m"
+    cfg.backend = "hip"
+    cfg.runtime = "client_amd"
+    cfg.version_policy.latest.num_versions = 2
+    cfg.max_batch_size = 32
+    inp = cfg.input.add()
+    inp.name = "IN"
+    inp.data_type = 11  # TYPE_FP32
+    inp.dims.extend([3, 224, 224])
+    inp.is_non_linear_format_io = True
+    out = cfg.output.add()
+    out.name = "OUT"
+    out.label_filename = "labels.txt"  # field 4 (reshape is 5)
+    bi = cfg.batch_input.add()
+    bi.kind = 3  # BATCH_MAX_ELEMENT_COUNT_AS_SHAPE
+    bi.target_name.append("ragged_shape")
+    cfg.optimization.cuda.graphs = True
+    spec = cfg.optimization.cuda.graph_spec.add()
+    spec.batch_size = 8
+    spec.input["IN"].dim.extend([3, 224, 224])
+    cfg.optimization.priority = 1  # PRIORITY_MAX
+    cfg.dynamic_batching.preferred_batch_size.extend([4, 8])
+    cfg.dynamic_batching.priority_levels = 3
+    cfg.dynamic_batching.default_queue_policy.timeout_action = 1  # DELAY
+    cfg.dynamic_batching.priority_queue_policy[1].max_queue_size = 16
+    ig = cfg.instance_group.add()
+    ig.kind = 1  # KIND_GPU
+    ig.gpus.extend([0, 1])
+    ig.rate_limiter.resources.add().name = "R1"
+    cfg.cc_model_filenames["gfx950"] = "model.so"
+    cfg.metric_tags["team"] = "serving"
+    cfg.parameters["key"].string_value = "val"
+    w = cfg.model_warmup.add()
+    w.name = "warm"
+    w.inputs["IN"].zero_data = True
+    cfg.model_transaction_policy.decoupled = True
+    cfg.model_repository_agents.agents.add().name = "agent"
+    cfg.response_cache.enable = True
+    mc = cfg.model_metrics.metric_control.add()
+    mc.metric_identifier.family = "latency"
+    mc.histogram_options.buckets.extend([0.1, 1.0])
+
+    data = cfg.SerializeToString()
+    cfg2 = service_pb2.ModelConfig()
+    cfg2.ParseFromString(data)
+    assert cfg2.runtime == "client_amd"
+    assert cfg2.version_policy.WhichOneof("policy_choice") == "latest"
+    assert cfg2.output[0].label_filename == "labels.txt"
+    assert cfg2.batch_input[0].kind == 3
+    assert list(cfg2.optimization.cuda.graph_spec[0].input["IN"].dim) == \
+        [3, 224, 224]
+    assert cfg2.dynamic_batching.priority_queue_policy[1].max_queue_size == 16
+    assert cfg2.instance_group[0].rate_limiter.resources[0].name == "R1"
+    assert cfg2.cc_model_filenames["gfx950"] == "model.so"
+    assert cfg2.model_warmup[0].inputs["IN"].zero_data is True
+    assert cfg2.model_metrics.metric_control[0].histogram_options.buckets[1] \
+        == 1.0
+    assert cfg2.WhichOneof("scheduling_choice") == "dynamic_batching"
+
+    # sequence batching oneof replaces dynamic batching
+    cfg2.sequence_batching.oldest.max_candidate_sequences = 4
+    st = cfg2.sequence_batching.state.add()
+    st.input_name = "S_IN"
+    st.initial_state.add().zero_data = True
+    assert cfg2.WhichOneof("scheduling_choice") == "sequence_batching"
+    cfg3 = service_pb2.ModelConfig()
+    cfg3.ParseFromString(cfg2.SerializeToString())
+    assert cfg3.sequence_batching.oldest.max_candidate_sequences == 4
+    assert cfg3.sequence_batching.state[0].initial_state[0].zero_data is True
