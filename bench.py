@@ -77,7 +77,7 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--concurrency", type=int, default=8)
-    ap.add_argument("--reqs-per-step", type=int, default=16)
+    ap.add_argument("--reqs-per-step", type=int, default=48)
     ap.add_argument("--no-dynamic-batching", action="store_true")
     args = ap.parse_args()
 

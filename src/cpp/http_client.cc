@@ -346,47 +346,62 @@ Error InferenceServerHttpClient::Create(
 //==============================================================================
 // TLS plumbing (sync path). OpenSSL is initialized lazily per client.
 
-Error InferenceServerHttpClient::SslConnect() {
-  if (ssl_ctx_ == nullptr) {
-    static std::once_flag ssl_init;
-    // SIGPIPE: SSL_write lacks MSG_NOSIGNAL; a peer reset would kill
-    // the process without this (libcurl installs the same ignore).
-    std::call_once(ssl_init, [] {
-      SSL_library_init();
-      signal(SIGPIPE, SIG_IGN);
-    });
-    SSL_CTX* ctx = SSL_CTX_new(TLS_client_method());
-    if (ctx == nullptr) return Error("SSL_CTX_new failed");
-    if (!ssl_options_.ca_info.empty()) {
-      if (SSL_CTX_load_verify_locations(ctx, ssl_options_.ca_info.c_str(),
-                                        nullptr) != 1) {
-        SSL_CTX_free(ctx);
-        return Error("failed to load CA bundle " + ssl_options_.ca_info);
-      }
-    } else {
-      SSL_CTX_set_default_verify_paths(ctx);
+Error InferenceServerHttpClient::EnsureSslCtx() {
+  if (ssl_ctx_ != nullptr) return Error::Success;
+  static std::once_flag ssl_init;
+  // SIGPIPE: SSL_write lacks MSG_NOSIGNAL; a peer reset would kill
+  // the process without this (libcurl installs the same ignore).
+  std::call_once(ssl_init, [] {
+    SSL_library_init();
+    signal(SIGPIPE, SIG_IGN);
+  });
+  SSL_CTX* ctx = SSL_CTX_new(TLS_client_method());
+  if (ctx == nullptr) return Error("SSL_CTX_new failed");
+  if (!ssl_options_.ca_info.empty()) {
+    if (SSL_CTX_load_verify_locations(ctx, ssl_options_.ca_info.c_str(),
+                                      nullptr) != 1) {
+      SSL_CTX_free(ctx);
+      return Error("failed to load CA bundle " + ssl_options_.ca_info);
     }
-    if (!ssl_options_.cert.empty()) {
-      if (SSL_CTX_use_certificate_file(ctx, ssl_options_.cert.c_str(),
-                                       SSL_FILETYPE_PEM) != 1 ||
-          SSL_CTX_use_PrivateKey_file(ctx, ssl_options_.key.c_str(),
-                                      SSL_FILETYPE_PEM) != 1) {
-        SSL_CTX_free(ctx);
-        return Error("failed to load client cert/key");
-      }
-    }
-    SSL_CTX_set_verify(
-        ctx, ssl_options_.verify_peer ? SSL_VERIFY_PEER : SSL_VERIFY_NONE,
-        nullptr);
-    ssl_ctx_ = ctx;
+  } else {
+    SSL_CTX_set_default_verify_paths(ctx);
   }
+  if (!ssl_options_.cert.empty()) {
+    if (SSL_CTX_use_certificate_file(ctx, ssl_options_.cert.c_str(),
+                                     SSL_FILETYPE_PEM) != 1 ||
+        SSL_CTX_use_PrivateKey_file(ctx, ssl_options_.key.c_str(),
+                                    SSL_FILETYPE_PEM) != 1) {
+      SSL_CTX_free(ctx);
+      return Error("failed to load client cert/key");
+    }
+  }
+  SSL_CTX_set_verify(
+      ctx, ssl_options_.verify_peer ? SSL_VERIFY_PEER : SSL_VERIFY_NONE,
+      nullptr);
+  // async transfers retry SSL_write after partial progress
+  SSL_CTX_set_mode(ctx, SSL_MODE_ENABLE_PARTIAL_WRITE |
+                            SSL_MODE_ACCEPT_MOVING_WRITE_BUFFER);
+  ssl_ctx_ = ctx;
+  return Error::Success;
+}
+
+Error InferenceServerHttpClient::NewSsl(int fd, void** ssl_out) {
+  RETURN_IF_ERROR(EnsureSslCtx());
   SSL* ssl = SSL_new((SSL_CTX*)ssl_ctx_);
   if (ssl == nullptr) return Error("SSL_new failed");
-  SSL_set_fd(ssl, sync_fd_);
+  SSL_set_fd(ssl, fd);
   SSL_set_tlsext_host_name(ssl, host_.c_str());
   if (ssl_options_.verify_host) {
     SSL_set1_host(ssl, host_.c_str());
   }
+  *ssl_out = ssl;
+  return Error::Success;
+}
+
+Error InferenceServerHttpClient::SslConnect() {
+  void* ssl_v = nullptr;
+  RETURN_IF_ERROR(NewSsl(sync_fd_, &ssl_v));
+  SSL* ssl = (SSL*)ssl_v;
   if (SSL_connect(ssl) != 1) {
     unsigned long err = ERR_get_error();
     char buf[256];
@@ -1057,6 +1072,8 @@ Error InferenceServerHttpClient::InferMulti(
 
 struct InferenceServerHttpClient::AsyncTransfer {
   int fd = -1;
+  SSL* ssl = nullptr;      // per-transfer TLS (worker thread only)
+  bool tls_ready = false;  // handshake finished
   std::string out;     // full request bytes
   size_t out_pos = 0;
   std::string in;      // accumulated response
@@ -1176,6 +1193,10 @@ void InferenceServerHttpClient::AsyncWorker() {
 
   auto finish = [&](std::unique_ptr<AsyncTransfer> t, int code,
                     const std::string& error_msg) {
+    if (t->ssl != nullptr) {
+      SSL_free(t->ssl);
+      t->ssl = nullptr;
+    }
     if (t->fd >= 0) {
       epoll_ctl(epfd, EPOLL_CTL_DEL, t->fd, nullptr);
       close(t->fd);
@@ -1213,6 +1234,15 @@ void InferenceServerHttpClient::AsyncWorker() {
           finish(std::move(t), 0, "failed to connect");
           continue;
         }
+        if (use_ssl_) {
+          void* ssl_v = nullptr;
+          Error serr = NewSsl(t->fd, &ssl_v);
+          if (!serr.IsOk()) {
+            finish(std::move(t), 0, serr.Message());
+            continue;
+          }
+          t->ssl = (SSL*)ssl_v;
+        }
         t->timer.CaptureTimestamp(RequestTimers::Kind::SEND_START);
         struct epoll_event tev;
         tev.events = EPOLLOUT | EPOLLIN;
@@ -1245,10 +1275,59 @@ void InferenceServerHttpClient::AsyncWorker() {
         finish(std::move(owned), 0, "connection error");
         continue;
       }
+      auto set_events = [&](uint32_t mask) {
+        struct epoll_event tev;
+        tev.events = mask;
+        tev.data.fd = fd;
+        epoll_ctl(epfd, EPOLL_CTL_MOD, fd, &tev);
+      };
+      // Non-blocking TLS handshake woven into the event machine: the
+      // first SSL_connect also absorbs TCP connect-in-progress (write
+      // returns EAGAIN until the connect completes -> WANT_WRITE).
+      if (t->ssl != nullptr && !t->tls_ready) {
+        int hr = SSL_connect(t->ssl);
+        if (hr == 1) {
+          t->tls_ready = true;
+          set_events(EPOLLOUT | EPOLLIN);
+        } else {
+          int e = SSL_get_error(t->ssl, hr);
+          if (e == SSL_ERROR_WANT_READ) {
+            set_events(EPOLLIN);  // avoid level-triggered EPOLLOUT spin
+            continue;
+          } else if (e == SSL_ERROR_WANT_WRITE) {
+            set_events(EPOLLOUT | EPOLLIN);
+            continue;
+          } else {
+            unsigned long ee = ERR_get_error();
+            char ebuf[256];
+            ERR_error_string_n(ee, ebuf, sizeof(ebuf));
+            auto owned = std::move(it->second);
+            active.erase(it);
+            finish(std::move(owned), 0,
+                   std::string("TLS handshake failed: ") + ebuf);
+            continue;
+          }
+        }
+      }
       if ((events[i].events & EPOLLOUT) && t->out_pos < t->out.size()) {
         while (t->out_pos < t->out.size()) {
-          ssize_t sent = send(fd, t->out.data() + t->out_pos,
-                              t->out.size() - t->out_pos, MSG_NOSIGNAL);
+          ssize_t sent;
+          if (t->ssl != nullptr) {
+            int r = SSL_write(t->ssl, t->out.data() + t->out_pos,
+                              (int)(t->out.size() - t->out_pos));
+            if (r > 0) {
+              sent = r;
+            } else {
+              int e = SSL_get_error(t->ssl, r);
+              sent = -1;
+              errno = (e == SSL_ERROR_WANT_WRITE || e == SSL_ERROR_WANT_READ)
+                          ? EAGAIN
+                          : EPIPE;
+            }
+          } else {
+            sent = send(fd, t->out.data() + t->out_pos,
+                        t->out.size() - t->out_pos, MSG_NOSIGNAL);
+          }
           if (sent > 0) {
             t->out_pos += (size_t)sent;
           } else if (sent < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
@@ -1259,16 +1338,29 @@ void InferenceServerHttpClient::AsyncWorker() {
         }
         if (t->out_pos >= t->out.size()) {
           t->timer.CaptureTimestamp(RequestTimers::Kind::SEND_END);
-          struct epoll_event tev;
-          tev.events = EPOLLIN;
-          tev.data.fd = fd;
-          epoll_ctl(epfd, EPOLL_CTL_MOD, fd, &tev);
+          set_events(EPOLLIN);
         }
       }
       if (events[i].events & EPOLLIN) {
         bool closed = false;
         while (true) {
-          ssize_t r = recv(fd, chunk, sizeof(chunk), 0);
+          ssize_t r;
+          if (t->ssl != nullptr) {
+            int sr = SSL_read(t->ssl, chunk, sizeof(chunk));
+            if (sr > 0) {
+              r = sr;
+            } else {
+              int e = SSL_get_error(t->ssl, sr);
+              if (e == SSL_ERROR_WANT_READ || e == SSL_ERROR_WANT_WRITE) {
+                r = -1;
+                errno = EAGAIN;
+              } else {
+                r = 0;  // clean close or fatal -> treat as closed
+              }
+            }
+          } else {
+            r = recv(fd, chunk, sizeof(chunk), 0);
+          }
           if (r > 0) {
             if (t->in.empty())
               t->timer.CaptureTimestamp(RequestTimers::Kind::RECV_START);

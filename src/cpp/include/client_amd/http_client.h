@@ -193,6 +193,8 @@ class InferenceServerHttpClient : public InferenceServerClient {
   HttpSslOptions ssl_options_;
   void* ssl_ctx_ = nullptr;  // SSL_CTX*
   void* ssl_ = nullptr;      // SSL* bound to sync_fd_
+  Error EnsureSslCtx();
+  Error NewSsl(int fd, void** ssl_out);  // SSL* with SNI + host checks
   Error SslConnect();
   void SslClose();
   bool IoSend(const char* data, size_t n);

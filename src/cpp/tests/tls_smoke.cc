@@ -1,7 +1,11 @@
-// HTTPS smoke: health + infer over TLS with peer verification off
-// (self-signed test cert). Usage: tls_smoke <host:port>
+// HTTPS smoke: health + sync infer + async (epoll worker) infers over
+// TLS with peer verification off (self-signed test cert).
+// Usage: tls_smoke <host:port>
+#include <chrono>
+#include <condition_variable>
 #include <iostream>
 #include <memory>
+#include <mutex>
 #include <vector>
 
 #include "client_amd/http_client.h"
@@ -42,6 +46,44 @@ int main(int argc, char** argv) {
     std::cerr << "wrong result over TLS" << std::endl;
     return 1;
   }
-  std::cout << "PASS : https" << std::endl;
+  // async path: several concurrent transfers, each with its own
+  // non-blocking TLS handshake inside the epoll worker
+  {
+    std::mutex mu;
+    std::condition_variable cv;
+    int done = 0;
+    bool all_ok = true;
+    const int kAsync = 6;
+    for (int r = 0; r < kAsync; ++r) {
+      err = client->AsyncInfer(
+          [&](InferResult* res) {
+            std::unique_ptr<InferResult> owned(res);
+            const uint8_t* b;
+            size_t bn;
+            if (!owned->RequestStatus().IsOk() ||
+                !owned->RawData("OUTPUT0", &b, &bn).IsOk() || bn != 64 ||
+                ((const int32_t*)b)[7] != 5) {
+              all_ok = false;
+            }
+            std::lock_guard<std::mutex> lock(mu);
+            done++;
+            cv.notify_all();
+          },
+          options, {input0, input1});
+      if (!err.IsOk()) { std::cerr << err.Message() << std::endl; return 1; }
+    }
+    std::unique_lock<std::mutex> lock(mu);
+    if (!cv.wait_for(lock, std::chrono::seconds(20),
+                     [&] { return done == kAsync; })) {
+      std::cerr << "async TLS transfers timed out (" << done << "/"
+                << kAsync << ")" << std::endl;
+      return 1;
+    }
+    if (!all_ok) {
+      std::cerr << "async TLS result mismatch" << std::endl;
+      return 1;
+    }
+  }
+  std::cout << "PASS : https sync+async" << std::endl;
   return 0;
 }
