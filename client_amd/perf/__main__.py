@@ -15,8 +15,20 @@ def main(argv=None):
     p.add_argument("-i", "--protocol", default="grpc",
                    choices=["grpc", "http"])
     p.add_argument("-b", "--batch-size", type=int, default=1)
-    p.add_argument("--concurrency-range", default="1:4:1",
-                   help="start:end:step")
+    p.add_argument("--concurrency-range", default=None,
+                   help="start:end:step (closed loop; default 1:4:1)")
+    p.add_argument("--request-rate-range", default=None,
+                   help="start:end:step req/s (open loop; overrides "
+                        "--concurrency-range)")
+    p.add_argument("--request-distribution", default="constant",
+                   choices=["constant", "poisson"])
+    p.add_argument("--max-threads", type=int, default=16,
+                   help="worker threads for request-rate mode")
+    p.add_argument("--input-data", default=None,
+                   help="JSON file with real request tensors "
+                        '({"data": [{name: [...] | {"content","shape"}}]})')
+    p.add_argument("--percentile", type=float, default=None,
+                   help="additionally report this latency percentile")
     p.add_argument("--shared-memory", default="none",
                    choices=["none", "cuda", "hip"])
     p.add_argument("--repack", action="store_true",
@@ -34,11 +46,20 @@ def main(argv=None):
     p.add_argument("-v", "--verbose", action="store_true")
     args = p.parse_args(argv)
 
-    parts = args.concurrency_range.split(":")
-    start = int(parts[0])
-    end = int(parts[1]) if len(parts) > 1 else start
-    step = int(parts[2]) if len(parts) > 2 else 1
-    concurrency_list = list(range(start, end + 1, step))
+    def parse_range(spec, cast=int):
+        parts = spec.split(":")
+        start = cast(parts[0])
+        end = cast(parts[1]) if len(parts) > 1 else start
+        step = cast(parts[2]) if len(parts) > 2 else cast(1)
+        out, v = [], start
+        while v <= end:
+            out.append(v)
+            v += step
+        return out
+
+    rate_list = (parse_range(args.request_rate_range, float)
+                 if args.request_rate_range else None)
+    concurrency_list = parse_range(args.concurrency_range or "1:4:1")
 
     from .analyzer import PerfAnalyzer
 
@@ -55,30 +76,54 @@ def main(argv=None):
         repack=args.repack,
         verbose=args.verbose,
         shapes=shapes,
+        input_data=args.input_data,
+        percentile_q=args.percentile,
     )
-    results = pa.run(
-        concurrency_list,
-        warmup_s=args.warmup,
-        window_s=args.measurement_interval,
-        max_windows=args.max_windows,
-        stability_pct=args.stability_percentage,
-    )
-    for r in results:
-        print(
-            f"Concurrency: {r['concurrency']}, throughput: "
-            f"{r['inferences_per_sec']} infer/sec, latency p99: "
-            f"{r['latency_us']['p99']} usec"
+    if rate_list is not None:
+        results = pa.run_request_rate(
+            rate_list,
+            warmup_s=args.warmup,
+            window_s=args.measurement_interval,
+            max_windows=args.max_windows,
+            stability_pct=args.stability_percentage,
+            distribution=args.request_distribution,
+            max_threads=args.max_threads,
         )
+        for r in results:
+            print(
+                f"Request rate: {r['target_request_rate']}, achieved: "
+                f"{r['request_rate_per_sec']} req/sec, throughput: "
+                f"{r['inferences_per_sec']} infer/sec, latency p99: "
+                f"{r['latency_us']['p99']} usec"
+            )
+    else:
+        results = pa.run(
+            concurrency_list,
+            warmup_s=args.warmup,
+            window_s=args.measurement_interval,
+            max_windows=args.max_windows,
+            stability_pct=args.stability_percentage,
+        )
+        for r in results:
+            print(
+                f"Concurrency: {r['concurrency']}, throughput: "
+                f"{r['inferences_per_sec']} infer/sec, latency p99: "
+                f"{r['latency_us']['p99']} usec"
+            )
     if args.json:
         with open(args.json, "w") as f:
             json.dump(results, f, indent=2)
     if args.csv:
         with open(args.csv, "w") as f:
-            f.write("Concurrency,Inferences/Second,Client Send,"
+            key = ("Concurrency" if rate_list is None
+                   else "Request Rate")
+            f.write(f"{key},Inferences/Second,Client Send,"
                     "p50 latency,p90 latency,p95 latency,p99 latency\n")
             for r in results:
                 lat = r["latency_us"]
-                f.write(f"{r['concurrency']},{r['inferences_per_sec']},0,"
+                first = (r["concurrency"] if rate_list is None
+                         else r["target_request_rate"])
+                f.write(f"{first},{r['inferences_per_sec']},0,"
                         f"{lat['p50']},{lat['p90']},{lat['p95']},"
                         f"{lat['p99']}\n")
 

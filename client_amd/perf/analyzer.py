@@ -117,11 +117,130 @@ class ConcurrencyDriver:
         return throughput, all_lat, total_err, len(window_results)
 
 
+class RequestRateDriver:
+    """Open-loop load (perf_analyzer --request-rate-range): a request
+    schedule at ``rate`` req/s with constant or Poisson gaps, executed
+    by a pool of worker threads that each claim the next schedule slot,
+    sleep until its deadline, then issue. Unlike the closed loop,
+    server slowdown does not throttle the offered load — late issues
+    are counted in ``delayed``."""
+
+    def __init__(self, issue_fn, rate, num_threads, distribution="constant",
+                 seed=0):
+        self._issue_fn = issue_fn
+        self._rate = rate
+        self._num_threads = num_threads
+        self._distribution = distribution
+        self._seed = seed
+        self._recorder = LatencyRecorder()
+        self._stop = threading.Event()
+        self._done = threading.Event()
+        self._lock = threading.Lock()
+        self._active = 0
+        self._next_idx = 0
+        self._start_ns = 0
+        self.delayed = 0
+        # Pre-generated gap schedule (regenerated lazily as needed).
+        self._rng = np.random.default_rng(seed)
+        self._deadlines = []
+
+    def _deadline(self, idx):
+        with self._lock:
+            while len(self._deadlines) <= idx:
+                if self._distribution == "poisson":
+                    gap = float(self._rng.exponential(1.0 / self._rate))
+                else:
+                    gap = 1.0 / self._rate
+                prev = self._deadlines[-1] if self._deadlines else 0.0
+                self._deadlines.append(prev + gap)
+            return self._deadlines[idx]
+
+    def _worker(self, slot):
+        while not self._stop.is_set():
+            with self._lock:
+                idx = self._next_idx
+                self._next_idx += 1
+            deadline_ns = self._start_ns + int(self._deadline(idx) * 1e9)
+            now = time.monotonic_ns()
+            if deadline_ns > now:
+                if self._stop.wait((deadline_ns - now) / 1e9):
+                    break
+            else:
+                with self._lock:
+                    self.delayed += 1
+            start = time.monotonic_ns()
+            try:
+                self._issue_fn(slot)
+                self._recorder.record(start, time.monotonic_ns())
+            except Exception as e:
+                self._recorder.record(start, time.monotonic_ns(), e)
+                time.sleep(0.01)
+        with self._lock:
+            self._active -= 1
+            if self._active == 0:
+                self._done.set()
+
+    def run(self, warmup_s, window_s, max_windows, stability_pct=10.0,
+            min_stable=3):
+        self._start_ns = time.monotonic_ns()
+        self._active = self._num_threads
+        for slot in range(self._num_threads):
+            threading.Thread(target=self._worker, args=(slot,),
+                             daemon=True).start()
+        time.sleep(warmup_s)
+        self._recorder.snapshot_and_reset()
+        window_results = []
+        all_lat = []
+        for _ in range(max_windows):
+            time.sleep(window_s)
+            lat, err = self._recorder.snapshot_and_reset()
+            window_results.append((len(lat) / window_s, lat, err))
+            all_lat.extend(lat)
+            if len(window_results) >= min_stable:
+                recent = [w[0] for w in window_results[-min_stable:]]
+                mean = sum(recent) / len(recent)
+                if mean > 0 and all(
+                    abs(r - mean) / mean * 100 <= stability_pct
+                    for r in recent
+                ):
+                    break
+        self._stop.set()
+        self._done.wait(timeout=60)
+        total_s = window_s * len(window_results)
+        total_req = sum(len(w[1]) for w in window_results)
+        total_err = sum(w[2] for w in window_results)
+        throughput = total_req / total_s if total_s else 0.0
+        all_lat.sort()
+        return throughput, all_lat, total_err, len(window_results)
+
+
+def load_input_data(path):
+    """Parse a perf_analyzer --input-data JSON file: {"data": [ {name:
+    flat_list | {"content": [...], "shape": [...]} , ...}, ... ]}.
+    Returns a list of per-request dicts name -> (array_values, shape|None)."""
+    import json as _json
+
+    with open(path) as f:
+        doc = _json.load(f)
+    entries = []
+    for item in doc.get("data", []):
+        entry = {}
+        for name, val in item.items():
+            if isinstance(val, dict):
+                entry[name] = (val.get("content", []), val.get("shape"))
+            else:
+                entry[name] = (val, None)
+        entries.append(entry)
+    if not entries:
+        raise ValueError(f"no 'data' entries in {path}")
+    return entries
+
+
 class PerfAnalyzer:
     def __init__(self, url, protocol="grpc", model_name="identity_fp32",
                  batch_size=1, shared_memory="none", input_dtype=None,
                  repack=False, verbose=False, device_id=0, shapes=None,
-                 int_range=(0, 127)):
+                 int_range=(0, 127), input_data=None, percentile_q=None):
         self.url = url
         self.protocol = protocol
         self.model_name = model_name
@@ -134,6 +253,12 @@ class PerfAnalyzer:
         # perf_analyzer-style --shape NAME:d1,d2 overrides for dynamic dims
         self.shapes = shapes or {}
         self.int_range = int_range
+        # --input-data file entries (round-robined across slots);
+        # None = synthetic random tensors
+        self.input_data = (load_input_data(input_data)
+                           if isinstance(input_data, str) else input_data)
+        self._input_data_idx = 0
+        self.percentile_q = percentile_q  # extra reported percentile
         self._client = None
         self._slots = []
 
@@ -177,8 +302,19 @@ class PerfAnalyzer:
         outputs = [(n, d, concrete(n, s)) for n, d, s in outputs]
         return inputs, outputs
 
-    def _synth_array(self, datatype, shape):
+    def _synth_array(self, datatype, shape, name=None):
         np_dt = triton_to_np_dtype(datatype)
+        if self.input_data is not None and name is not None:
+            entry = self.input_data[self._input_data_idx % len(self.input_data)]
+            if name in entry:
+                content, dshape = entry[name]
+                if np_dt == np.object_:
+                    arr = np.array(
+                        [c.encode() if isinstance(c, str) else c
+                         for c in content], dtype=np.object_)
+                else:
+                    arr = np.array(content, dtype=np_dt)
+                return arr.reshape(dshape if dshape else shape)
         if np_dt == np.object_:
             return np.array(
                 [b"x" * 8] * int(np.prod(shape)), dtype=np.object_
@@ -195,9 +331,11 @@ class PerfAnalyzer:
         for _ in range(concurrency):
             infer_inputs = []
             for name, datatype, shape in inputs:
-                ii = mod.InferInput(name, shape, datatype)
-                ii.set_data_from_numpy(self._synth_array(datatype, shape))
+                arr = self._synth_array(datatype, shape, name)
+                ii = mod.InferInput(name, list(arr.shape), datatype)
+                ii.set_data_from_numpy(arr)
                 infer_inputs.append(ii)
+            self._input_data_idx += 1
             infer_outputs = [mod.InferRequestedOutput(n) for n, _, _ in outputs]
             slots.append((infer_inputs, infer_outputs, []))
         return slots
@@ -278,86 +416,125 @@ class PerfAnalyzer:
 
     # ---- measurement ----
 
+    def _build_issue(self, client, mod, n_slots):
+        """Build per-slot request objects + the blocking issue(slot) fn."""
+        inputs, outputs = self._model_io(client)
+        if self.shared_memory in ("cuda", "hip"):
+            slots = self._setup_hipshm_slots(
+                client, mod, inputs, outputs, n_slots
+            )
+        else:
+            slots = self._setup_wire_slots(mod, inputs, outputs, n_slots)
+
+        def repack_slot(staged):
+            if self.repack and staged:
+                import client_amd.utils.hip_shared_memory as hs
+
+                for region, data, datatype in staged:
+                    if datatype == "BF16":
+                        hs.set_shared_memory_region_cast(region, data, "BF16")
+                    else:
+                        hs.set_shared_memory_region(region, [data])
+
+        if self.protocol == "grpc":
+            def issue(slot_idx):
+                infer_inputs, infer_outputs, staged = slots[slot_idx]
+                repack_slot(staged)
+                done = threading.Event()
+                box = {}
+
+                def cb(result, error):
+                    box["error"] = error
+                    done.set()
+
+                client.async_infer(
+                    self.model_name, infer_inputs, callback=cb,
+                    outputs=infer_outputs,
+                )
+                if not done.wait(timeout=120):
+                    raise TimeoutError("request timed out")
+                if box["error"] is not None:
+                    raise box["error"]
+        else:
+            def issue(slot_idx):
+                infer_inputs, infer_outputs, staged = slots[slot_idx]
+                repack_slot(staged)
+                client.infer(
+                    self.model_name, infer_inputs, outputs=infer_outputs,
+                )
+        return issue
+
+    def _result_dict(self, throughput, lat, errors, windows, **extra):
+        latency_us = {
+            "avg": int(np.mean(lat) / 1000) if lat else 0,
+            "p50": int(percentile(lat, 50) / 1000),
+            "p90": int(percentile(lat, 90) / 1000),
+            "p95": int(percentile(lat, 95) / 1000),
+            "p99": int(percentile(lat, 99) / 1000),
+        }
+        if self.percentile_q is not None:
+            latency_us[f"p{self.percentile_q:g}"] = int(
+                percentile(lat, self.percentile_q) / 1000
+            )
+        result = {
+            "batch_size": self.batch_size,
+            "request_rate_per_sec": round(throughput, 2),
+            "inferences_per_sec": round(throughput * self.batch_size, 2),
+            "latency_us": latency_us,
+            "errors": errors,
+            "measurement_windows": windows,
+            "shared_memory": self.shared_memory,
+            "protocol": self.protocol,
+        }
+        result.update(extra)
+        return result
+
     def run(self, concurrency_list, warmup_s=1.0, window_s=2.0, max_windows=6,
             stability_pct=10.0):
-        """Sweep concurrency; returns list of result dicts."""
+        """Closed-loop concurrency sweep; returns list of result dicts."""
         results = []
         for concurrency in concurrency_list:
             client, mod = self._make_client(concurrency)
             try:
-                inputs, outputs = self._model_io(client)
-                if self.shared_memory in ("cuda", "hip"):
-                    slots = self._setup_hipshm_slots(
-                        client, mod, inputs, outputs, concurrency
-                    )
-                else:
-                    slots = self._setup_wire_slots(
-                        mod, inputs, outputs, concurrency
-                    )
-
-                def repack_slot(staged):
-                    if self.repack and staged:
-                        import client_amd.utils.hip_shared_memory as hs
-
-                        for region, data, datatype in staged:
-                            if datatype == "BF16":
-                                hs.set_shared_memory_region_cast(
-                                    region, data, "BF16"
-                                )
-                            else:
-                                hs.set_shared_memory_region(region, [data])
-
-                if self.protocol == "grpc":
-                    def issue(slot_idx):
-                        infer_inputs, infer_outputs, staged = slots[slot_idx]
-                        repack_slot(staged)
-                        done = threading.Event()
-                        box = {}
-
-                        def cb(result, error):
-                            box["error"] = error
-                            done.set()
-
-                        client.async_infer(
-                            self.model_name, infer_inputs, callback=cb,
-                            outputs=infer_outputs,
-                        )
-                        if not done.wait(timeout=120):
-                            raise TimeoutError("request timed out")
-                        if box["error"] is not None:
-                            raise box["error"]
-                else:
-                    def issue(slot_idx):
-                        infer_inputs, infer_outputs, staged = slots[slot_idx]
-                        repack_slot(staged)
-                        client.infer(
-                            self.model_name, infer_inputs,
-                            outputs=infer_outputs,
-                        )
-
+                issue = self._build_issue(client, mod, concurrency)
                 driver = ConcurrencyDriver(issue, concurrency)
                 throughput, lat, errors, windows = driver.run(
                     warmup_s, window_s, max_windows, stability_pct
                 )
-                result = {
-                    "concurrency": concurrency,
-                    "batch_size": self.batch_size,
-                    "request_rate_per_sec": round(throughput, 2),
-                    "inferences_per_sec": round(
-                        throughput * self.batch_size, 2
-                    ),
-                    "latency_us": {
-                        "avg": int(np.mean(lat) / 1000) if lat else 0,
-                        "p50": int(percentile(lat, 50) / 1000),
-                        "p90": int(percentile(lat, 90) / 1000),
-                        "p95": int(percentile(lat, 95) / 1000),
-                        "p99": int(percentile(lat, 99) / 1000),
-                    },
-                    "errors": errors,
-                    "measurement_windows": windows,
-                    "shared_memory": self.shared_memory,
-                    "protocol": self.protocol,
-                }
+                result = self._result_dict(
+                    throughput, lat, errors, windows, concurrency=concurrency
+                )
+                results.append(result)
+                if self.verbose:
+                    print(result)
+                if self.shared_memory in ("cuda", "hip"):
+                    self._teardown_hipshm(client)
+            finally:
+                client.close()
+        return results
+
+    def run_request_rate(self, rate_list, warmup_s=1.0, window_s=2.0,
+                         max_windows=6, stability_pct=10.0,
+                         distribution="constant", max_threads=16):
+        """Open-loop request-rate sweep (perf_analyzer
+        --request-rate-range / --request-distribution)."""
+        results = []
+        for rate in rate_list:
+            client, mod = self._make_client(max_threads)
+            try:
+                issue = self._build_issue(client, mod, max_threads)
+                driver = RequestRateDriver(
+                    issue, rate, max_threads, distribution
+                )
+                throughput, lat, errors, windows = driver.run(
+                    warmup_s, window_s, max_windows, stability_pct
+                )
+                result = self._result_dict(
+                    throughput, lat, errors, windows,
+                    target_request_rate=rate,
+                    request_distribution=distribution,
+                    delayed_requests=driver.delayed,
+                )
                 results.append(result)
                 if self.verbose:
                     print(result)
