@@ -452,6 +452,18 @@ Error InferenceServerGrpcClient::ModelMetadata(
   return Error::Success;
 }
 
+Error InferenceServerGrpcClient::ModelConfig(
+    kserve::ModelConfigPb* config, const std::string& model_name,
+    const std::string& model_version) {
+  std::string resp;
+  RETURN_IF_ERROR(UnaryCall(
+      "ModelConfig", kserve::EncodeNameVersion(model_name, model_version),
+      &resp));
+  *config = kserve::ModelConfigPb::Decode(
+      (const uint8_t*)resp.data(), resp.size());
+  return Error::Success;
+}
+
 Error InferenceServerGrpcClient::ModelRepositoryIndex(
     std::vector<kserve::RepositoryIndexEntryPb>* index) {
   std::string resp;

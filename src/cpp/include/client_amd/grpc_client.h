@@ -61,6 +61,9 @@ class InferenceServerGrpcClient : public InferenceServerClient {
   Error ModelMetadata(
       kserve::ModelMetadataPb* metadata, const std::string& model_name,
       const std::string& model_version = "");
+  Error ModelConfig(
+      kserve::ModelConfigPb* config, const std::string& model_name,
+      const std::string& model_version = "");
   Error ModelRepositoryIndex(
       std::vector<kserve::RepositoryIndexEntryPb>* index);
   Error LoadModel(const std::string& model_name);

@@ -124,6 +124,61 @@ struct ModelMetadataPb {
   static ModelMetadataPb Decode(const uint8_t* data, size_t n);
 };
 
+//==============================================================================
+// ModelConfig (model_config.proto:1971-2180). Decoded subset covering
+// the commonly-consumed fields; unrecognized fields are skipped (the
+// full schema lives on the Python side, client_amd/grpc/_proto.py).
+struct ModelTensorConfigPb {
+  std::string name;
+  int32_t data_type = 0;  // DataType enum value
+  std::vector<int64_t> dims;
+  std::string label_filename;    // outputs only (field 4)
+  bool is_shape_tensor = false;
+  bool optional_input = false;   // inputs only (field 8)
+};
+
+struct ModelInstanceGroupPb {
+  std::string name;
+  int32_t kind = 0;  // ModelInstanceGroup.Kind
+  int32_t count = 0;
+  std::vector<int32_t> gpus;
+};
+
+struct ModelDynamicBatchingPb {
+  std::vector<int32_t> preferred_batch_size;
+  uint64_t max_queue_delay_microseconds = 0;
+  bool preserve_ordering = false;
+};
+
+struct EnsembleStepPb {
+  std::string model_name;
+  int64_t model_version = -1;
+  std::map<std::string, std::string> input_map;
+  std::map<std::string, std::string> output_map;
+};
+
+struct ModelConfigPb {
+  std::string name;
+  std::string platform;
+  std::string backend;
+  std::string runtime;
+  std::string default_model_filename;
+  int32_t max_batch_size = 0;
+  std::vector<ModelTensorConfigPb> input;
+  std::vector<ModelTensorConfigPb> output;
+  std::vector<ModelInstanceGroupPb> instance_group;
+  bool has_dynamic_batching = false;   // scheduling_choice oneof
+  bool has_sequence_batching = false;
+  bool has_ensemble_scheduling = false;
+  ModelDynamicBatchingPb dynamic_batching;
+  std::vector<EnsembleStepPb> ensemble_steps;
+  std::map<std::string, std::string> parameters;  // name -> string_value
+  bool decoupled = false;              // model_transaction_policy
+  bool response_cache_enable = false;
+  // Decodes a ModelConfigResponse (config = field 1).
+  static ModelConfigPb Decode(const uint8_t* data, size_t n);
+};
+
 struct RepositoryIndexEntryPb {
   std::string name;
   std::string version;

@@ -381,5 +381,246 @@ std::vector<ModelStatisticsPb> DecodeModelStatistics(const uint8_t* data,
   return out;
 }
 
+//==============================================================================
+// ModelConfig decode (subset; see kserve_pb.h).
+
+namespace {
+
+void DecodePackedOrSingleI32(Reader* r, int wire, std::vector<int32_t>* out) {
+  if (wire == pb::LEN) {
+    auto [ptr, len] = r->bytes();
+    Reader sub(ptr, len);
+    while (!sub.done()) out->push_back((int32_t)sub.varint());
+  } else {
+    out->push_back((int32_t)r->varint());
+  }
+}
+
+void DecodePackedOrSingleI64(Reader* r, int wire, std::vector<int64_t>* out) {
+  if (wire == pb::LEN) {
+    auto [ptr, len] = r->bytes();
+    Reader sub(ptr, len);
+    while (!sub.done()) out->push_back((int64_t)sub.varint());
+  } else {
+    out->push_back((int64_t)r->varint());
+  }
+}
+
+ModelTensorConfigPb DecodeTensorConfig(const uint8_t* data, size_t n,
+                                       bool is_input) {
+  ModelTensorConfigPb t;
+  Reader r(data, n);
+  int field, wire;
+  while (r.next(&field, &wire)) {
+    switch (field) {
+      case 1: t.name = r.str(); break;
+      case 2: t.data_type = (int32_t)r.varint(); break;
+      case 3:
+        // ModelInput.format (varint) vs ModelOutput.dims (packed)
+        if (is_input) r.skip(wire);
+        else DecodePackedOrSingleI64(&r, wire, &t.dims);
+        break;
+      case 4:
+        if (is_input) DecodePackedOrSingleI64(&r, wire, &t.dims);
+        else t.label_filename = r.str();
+        break;
+      case 6: t.is_shape_tensor = r.varint() != 0; break;
+      case 8:
+        if (is_input) t.optional_input = r.varint() != 0;
+        else r.skip(wire);
+        break;
+      default: r.skip(wire);
+    }
+  }
+  return t;
+}
+
+ModelInstanceGroupPb DecodeInstanceGroup(const uint8_t* data, size_t n) {
+  ModelInstanceGroupPb g;
+  Reader r(data, n);
+  int field, wire;
+  while (r.next(&field, &wire)) {
+    switch (field) {
+      case 1: g.name = r.str(); break;
+      case 2: g.count = (int32_t)r.varint(); break;
+      case 3: DecodePackedOrSingleI32(&r, wire, &g.gpus); break;
+      case 4: g.kind = (int32_t)r.varint(); break;
+      default: r.skip(wire);
+    }
+  }
+  return g;
+}
+
+ModelDynamicBatchingPb DecodeDynBatch(const uint8_t* data, size_t n) {
+  ModelDynamicBatchingPb d;
+  Reader r(data, n);
+  int field, wire;
+  while (r.next(&field, &wire)) {
+    switch (field) {
+      case 1: DecodePackedOrSingleI32(&r, wire, &d.preferred_batch_size); break;
+      case 2: d.max_queue_delay_microseconds = r.varint(); break;
+      case 3: d.preserve_ordering = r.varint() != 0; break;
+      default: r.skip(wire);
+    }
+  }
+  return d;
+}
+
+void DecodeStrMapEntry(const uint8_t* data, size_t n,
+                       std::map<std::string, std::string>* out) {
+  Reader r(data, n);
+  int field, wire;
+  std::string k, v;
+  while (r.next(&field, &wire)) {
+    if (field == 1) k = r.str();
+    else if (field == 2) v = r.str();
+    else r.skip(wire);
+  }
+  (*out)[k] = v;
+}
+
+EnsembleStepPb DecodeEnsembleStep(const uint8_t* data, size_t n) {
+  EnsembleStepPb e;
+  Reader r(data, n);
+  int field, wire;
+  while (r.next(&field, &wire)) {
+    switch (field) {
+      case 1: e.model_name = r.str(); break;
+      case 2: e.model_version = (int64_t)r.varint(); break;
+      case 3: {
+        auto [ptr, len] = r.bytes();
+        DecodeStrMapEntry(ptr, len, &e.input_map);
+        break;
+      }
+      case 4: {
+        auto [ptr, len] = r.bytes();
+        DecodeStrMapEntry(ptr, len, &e.output_map);
+        break;
+      }
+      default: r.skip(wire);
+    }
+  }
+  return e;
+}
+
+void DecodeModelParamEntry(const uint8_t* data, size_t n,
+                           std::map<std::string, std::string>* out) {
+  Reader r(data, n);
+  int field, wire;
+  std::string k, v;
+  while (r.next(&field, &wire)) {
+    if (field == 1) {
+      k = r.str();
+    } else if (field == 2 && wire == pb::LEN) {
+      auto [ptr, len] = r.bytes();
+      Reader sub(ptr, len);
+      int f2, w2;
+      while (sub.next(&f2, &w2)) {
+        if (f2 == 1) v = sub.str();
+        else sub.skip(w2);
+      }
+    } else {
+      r.skip(wire);
+    }
+  }
+  (*out)[k] = v;
+}
+
+ModelConfigPb DecodeModelConfigMsg(const uint8_t* data, size_t n) {
+  ModelConfigPb c;
+  Reader r(data, n);
+  int field, wire;
+  while (r.next(&field, &wire)) {
+    switch (field) {
+      case 1: c.name = r.str(); break;
+      case 2: c.platform = r.str(); break;
+      case 4: c.max_batch_size = (int32_t)r.varint(); break;
+      case 5: {
+        auto [ptr, len] = r.bytes();
+        c.input.push_back(DecodeTensorConfig(ptr, len, true));
+        break;
+      }
+      case 6: {
+        auto [ptr, len] = r.bytes();
+        c.output.push_back(DecodeTensorConfig(ptr, len, false));
+        break;
+      }
+      case 7: {
+        auto [ptr, len] = r.bytes();
+        c.instance_group.push_back(DecodeInstanceGroup(ptr, len));
+        break;
+      }
+      case 8: c.default_model_filename = r.str(); break;
+      case 11: {
+        auto [ptr, len] = r.bytes();
+        c.has_dynamic_batching = true;
+        c.dynamic_batching = DecodeDynBatch(ptr, len);
+        break;
+      }
+      case 13: c.has_sequence_batching = true; r.skip(wire); break;
+      case 14: {
+        auto [ptr, len] = r.bytes();
+        DecodeModelParamEntry(ptr, len, &c.parameters);
+        break;
+      }
+      case 15: {
+        auto [ptr, len] = r.bytes();
+        c.has_ensemble_scheduling = true;
+        Reader sub(ptr, len);
+        int f2, w2;
+        while (sub.next(&f2, &w2)) {
+          if (f2 == 1 && w2 == pb::LEN) {
+            auto [p2, l2] = sub.bytes();
+            c.ensemble_steps.push_back(DecodeEnsembleStep(p2, l2));
+          } else {
+            sub.skip(w2);
+          }
+        }
+        break;
+      }
+      case 17: c.backend = r.str(); break;
+      case 19: {
+        auto [ptr, len] = r.bytes();
+        Reader sub(ptr, len);
+        int f2, w2;
+        while (sub.next(&f2, &w2)) {
+          if (f2 == 1) c.decoupled = sub.varint() != 0;
+          else sub.skip(w2);
+        }
+        break;
+      }
+      case 24: {
+        auto [ptr, len] = r.bytes();
+        Reader sub(ptr, len);
+        int f2, w2;
+        while (sub.next(&f2, &w2)) {
+          if (f2 == 1) c.response_cache_enable = sub.varint() != 0;
+          else sub.skip(w2);
+        }
+        break;
+      }
+      case 25: c.runtime = r.str(); break;
+      default: r.skip(wire);
+    }
+  }
+  return c;
+}
+
+}  // namespace
+
+ModelConfigPb ModelConfigPb::Decode(const uint8_t* data, size_t n) {
+  // ModelConfigResponse { ModelConfig config = 1; }
+  Reader r(data, n);
+  int field, wire;
+  while (r.next(&field, &wire)) {
+    if (field == 1 && wire == pb::LEN) {
+      auto [ptr, len] = r.bytes();
+      return DecodeModelConfigMsg(ptr, len);
+    }
+    r.skip(wire);
+  }
+  return ModelConfigPb();
+}
+
 }  // namespace kserve
 }  // namespace client_amd

@@ -59,6 +59,13 @@ int main(int argc, char** argv) {
   CHECK(model_meta.inputs.size() == 2);
   CHECK(model_meta.inputs[0].datatype == "INT32");
 
+  // ---- model config ----
+  kserve::ModelConfigPb cfg;
+  CHECK_OK(client->ModelConfig(&cfg, "simple"));
+  CHECK(cfg.name == "simple");
+  CHECK(cfg.input.size() == 2 && cfg.output.size() == 2);
+  CHECK(cfg.input[0].dims.size() == 2);
+
   // ---- repository ----
   std::vector<kserve::RepositoryIndexEntryPb> index;
   CHECK_OK(client->ModelRepositoryIndex(&index));
