@@ -21,17 +21,27 @@ using namespace client_amd;
     }                                                       \
   } while (0)
 
+static std::string ReadFile(const char* path) {
+  std::ifstream f(path);
+  std::stringstream ss;
+  ss << f.rdbuf();
+  return ss.str();
+}
+
 int main(int argc, char** argv) {
-  if (argc != 4) {
-    fprintf(stderr, "usage: %s host port root_cert.pem\n", argv[0]);
+  if (argc != 4 && argc != 6) {
+    fprintf(stderr,
+            "usage: %s host port root_cert.pem [client_key.pem "
+            "client_cert.pem]\n",
+            argv[0]);
     return 2;
   }
-  std::ifstream f(argv[3]);
-  std::stringstream pem;
-  pem << f.rdbuf();
-
   SslOptions ssl;
-  ssl.root_certificates = pem.str();
+  ssl.root_certificates = ReadFile(argv[3]);
+  if (argc == 6) {  // mTLS
+    ssl.private_key = ReadFile(argv[4]);
+    ssl.certificate_chain = ReadFile(argv[5]);
+  }
   std::unique_ptr<InferenceServerGrpcClient> client;
   CHECK_OK(InferenceServerGrpcClient::Create(
       &client, std::string(argv[1]) + ":" + argv[2], false, true, ssl));

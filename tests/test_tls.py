@@ -146,3 +146,72 @@ def test_cpp_grpc_tls(grpc_tls_server, tls_cert):
     )
     assert proc.returncode == 0, proc.stdout + proc.stderr
     assert "PASS" in proc.stdout
+
+
+@pytest.fixture(scope="module")
+def grpc_mtls_server(tls_cert):
+    """Secure port that REQUIRES a client certificate (mTLS); the
+    self-signed cert doubles as the client identity and trust root."""
+    import grpc
+
+    from client_amd.server.grpc_server import GrpcServer
+
+    cert, key = tls_cert
+    creds = grpc.ssl_server_credentials(
+        [(key.read_bytes(), cert.read_bytes())],
+        root_certificates=cert.read_bytes(),
+        require_client_auth=True,
+    )
+    server = GrpcServer(
+        _addsub_core(), host="127.0.0.1", port=0, ssl_credentials=creds
+    )
+    server.start()
+    yield "127.0.0.1", server.port
+    server.stop()
+
+
+def test_python_grpc_mtls(grpc_mtls_server, tls_cert):
+    import client_amd.grpc as grpcclient
+
+    host, port = grpc_mtls_server
+    cert, key = tls_cert
+    client = grpcclient.InferenceServerClient(
+        f"{host}:{port}",
+        ssl=True,
+        root_certificates=str(cert),
+        private_key=str(key),
+        certificate_chain=str(cert),
+        channel_args=[("grpc.ssl_target_name_override", "localhost")],
+    )
+    try:
+        assert client.is_server_live()
+    finally:
+        client.close()
+
+
+def test_cpp_grpc_mtls(grpc_mtls_server, tls_cert):
+    from tests.test_cpp_client import _compile, CPP
+
+    host, port = grpc_mtls_server
+    cert, key = tls_cert
+    binary = _compile("grpc_tls_smoke", CPP / "tests" / "grpc_tls_smoke.cc")
+    proc = subprocess.run(
+        [str(binary), host, str(port), str(cert), str(key), str(cert)],
+        capture_output=True, text=True, timeout=60,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "PASS" in proc.stdout
+
+
+def test_cpp_grpc_tls_rejects_without_client_cert(grpc_mtls_server, tls_cert):
+    """The mTLS server must refuse a client that presents no cert."""
+    from tests.test_cpp_client import _compile, CPP
+
+    host, port = grpc_mtls_server
+    cert, _ = tls_cert
+    binary = _compile("grpc_tls_smoke", CPP / "tests" / "grpc_tls_smoke.cc")
+    proc = subprocess.run(
+        [str(binary), host, str(port), str(cert)],
+        capture_output=True, text=True, timeout=60,
+    )
+    assert proc.returncode != 0
