@@ -215,3 +215,19 @@ def test_cpp_grpc_tls_rejects_without_client_cert(grpc_mtls_server, tls_cert):
         capture_output=True, text=True, timeout=60,
     )
     assert proc.returncode != 0
+
+
+def test_cpp_grpc_ssl_example(grpc_tls_server, tls_cert):
+    from tests.test_cpp_client import _compile, CPP
+
+    host, port = grpc_tls_server
+    cert, _ = tls_cert
+    binary = _compile("simple_grpc_ssl_infer_client",
+                      CPP / "examples" / "simple_grpc_ssl_infer_client.cc")
+    proc = subprocess.run(
+        [str(binary), "-u", f"{host}:{port}", "-ssl",
+         "--root-certificates", str(cert)],
+        capture_output=True, text=True, timeout=60,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "PASS" in proc.stdout
