@@ -28,6 +28,7 @@ def http_fixture_server():
     core.add_model(IdentityModel("identity_fp32", "FP32"))
     core.add_model(IdentityModel("identity_bf16", "BF16"))
     core.add_model(IdentityModel("identity_bytes", "BYTES"))
+    core.add_model(IdentityModel("identity_int8", "INT8"))
     core.add_model(AddSubModel("simple", "INT32", (-1, 16)))
     core.add_model(AddSubModel("simple_string", "BYTES", (-1, 16)))
     core.add_model(SequenceModel())
@@ -51,6 +52,7 @@ def _make_fixture_core():
     core.add_model(IdentityModel("identity_fp32", "FP32"))
     core.add_model(IdentityModel("identity_bf16", "BF16"))
     core.add_model(IdentityModel("identity_bytes", "BYTES"))
+    core.add_model(IdentityModel("identity_int8", "INT8"))
     core.add_model(AddSubModel("simple", "INT32", (-1, 16)))
     core.add_model(AddSubModel("simple_string", "BYTES", (-1, 16)))
     core.add_model(SequenceModel())

@@ -19,6 +19,8 @@ HTTP_EXAMPLES = [
     "simple_http_health_metadata.py",
     "simple_http_model_control.py",
     "simple_http_shm_client.py",
+    "simple_http_shm_string_client.py",
+    "simple_http_sequence_sync_infer_client.py",
     "simple_http_aio_infer_client.py",
     "reuse_infer_objects_client.py",
 ]
@@ -33,6 +35,15 @@ GRPC_EXAMPLES = [
     "simple_grpc_keepalive_client.py",
     "simple_grpc_custom_args_client.py",
     "simple_grpc_aio_infer_client.py",
+    "simple_grpc_string_infer_client.py",
+    "simple_grpc_shm_client.py",
+    "simple_grpc_shm_string_client.py",
+    "simple_grpc_sequence_sync_infer_client.py",
+    "simple_grpc_aio_sequence_stream_infer_client.py",
+    "grpc_client.py",
+    "grpc_explicit_int_content_client.py",
+    "grpc_explicit_int8_content_client.py",
+    "grpc_explicit_byte_content_client.py",
 ]
 
 
