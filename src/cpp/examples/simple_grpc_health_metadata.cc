@@ -1,0 +1,46 @@
+// Server health + metadata over gRPC
+// (reference: src/c++/examples/simple_grpc_health_metadata.cc).
+#include "client_amd/grpc_client.h"
+#include <iostream>
+#include <memory>
+#include <vector>
+
+namespace ca = client_amd;
+
+#define FAIL_IF_ERR(X, MSG)                                      \
+  {                                                              \
+    ca::Error err = (X);                                         \
+    if (!err.IsOk()) {                                           \
+      std::cerr << "error: " << (MSG) << ": " << err.Message()   \
+                << std::endl;                                    \
+      exit(1);                                                   \
+    }                                                            \
+  }
+
+int main(int argc, char** argv) {
+  std::string url = "127.0.0.1:8001";
+  for (int i = 1; i < argc - 1; ++i)
+    if (std::string(argv[i]) == "-u") url = argv[i + 1];
+
+  std::unique_ptr<ca::InferenceServerGrpcClient> client;
+  FAIL_IF_ERR(ca::InferenceServerGrpcClient::Create(&client, url), "create");
+  bool live = false, ready = false, model_ready = false;
+  FAIL_IF_ERR(client->IsServerLive(&live), "live");
+  FAIL_IF_ERR(client->IsServerReady(&ready), "ready");
+  FAIL_IF_ERR(client->IsModelReady(&model_ready, "simple"), "model ready");
+  if (!live || !ready || !model_ready) {
+    std::cerr << "server/model not ready" << std::endl;
+    return 1;
+  }
+  ca::kserve::ServerMetadataPb meta;
+  FAIL_IF_ERR(client->ServerMetadata(&meta), "server metadata");
+  std::cout << "server: " << meta.name << " " << meta.version << std::endl;
+  ca::kserve::ModelMetadataPb mmeta;
+  FAIL_IF_ERR(client->ModelMetadata(&mmeta, "simple"), "model metadata");
+  std::cout << "model: " << mmeta.name << " inputs=" << mmeta.inputs.size()
+            << std::endl;
+  ca::kserve::ModelConfigPb cfg;
+  FAIL_IF_ERR(client->ModelConfig(&cfg, "simple"), "model config");
+  std::cout << "PASS : health+metadata" << std::endl;
+  return 0;
+}

@@ -156,8 +156,35 @@ HTTP_CC_EXAMPLES = [
     "simple_http_async_infer_client.cc",
     "simple_http_string_infer_client.cc",
     "simple_http_shm_client.cc",
+    "simple_http_health_metadata.cc",
+    "simple_http_model_control.cc",
+    "simple_http_sequence_sync_infer_client.cc",
     "reuse_infer_objects_client.cc",
 ]
+
+GRPC_CC_EXAMPLES = [
+    "simple_grpc_health_metadata.cc",
+    "simple_grpc_model_control.cc",
+    "simple_grpc_string_infer_client.cc",
+    "simple_grpc_async_infer_client.cc",
+    "simple_grpc_sequence_sync_infer_client.cc",
+    "simple_grpc_shm_client.cc",
+    "simple_grpc_custom_args_client.cc",
+    "simple_grpc_custom_repeat.cc",
+    "simple_grpc_keepalive_client.cc",
+]
+
+
+@pytest.mark.parametrize("src", GRPC_CC_EXAMPLES)
+def test_cc_grpc_examples(src, grpc_fixture_server):
+    host, port, _ = grpc_fixture_server
+    binary = _compile(src[:-3], CPP / "examples" / src)
+    proc = subprocess.run(
+        [str(binary), "-u", f"{host}:{port}"], capture_output=True, text=True,
+        timeout=60,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "PASS" in proc.stdout
 
 
 @pytest.mark.parametrize("src", HTTP_CC_EXAMPLES)
@@ -279,6 +306,55 @@ def test_cc_hipshm_example_gpu():
         assert ready, "server not ready"
         proc = subprocess.run(
             [str(binary), "-u", "127.0.0.1:18511"], capture_output=True,
+            text=True, timeout=120,
+        )
+        assert proc.returncode == 0, proc.stdout + proc.stderr
+        assert "PASS" in proc.stdout
+    finally:
+        server.terminate()
+        server.wait(timeout=10)
+
+
+@pytest.mark.gpu
+def test_cc_grpc_hipshm_example_gpu():
+    """gRPC HIP-IPC example (raw 64-byte handle in the proto) compiled
+    with hipcc against an out-of-process server — the gRPC analog of
+    the cudashm example round trip (GPU)."""
+    import time
+
+    from client_amd.ops import gpu_available
+
+    if not gpu_available():
+        pytest.skip("no HIP device")
+    BUILD.mkdir(exist_ok=True)
+    binary = BUILD / "simple_grpc_hipshm_client"
+    srcs = [CPP / s for s in ("common.cc", "json.cc", "shm_utils.cc",
+                              "http_client.cc", "h2.cc", "kserve_pb.cc",
+                              "grpc_client.cc", "hip_shm.cc")]
+    srcs.append(CPP / "examples" / "simple_grpc_hipshm_client.cc")
+    subprocess.run(
+        ["/opt/rocm/bin/hipcc", "--offload-arch=gfx950", "-std=c++17", "-O1",
+         f"-I{CPP}/include", "-DTRITON_ENABLE_HIP", "-x", "hip",
+         *map(str, srcs), "-o", str(binary), "-lpthread", "-lrt", "-lz",
+         "-lssl", "-lcrypto", "-l:libnghttp2.so.14"],
+        check=True, capture_output=True, text=True)
+    server = subprocess.Popen(
+        [sys.executable, "-m", "client_amd.server", "--grpc-port", "18512",
+         "--models", "simple"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        cwd=REPO,
+    )
+    try:
+        deadline = time.time() + 60
+        ready = False
+        while time.time() < deadline:
+            line = server.stdout.readline()
+            if line.startswith("GRPC_READY"):
+                ready = True
+                break
+        assert ready, "server not ready"
+        proc = subprocess.run(
+            [str(binary), "-u", "127.0.0.1:18512"], capture_output=True,
             text=True, timeout=120,
         )
         assert proc.returncode == 0, proc.stdout + proc.stderr
