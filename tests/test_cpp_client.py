@@ -362,3 +362,11 @@ def test_cc_grpc_hipshm_example_gpu():
     finally:
         server.terminate()
         server.wait(timeout=10)
+
+
+def test_cc_image_clients_compile():
+    """image_client / ensemble_image_client need a GPU-backed resnet50
+    to run; on CPU we verify they build and link."""
+    _compile("image_client", CPP / "examples" / "image_client.cc")
+    _compile("ensemble_image_client",
+             CPP / "examples" / "ensemble_image_client.cc")
