@@ -42,6 +42,11 @@ def _shm_path(shm_key):
 def create_shared_memory_region(triton_shm_name, shm_key, byte_size,
                                 create_only=False):
     """Create (or open) a POSIX shm region; returns the region handle."""
+    if byte_size <= 0:
+        raise SharedMemoryException(
+            "unable to create the shared memory region: byte_size must be "
+            "positive"
+        )
     shm_handle = SharedMemoryRegion(triton_shm_name, shm_key)
     path = _shm_path(shm_key)
     exists = os.path.exists(path)
@@ -92,6 +97,11 @@ def set_shared_memory_region(shm_handle, input_values, offset=0):
             byte_data = serialize_byte_tensor(input_value).item()
         else:
             byte_data = input_value.tobytes()
+        if offset_current + len(byte_data) > shm_handle._byte_size:
+            raise SharedMemoryException(
+                "unable to set the shared memory region: tensors exceed "
+                f"region size ({shm_handle._byte_size} bytes)"
+            )
         mem[offset_current : offset_current + len(byte_data)] = byte_data
         offset_current += len(byte_data)
 

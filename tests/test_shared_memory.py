@@ -114,3 +114,35 @@ def test_shm_inference_roundtrip(protocol, http_fixture_server,
     finally:
         shm.destroy_shared_memory_region(handle)
         client.close()
+
+
+def test_set_region_oversize_raises():
+    """Writing past the region boundary is a SharedMemoryException, not a
+    raw mmap IndexError (reference tests/test_shared_memory.py:92)."""
+    import numpy as np
+
+    import client_amd.utils.shared_memory as shm
+    from client_amd.utils.shared_memory import SharedMemoryException
+
+    handle = shm.create_shared_memory_region("over", "/shm_over_test", 32)
+    try:
+        big = np.zeros(64, dtype=np.int32)
+        with pytest.raises(SharedMemoryException):
+            shm.set_shared_memory_region(handle, [big])
+        with pytest.raises(SharedMemoryException):
+            shm.set_shared_memory_region(
+                handle, [np.zeros(4, dtype=np.int32)], offset=24)
+    finally:
+        shm.destroy_shared_memory_region(handle)
+
+
+def test_invalid_create_raises():
+    """Zero/negative byte sizes are rejected up front (reference
+    tests/test_shared_memory.py:65)."""
+    import client_amd.utils.shared_memory as shm
+    from client_amd.utils.shared_memory import SharedMemoryException
+
+    with pytest.raises(SharedMemoryException):
+        shm.create_shared_memory_region("bad", "/shm_bad_test", 0)
+    with pytest.raises(SharedMemoryException):
+        shm.create_shared_memory_region("bad", "/shm_bad_test", -4)
