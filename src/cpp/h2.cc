@@ -277,11 +277,14 @@ void H2Connection::Close() {
     exiting_ = true;
   }
   alive_ = false;
-  ka_cv_.notify_all();
-  if (keepalive_.joinable()) keepalive_.join();
+  // shutdown() BEFORE joining: the keepalive thread may be blocked in a
+  // socket send (WriteFrame) and the reader in recv — both need the fd
+  // torn down to return.
   if (fd_ >= 0) {
     shutdown(fd_, SHUT_RDWR);
   }
+  ka_cv_.notify_all();
+  if (keepalive_.joinable()) keepalive_.join();
   if (reader_.joinable()) reader_.join();
   if (ssl_ != nullptr) {
     SSL_free((SSL*)ssl_);
