@@ -114,12 +114,13 @@ def main():
                                      server_args)
     try:
         client = grpcclient.InferenceServerClient(f"127.0.0.1:{port}")
-        for _ in range(120):
+        for _ in range(240):
             try:
                 if client.is_server_ready():
                     break
             except Exception:
-                time.sleep(0.25)
+                pass
+            time.sleep(0.25)
 
         # per-slot region pairs (bf16 in HBM3E)
         in_elems = int(np.prod(IN_SHAPE))
