@@ -231,3 +231,58 @@ def test_cpp_grpc_ssl_example(grpc_tls_server, tls_cert):
     )
     assert proc.returncode == 0, proc.stdout + proc.stderr
     assert "PASS" in proc.stdout
+
+
+def test_python_aio_https(tls_server, tls_cert):
+    """asyncio HTTP client over TLS with full verification against the
+    self-signed root (IP SAN)."""
+    import asyncio
+
+    import client_amd.http.aio as aiohttpclient
+
+    host, port = tls_server
+    cert, _ = tls_cert
+
+    async def run():
+        ctx = ssl_mod.create_default_context(cafile=str(cert))
+        client = aiohttpclient.InferenceServerClient(
+            f"{host}:{port}", ssl=True, ssl_context=ctx
+        )
+        try:
+            assert await client.is_server_live()
+            a, b = _addsub_io()
+            inputs = [
+                aiohttpclient.InferInput("INPUT0", [1, 16], "INT32"),
+                aiohttpclient.InferInput("INPUT1", [1, 16], "INT32"),
+            ]
+            inputs[0].set_data_from_numpy(a)
+            inputs[1].set_data_from_numpy(b)
+            result = await client.infer("simple", inputs)
+            np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), a + b)
+        finally:
+            await client.close()
+
+    asyncio.run(run())
+
+
+def test_python_aio_grpc_tls(grpc_tls_server, tls_cert):
+    import asyncio
+
+    import client_amd.grpc.aio as aiogrpcclient
+
+    host, port = grpc_tls_server
+    cert, _ = tls_cert
+
+    async def run():
+        client = aiogrpcclient.InferenceServerClient(
+            f"{host}:{port}",
+            ssl=True,
+            root_certificates=str(cert),
+            channel_args=[("grpc.ssl_target_name_override", "localhost")],
+        )
+        try:
+            assert await client.is_server_live()
+        finally:
+            await client.close()
+
+    asyncio.run(run())
