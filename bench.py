@@ -79,6 +79,7 @@ def main():
     ap.add_argument("--concurrency", type=int, default=8)
     ap.add_argument("--reqs-per-step", type=int, default=48)
     ap.add_argument("--no-dynamic-batching", action="store_true")
+    ap.add_argument("--preferred-batch-size", type=int, default=32)
     args = ap.parse_args()
 
     import torch
@@ -107,7 +108,8 @@ def main():
     import client_amd.utils.hip_shared_memory as hipshm
 
     server_args = [] if args.no_dynamic_batching else [
-        "--dynamic-batching", "--preferred-batch-size", "32",
+        "--dynamic-batching",
+        "--preferred-batch-size", str(args.preferred_batch_size),
         "--max-queue-delay-us", "400",
     ]
     server_proc, port = start_server(local_rank, 8101 + local_rank,
