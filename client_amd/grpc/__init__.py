@@ -13,6 +13,7 @@ from ._client import (
 )
 from ._infer_input import InferInput
 from ._infer_result import InferResult
+from . import model_config_pb2
 from ._proto import service_pb2
 from ._requested_output import InferRequestedOutput
 

@@ -57,3 +57,25 @@ def test_deprecation_shims_warn():
             warnings.simplefilter("always")
             importlib.import_module(name)
         assert any(issubclass(w.category, DeprecationWarning) for w in caught), name
+
+
+def test_model_config_pb2_alias():
+    """Reference user code: from tritonclient.grpc import model_config_pb2."""
+    from tritonclient.grpc import model_config_pb2 as mc
+
+    cfg = mc.ModelConfig()
+    cfg.name = "m"
+    inp = cfg.input.add()
+    inp.data_type = mc.TYPE_FP32
+    assert mc.TYPE_FP32 == 11
+    ig = cfg.instance_group.add()
+    ig.kind = mc.ModelInstanceGroup.KIND_GPU
+    data = cfg.SerializeToString()
+    cfg2 = mc.ModelConfig()
+    cfg2.ParseFromString(data)
+    assert cfg2.input[0].data_type == mc.TYPE_FP32
+    assert cfg2.instance_group[0].kind == mc.ModelInstanceGroup.KIND_GPU
+    # classes are the same objects the service schema uses
+    from tritonclient.grpc import service_pb2
+
+    assert mc.ModelConfig is service_pb2.ModelConfig
