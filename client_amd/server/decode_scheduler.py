@@ -74,6 +74,26 @@ class DecodeScheduler:
         self._worker = threading.Thread(target=self._run, daemon=True)
         self._worker.start()
 
+    @property
+    def kv_utilization(self):
+        """Fraction of decode slots occupied (ORCA endpoint-load-metrics
+        source; reference README.md:352-366)."""
+        busy = sum(1 for s in self.slots if s.state != _Slot.FREE)
+        return busy / max(1, len(self.slots))
+
+    def prewarm(self, n_buckets=2):
+        """Capture the first n decode-graph buckets up front. A cold
+        capture costs ~0.5 s and otherwise lands in the first stream's
+        TTFT (measured: TTFT p90 590 ms cold vs ~50 ms warm) — the
+        MI355X analog of the reference's ModelWarmup."""
+        if not self.use_graph:
+            return
+        for i in range(1, n_buckets + 1):
+            bucket = min(i * self.len_bucket, self.model.cfg.max_seq)
+            self._get_graph(bucket)
+            if bucket >= self.model.cfg.max_seq:
+                break
+
     def shutdown(self):
         with self._cv:
             self._alive = False

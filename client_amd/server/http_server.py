@@ -229,6 +229,20 @@ class HttpServer:
         response, binary_parts = self.core.infer(model_name, request, binary_buf)
         resp_json = json.dumps(response).encode()
         resp_headers = {}
+        # ORCA per-response load metrics (reference README.md:352-366):
+        # the client opts in via endpoint-load-metrics-format: text|json.
+        fmt = headers.get("endpoint-load-metrics-format")
+        model = self.core.models.get(model_name)
+        metrics_fn = getattr(model, "load_metrics", None)
+        if fmt in ("text", "json") and metrics_fn is not None:
+            metrics = metrics_fn()
+            if metrics:
+                if fmt == "json":
+                    resp_headers["endpoint-load-metrics"] = (
+                        "JSON " + json.dumps(metrics))
+                else:
+                    resp_headers["endpoint-load-metrics"] = "TEXT " + ", ".join(
+                        f"{k}={v:g}" for k, v in metrics.items())
         if binary_parts:
             resp_headers["Inference-Header-Content-Length"] = str(len(resp_json))
             resp_body = resp_json + b"".join(binary_parts)

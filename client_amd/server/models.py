@@ -209,6 +209,15 @@ class GenerateModel(Model):
             self._scheduler = DecodeScheduler(
                 self.module, max_batch=max_batch, device=device
             )
+            # capture the first decode-graph buckets now, not inside the
+            # first request's TTFT (cold capture ≈0.5 s per bucket)
+            self._scheduler.prewarm()
+
+    def load_metrics(self):
+        """ORCA named metrics for the endpoint-load-metrics header."""
+        if self._scheduler is None:
+            return {}
+        return {"kv_cache_utilization": self._scheduler.kv_utilization}
 
     def execute_decoupled(self, inputs, parameters):
         torch = self._torch

@@ -257,3 +257,45 @@ def test_large_tensor_roundtrip(client):
     np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
     # connection reuse after the big transfer
     assert client.is_server_live()
+
+
+def test_orca_load_metrics_header():
+    """Server emits ORCA endpoint-load-metrics when the client opts in
+    via endpoint-load-metrics-format (reference README.md:352-366);
+    surfaced by InferResult.get_response_header()."""
+    import numpy as np
+
+    import client_amd.http as httpclient
+    from client_amd.server import HttpServer
+    from client_amd.server.__main__ import build_core
+
+    core = build_core(["llama_tiny"], device="cpu", dtype="fp32")
+    server = HttpServer(core, host="127.0.0.1", port=0)
+    stop = server.serve_forever_in_thread()
+    try:
+        client = httpclient.InferenceServerClient(f"127.0.0.1:{server.port}")
+        inputs = [
+            httpclient.InferInput("input_ids", [4], "INT64"),
+            httpclient.InferInput("max_tokens", [1], "INT32"),
+        ]
+        inputs[0].set_data_from_numpy(np.array([1, 2, 3, 4], dtype=np.int64))
+        inputs[1].set_data_from_numpy(np.array([2], dtype=np.int32))
+        result = client.infer(
+            "llama_tiny", inputs,
+            headers={"endpoint-load-metrics-format": "text"},
+        )
+        hdr = result.get_response_header("endpoint-load-metrics")
+        assert hdr is not None and hdr.startswith("TEXT ")
+        assert "kv_cache_utilization=" in hdr
+        result_json = client.infer(
+            "llama_tiny", inputs,
+            headers={"endpoint-load-metrics-format": "json"},
+        )
+        hdr2 = result_json.get_response_header("endpoint-load-metrics")
+        assert hdr2.startswith("JSON ")
+        # no opt-in -> no header
+        result_none = client.infer("llama_tiny", inputs)
+        assert result_none.get_response_header("endpoint-load-metrics") is None
+        client.close()
+    finally:
+        stop()
