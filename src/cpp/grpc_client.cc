@@ -382,25 +382,32 @@ Error InferenceServerGrpcClient::AsyncUnaryCall(
 Error InferenceServerGrpcClient::UnaryCall(
     const std::string& method, const std::string& request,
     std::string* response, uint64_t timeout_us) {
-  std::mutex mu;
-  std::condition_variable cv;
-  bool done = false;
-  Error result_err("");
-  std::string result_body;
+  // Heap-allocated wait state: the completion callback may retain it
+  // past this frame on error paths, and fresh heap addresses keep
+  // TSAN's mutex shadow clean (stack-reused std::mutex never runs
+  // pthread_mutex_destroy and poisons later reports).
+  struct WaitState {
+    std::mutex mu;
+    std::condition_variable cv;
+    bool done = false;
+    Error err{""};
+    std::string body;
+  };
+  auto st = std::make_shared<WaitState>();
   RETURN_IF_ERROR(AsyncUnaryCall(
       method, request,
-      [&](Error err, std::string body) {
-        std::lock_guard<std::mutex> lock(mu);
-        result_err = std::move(err);
-        result_body = std::move(body);
-        done = true;
-        cv.notify_all();
+      [st](Error err, std::string body) {
+        std::lock_guard<std::mutex> lock(st->mu);
+        st->err = std::move(err);
+        st->body = std::move(body);
+        st->done = true;
+        st->cv.notify_all();
       },
       timeout_us));
-  std::unique_lock<std::mutex> lock(mu);
-  cv.wait(lock, [&] { return done; });
-  RETURN_IF_ERROR(result_err);
-  *response = std::move(result_body);
+  std::unique_lock<std::mutex> lock(st->mu);
+  st->cv.wait(lock, [&] { return st->done; });
+  RETURN_IF_ERROR(st->err);
+  *response = std::move(st->body);
   return Error::Success;
 }
 
@@ -665,22 +672,25 @@ Error InferenceServerGrpcClient::Infer(
     InferResult** result, const InferOptions& options,
     const std::vector<InferInput*>& inputs,
     const std::vector<const InferRequestedOutput*>& outputs) {
-  std::mutex mu;
-  std::condition_variable cv;
-  InferResult* res = nullptr;
-  bool done = false;
+  struct WaitState {  // heap state, same rationale as UnaryCall
+    std::mutex mu;
+    std::condition_variable cv;
+    InferResult* res = nullptr;
+    bool done = false;
+  };
+  auto st = std::make_shared<WaitState>();
   RETURN_IF_ERROR(AsyncInfer(
-      [&](InferResult* r) {
-        std::lock_guard<std::mutex> lock(mu);
-        res = r;
-        done = true;
-        cv.notify_all();
+      [st](InferResult* r) {
+        std::lock_guard<std::mutex> lock(st->mu);
+        st->res = r;
+        st->done = true;
+        st->cv.notify_all();
       },
       options, inputs, outputs));
-  std::unique_lock<std::mutex> lock(mu);
-  cv.wait(lock, [&] { return done; });
-  *result = res;
-  return res->RequestStatus();
+  std::unique_lock<std::mutex> lock(st->mu);
+  st->cv.wait(lock, [&] { return st->done; });
+  *result = st->res;
+  return st->res->RequestStatus();
 }
 
 Error InferenceServerGrpcClient::InferMulti(

@@ -7,6 +7,7 @@
 // — the GPU surface is hip_shm.h.
 #pragma once
 
+#include <atomic>
 #include <chrono>
 #include <cstdint>
 #include <functional>
@@ -320,7 +321,9 @@ class InferenceServerClient {
   Error UpdateInferStat(const RequestTimers& timer);
 
   bool verbose_;
-  bool exiting_;
+  // read lock-free by worker loops (while (!exiting_)), written by the
+  // destructor thread — must be atomic (TSAN-verified)
+  std::atomic<bool> exiting_;
   mutable std::mutex stat_mu_;
   InferStat infer_stat_;
 };

@@ -110,7 +110,7 @@ class H2Connection {
   std::mutex io_mu_;         // guards every SSL_read/SSL_write
   std::thread reader_;
   std::atomic_bool alive_{false};
-  bool exiting_ = false;
+  std::atomic_bool exiting_{false};  // also read outside mu_
 
   std::mutex write_mu_;
   int32_t next_stream_id_ = 1;

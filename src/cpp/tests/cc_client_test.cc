@@ -198,8 +198,10 @@ int main(int argc, char** argv) {
 
   // ---- async infer ----
   {
-    std::mutex mu;
-    std::condition_variable cv;
+    // static: stack-reused std::mutex never runs pthread_mutex_destroy,
+    // which poisons TSAN mutex shadow across test sections
+    static std::mutex mu;
+    static std::condition_variable cv;
     int completed = 0;
     const int kAsync = 8;
     bool all_ok = true;
@@ -207,8 +209,9 @@ int main(int argc, char** argv) {
       Error err = client->AsyncInfer(
           [&](InferResult* res) {
             std::unique_ptr<InferResult> owned(res);
-            if (!owned->RequestStatus().IsOk()) all_ok = false;
+            bool ok = owned->RequestStatus().IsOk();
             std::lock_guard<std::mutex> lock(mu);
+            if (!ok) all_ok = false;
             completed++;
             cv.notify_all();
           },

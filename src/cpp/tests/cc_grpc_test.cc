@@ -120,8 +120,10 @@ int main(int argc, char** argv) {
 
   // ---- async infer ----
   {
-    std::mutex mu;
-    std::condition_variable cv;
+    // static: stack-reused std::mutex never runs pthread_mutex_destroy,
+    // which poisons TSAN mutex shadow across test sections
+    static std::mutex mu;
+    static std::condition_variable cv;
     int completed = 0;
     bool all_ok = true;
     const int kAsync = 8;
@@ -129,8 +131,9 @@ int main(int argc, char** argv) {
       CHECK_OK(client->AsyncInfer(
           [&](InferResult* res) {
             std::unique_ptr<InferResult> owned(res);
-            if (!owned->RequestStatus().IsOk()) all_ok = false;
+            bool ok = owned->RequestStatus().IsOk();
             std::lock_guard<std::mutex> lock(mu);
+            if (!ok) all_ok = false;
             completed++;
             cv.notify_all();
           },
@@ -176,8 +179,10 @@ int main(int argc, char** argv) {
 
   // ---- bi-di stream: sequence accumulation ----
   {
-    std::mutex mu;
-    std::condition_variable cv;
+    // static: stack-reused std::mutex never runs pthread_mutex_destroy,
+    // which poisons TSAN mutex shadow across test sections
+    static std::mutex mu;
+    static std::condition_variable cv;
     std::vector<int32_t> seen;
     bool stream_error = false;
     CHECK_OK(client->StartStream([&](InferResult* res) {

@@ -49,8 +49,10 @@ int main(int argc, char** argv) {
   // async path: several concurrent transfers, each with its own
   // non-blocking TLS handshake inside the epoll worker
   {
-    std::mutex mu;
-    std::condition_variable cv;
+    // static: stack-reused std::mutex never runs pthread_mutex_destroy,
+    // which poisons TSAN mutex shadow across test sections
+    static std::mutex mu;
+    static std::condition_variable cv;
     int done = 0;
     bool all_ok = true;
     const int kAsync = 6;
