@@ -370,3 +370,53 @@ def test_cc_image_clients_compile():
     _compile("image_client", CPP / "examples" / "image_client.cc")
     _compile("ensemble_image_client",
              CPP / "examples" / "ensemble_image_client.cc")
+
+
+def test_cc_keepalive_watchdog_kills_dead_connection():
+    """A server that accepts h2 but never ACKs PINGs must be declared
+    dead by the keepalive watchdog, failing the in-flight Infer fast
+    (grpc keepalive.md semantics)."""
+    import socket
+    import threading
+
+    binary = _compile("keepalive_timeout_smoke",
+                      CPP / "tests" / "keepalive_timeout_smoke.cc")
+
+    srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(1)
+    port = srv.getsockname()[1]
+    stop = threading.Event()
+
+    def fake_h2_server():
+        try:
+            conn, _ = srv.accept()
+            conn.settimeout(0.5)
+            # empty SETTINGS so the client sees a live h2 peer
+            conn.sendall(b"\x00\x00\x00\x04\x00\x00\x00\x00\x00")
+            while not stop.is_set():
+                try:
+                    if conn.recv(65536) == b"":
+                        break  # drained; never answer anything
+                except socket.timeout:
+                    continue
+                except OSError:
+                    break
+            conn.close()
+        except Exception:
+            pass
+
+    t = threading.Thread(target=fake_h2_server, daemon=True)
+    t.start()
+    try:
+        proc = subprocess.run(
+            [str(binary), "127.0.0.1", str(port)], capture_output=True,
+            text=True, timeout=30,
+        )
+        assert proc.returncode == 0, proc.stdout + proc.stderr
+        assert "PASS" in proc.stdout
+    finally:
+        stop.set()
+        srv.close()
+        t.join(timeout=5)
