@@ -187,6 +187,9 @@ def main(argv=None):
     parser.add_argument("--max-queue-delay-us", type=int, default=500)
     parser.add_argument("--decode-max-batch", type=int, default=8,
                         help="continuous-batching slots for generate models")
+    parser.add_argument("--model-warmup", action="store_true",
+                        help="pre-capture hipGraphs / prime MIOpen for the "
+                             "serving batch sizes before READY is printed")
     args = parser.parse_args(argv)
 
     core = build_core(
@@ -199,6 +202,12 @@ def main(argv=None):
                 model.enable_dynamic_batching(
                     args.preferred_batch_size, args.max_queue_delay_us
                 )
+    if args.model_warmup:
+        sizes = (8, args.preferred_batch_size) if args.dynamic_batching \
+            else (8,)
+        for model in core.models.values():
+            if hasattr(model, "warmup"):
+                model.warmup(sizes)
 
     stoppers = []
     if args.grpc_port != 0:

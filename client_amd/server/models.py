@@ -401,6 +401,31 @@ class TorchModel(Model):
             self.max_batch_size = max_batch_size
         return self
 
+    def warmup(self, batch_sizes=(8, 32)):
+        """Pre-capture hipGraphs (and prime MIOpen kernel caches) for
+        the given batch sizes so cold requests don't pay capture
+        latency — the reference ModelWarmup analog (model_config.proto
+        ModelWarmup; our decode path has the same in
+        DecodeScheduler.prewarm)."""
+        if not self.device.startswith("cuda"):
+            return
+        torch = self._torch
+        for bs in batch_sizes:
+            tensors = []
+            for name, datatype, shape in self.inputs:
+                dims = [bs if d == -1 else d for d in shape]
+                if len(shape) > 0 and shape[0] == -1:
+                    dims[0] = bs
+                npdt = triton_to_np_dtype(datatype)
+                t = torch.zeros(dims, device=self.device,
+                                dtype=torch.float32 if npdt in (np.float32,)
+                                else torch.from_numpy(
+                                    np.zeros(1, npdt)).dtype)
+                if self.dtype is not None and t.is_floating_point():
+                    t = t.to(self.dtype)
+                tensors.append(t)
+            self._execute_direct(tensors)
+
     def execute(self, inputs, parameters):
         torch = self._torch
         with torch.inference_mode():

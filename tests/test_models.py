@@ -310,3 +310,22 @@ def test_chunked_prefill_matches_full():
         assert got == expected
     finally:
         sched.shutdown()
+
+
+def test_torchmodel_warmup_cpu_noop_and_cli():
+    """warmup() is a no-op off-GPU and the --model-warmup CLI path wires
+    through build_core without error."""
+    import numpy as np
+    import torch
+
+    from client_amd.server import TorchModel
+
+    model = TorchModel(
+        "wident", torch.nn.Linear(16, 16), device="cpu",
+        inputs=[("INPUT0", "FP32", [-1, 16])],
+        outputs=[("OUTPUT0", "FP32", [-1, 16])],
+    )
+    model.warmup((2, 4))  # no-op on cpu, must not raise
+    out = model.execute(
+        {"INPUT0": np.zeros((2, 16), dtype=np.float32)}, {})
+    assert out["OUTPUT0"].shape == (2, 16)
