@@ -420,3 +420,34 @@ def test_cc_keepalive_watchdog_kills_dead_connection():
         stop.set()
         srv.close()
         t.join(timeout=5)
+
+
+def test_cc_client_timeout_suite(http_fixture_server, grpc_fixture_server):
+    """Reference client_timeout_test analog: every API with microscopic
+    deadlines fails cleanly with Deadline Exceeded on both transports."""
+    http_host, http_port, _ = http_fixture_server
+    grpc_host, grpc_port, _ = grpc_fixture_server
+    binary = _compile("client_timeout_test",
+                      CPP / "tests" / "client_timeout_test.cc")
+    proc = subprocess.run(
+        [str(binary), f"{http_host}:{http_port}", f"{grpc_host}:{grpc_port}"],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "ALL PASSED" in proc.stdout
+
+
+def test_cc_memory_leak_suite(http_fixture_server, grpc_fixture_server):
+    """Reference memory_leak_test analog: repeated inference + shm churn
+    with an RSS growth budget."""
+    http_host, http_port, _ = http_fixture_server
+    grpc_host, grpc_port, _ = grpc_fixture_server
+    binary = _compile("memory_leak_test",
+                      CPP / "tests" / "memory_leak_test.cc")
+    proc = subprocess.run(
+        [str(binary), f"{http_host}:{http_port}", f"{grpc_host}:{grpc_port}",
+         "300"],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "ALL PASSED" in proc.stdout
