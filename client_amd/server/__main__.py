@@ -93,7 +93,10 @@ def build_core(model_names, device="cuda:0", dtype="bf16",
                     outputs=[("pooled", io_dt, [-1, hidden])],
                     device=device,
                     dtype=tdt if device.startswith("cuda") else None,
-                    use_graph=False,  # dynamic seq lengths
+                    # per-signature capture: each new (batch, seq) shape
+                    # captures once (~0.5 s) then replays; serving
+                    # traffic with few shapes amortizes immediately
+                    use_graph=True,
                 )
             )
         elif name in ("llama3_8b", "llama_tiny"):
