@@ -352,7 +352,7 @@ class PerfAnalyzer:
             for name, datatype, shape in inputs:
                 elem = TRITON_DTYPE_SIZES.get(datatype, 4)
                 nbytes = int(np.prod(shape)) * elem
-                rname = f"pa_in_{slot}_{name}"
+                rname = f"pa_{self.model_name}_in_{slot}_{name}"
                 region = hipshm.create_shared_memory_region(
                     rname, nbytes, self.device_id
                 )
@@ -380,7 +380,7 @@ class PerfAnalyzer:
             for name, datatype, shape in outputs:
                 elem = TRITON_DTYPE_SIZES.get(datatype, 4)
                 nbytes = int(np.prod(shape)) * elem
-                rname = f"pa_out_{slot}_{name}"
+                rname = f"pa_{self.model_name}_out_{slot}_{name}"
                 region = hipshm.create_shared_memory_region(
                     rname, nbytes, self.device_id
                 )
@@ -403,11 +403,13 @@ class PerfAnalyzer:
     def _teardown_hipshm(self, client):
         import client_amd.utils.hip_shared_memory as hipshm
 
-        try:
-            client.unregister_cuda_shared_memory()
-        except Exception:
-            pass
         for region in getattr(self, "_regions", []):
+            try:
+                # unregister by name: co-serving load generators must not
+                # blow away each other's regions (model-scoped names)
+                client.unregister_cuda_shared_memory(region._triton_shm_name)
+            except Exception:
+                pass
             try:
                 hipshm.destroy_shared_memory_region(region)
             except Exception:
