@@ -494,6 +494,25 @@ Error InferenceServerGrpcClient::UnloadModel(const std::string& model_name) {
       kserve::EncodeRepositoryModelRequest(model_name), &resp);
 }
 
+Error InferenceServerGrpcClient::UpdateTraceSettings(
+    kserve::TraceSettingsPb* response, const std::string& model_name,
+    const kserve::TraceSettingsPb& settings) {
+  std::string resp;
+  RETURN_IF_ERROR(UnaryCall(
+      "TraceSetting",
+      kserve::EncodeTraceSettingRequest(settings, model_name), &resp));
+  if (response != nullptr) {
+    *response = kserve::DecodeTraceSettingResponse(
+        (const uint8_t*)resp.data(), resp.size());
+  }
+  return Error::Success;
+}
+
+Error InferenceServerGrpcClient::GetTraceSettings(
+    kserve::TraceSettingsPb* settings, const std::string& model_name) {
+  return UpdateTraceSettings(settings, model_name, {});
+}
+
 Error InferenceServerGrpcClient::ModelInferenceStatistics(
     std::vector<kserve::ModelStatisticsPb>* stats,
     const std::string& model_name, const std::string& version) {

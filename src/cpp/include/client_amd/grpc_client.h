@@ -68,6 +68,14 @@ class InferenceServerGrpcClient : public InferenceServerClient {
       std::vector<kserve::RepositoryIndexEntryPb>* index);
   Error LoadModel(const std::string& model_name);
   Error UnloadModel(const std::string& model_name);
+  // Reference grpc_client.h:330-352 trace-settings RPCs (the reference
+  // has no gRPC log-settings; ours adds none either — Python covers it).
+  Error UpdateTraceSettings(
+      kserve::TraceSettingsPb* response, const std::string& model_name = "",
+      const kserve::TraceSettingsPb& settings = {});
+  Error GetTraceSettings(
+      kserve::TraceSettingsPb* settings, const std::string& model_name = "");
+
   Error ModelInferenceStatistics(
       std::vector<kserve::ModelStatisticsPb>* stats,
       const std::string& model_name = "", const std::string& version = "");

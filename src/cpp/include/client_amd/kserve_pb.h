@@ -179,6 +179,13 @@ struct ModelConfigPb {
   static ModelConfigPb Decode(const uint8_t* data, size_t n);
 };
 
+// Trace settings (grpc_service.proto:1673-1741): name -> list of
+// string values, same map on request and response.
+using TraceSettingsPb = std::map<std::string, std::vector<std::string>>;
+std::string EncodeTraceSettingRequest(const TraceSettingsPb& settings,
+                                      const std::string& model_name);
+TraceSettingsPb DecodeTraceSettingResponse(const uint8_t* data, size_t n);
+
 struct RepositoryIndexEntryPb {
   std::string name;
   std::string version;

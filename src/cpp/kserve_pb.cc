@@ -382,6 +382,58 @@ std::vector<ModelStatisticsPb> DecodeModelStatistics(const uint8_t* data,
 }
 
 //==============================================================================
+// Trace settings
+
+std::string EncodeTraceSettingRequest(const TraceSettingsPb& settings,
+                                      const std::string& model_name) {
+  pb::Writer w;
+  for (const auto& kv : settings) {
+    pb::Writer val;  // SettingValue { repeated string value = 1; }
+    for (const auto& v : kv.second) val.put_str(1, v);
+    pb::Writer entry;
+    entry.put_str(1, kv.first);
+    entry.put_msg(2, val.out);
+    w.put_msg(1, entry.out);
+  }
+  w.put_str(2, model_name);
+  return w.out;
+}
+
+TraceSettingsPb DecodeTraceSettingResponse(const uint8_t* data, size_t n) {
+  TraceSettingsPb out;
+  Reader r(data, n);
+  int field, wire;
+  while (r.next(&field, &wire)) {
+    if (field == 1 && wire == pb::LEN) {
+      auto [ptr, len] = r.bytes();
+      Reader entry(ptr, len);
+      std::string key;
+      std::vector<std::string> values;
+      int f2, w2;
+      while (entry.next(&f2, &w2)) {
+        if (f2 == 1) {
+          key = entry.str();
+        } else if (f2 == 2 && w2 == pb::LEN) {
+          auto [vp, vl] = entry.bytes();
+          Reader val(vp, vl);
+          int f3, w3;
+          while (val.next(&f3, &w3)) {
+            if (f3 == 1) values.push_back(val.str());
+            else val.skip(w3);
+          }
+        } else {
+          entry.skip(w2);
+        }
+      }
+      out[key] = std::move(values);
+    } else {
+      r.skip(wire);
+    }
+  }
+  return out;
+}
+
+//==============================================================================
 // ModelConfig decode (subset; see kserve_pb.h).
 
 namespace {

@@ -296,6 +296,19 @@ int main(int argc, char** argv) {
   CHECK(!bad_err.IsOk());
   delete bad_result;
 
+  // ---- trace settings (reference cc_client_test trace fixtures) ----
+  {
+    kserve::TraceSettingsPb settings;
+    CHECK_OK(client->GetTraceSettings(&settings));
+    kserve::TraceSettingsPb update{{"trace_level", {"TIMESTAMPS"}},
+                                   {"trace_rate", {"6"}}};
+    kserve::TraceSettingsPb after;
+    CHECK_OK(client->UpdateTraceSettings(&after, "", update));
+    CHECK(after.count("trace_rate") && after["trace_rate"][0] == "6");
+    CHECK_OK(client->GetTraceSettings(&after));
+    CHECK(after.count("trace_rate") && after["trace_rate"][0] == "6");
+  }
+
   // ---- client-side stat ----
   InferStat stat;
   CHECK_OK(client->ClientInferStat(&stat));
