@@ -22,7 +22,7 @@ sched.prewarm()
 
 outs = []
 for i in range(8):
-    ids = np.random.randint(0, cfg.vocab, 128).astype(np.int64)
+    ids = np.random.randint(0, cfg.vocab_size, 128).astype(np.int64)
     outs.append(sched.submit(ids, 64))
 total = 0
 for q in outs:
