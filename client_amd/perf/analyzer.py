@@ -466,6 +466,56 @@ class PerfAnalyzer:
                 )
         return issue
 
+    def _server_stats_snapshot(self, client):
+        """Cumulative server-side durations for the model (KServe-v2
+        statistics; perf_analyzer's Server Queue / Compute breakdown)."""
+        try:
+            stats = client.get_inference_statistics(self.model_name)
+            if self.protocol == "grpc":
+                m = stats.model_stats[0]
+                inf = m.inference_stats
+                return {
+                    "count": inf.success.count,
+                    "queue_ns": inf.queue.ns,
+                    "compute_input_ns": inf.compute_input.ns,
+                    "compute_infer_ns": inf.compute_infer.ns,
+                    "compute_output_ns": inf.compute_output.ns,
+                }
+            m = stats["model_stats"][0]
+            inf = m["inference_stats"]
+            return {
+                "count": inf["success"]["count"],
+                "queue_ns": inf["queue"]["ns"],
+                "compute_input_ns": inf["compute_input"]["ns"],
+                "compute_infer_ns": inf["compute_infer"]["ns"],
+                "compute_output_ns": inf["compute_output"]["ns"],
+            }
+        except Exception:
+            return None
+
+    @staticmethod
+    def _server_breakdown(before, after):
+        """Delta two snapshots -> avg per-request server-side times."""
+        if not before or not after:
+            return None
+        n = after["count"] - before["count"]
+        if n <= 0:
+            return None
+        return {
+            "requests": n,
+            "avg_queue_us": int((after["queue_ns"] - before["queue_ns"])
+                                / n / 1000),
+            "avg_compute_input_us": int(
+                (after["compute_input_ns"] - before["compute_input_ns"])
+                / n / 1000),
+            "avg_compute_infer_us": int(
+                (after["compute_infer_ns"] - before["compute_infer_ns"])
+                / n / 1000),
+            "avg_compute_output_us": int(
+                (after["compute_output_ns"] - before["compute_output_ns"])
+                / n / 1000),
+        }
+
     def _result_dict(self, throughput, lat, errors, windows, **extra):
         latency_us = {
             "avg": int(np.mean(lat) / 1000) if lat else 0,
@@ -500,11 +550,15 @@ class PerfAnalyzer:
             try:
                 issue = self._build_issue(client, mod, concurrency)
                 driver = ConcurrencyDriver(issue, concurrency)
+                stats_before = self._server_stats_snapshot(client)
                 throughput, lat, errors, windows = driver.run(
                     warmup_s, window_s, max_windows, stability_pct
                 )
+                server = self._server_breakdown(
+                    stats_before, self._server_stats_snapshot(client))
                 result = self._result_dict(
-                    throughput, lat, errors, windows, concurrency=concurrency
+                    throughput, lat, errors, windows, concurrency=concurrency,
+                    **({"server": server} if server else {}),
                 )
                 results.append(result)
                 if self.verbose:

@@ -163,3 +163,20 @@ def test_perf_cli_request_rate(grpc_fixture_server, tmp_path, capsys):
     assert "Request rate: 20" in captured.out
     assert (tmp_path / "rate.csv").read_text().startswith(
         "Request Rate,Inferences/Second")
+
+
+def test_perf_server_breakdown(grpc_fixture_server):
+    """Result includes perf_analyzer's server-side queue/compute
+    breakdown derived from KServe-v2 statistics deltas."""
+    host, port, _ = grpc_fixture_server
+    pa = PerfAnalyzer(
+        url=f"{host}:{port}", protocol="grpc", model_name="simple",
+        batch_size=1,
+    )
+    r = pa.run([2], warmup_s=0.1, window_s=0.3, max_windows=2)[0]
+    assert "server" in r
+    srv = r["server"]
+    assert srv["requests"] > 0
+    for k in ("avg_queue_us", "avg_compute_input_us",
+              "avg_compute_infer_us", "avg_compute_output_us"):
+        assert srv[k] >= 0
