@@ -582,14 +582,18 @@ class PerfAnalyzer:
                 driver = RequestRateDriver(
                     issue, rate, max_threads, distribution
                 )
+                stats_before = self._server_stats_snapshot(client)
                 throughput, lat, errors, windows = driver.run(
                     warmup_s, window_s, max_windows, stability_pct
                 )
+                server = self._server_breakdown(
+                    stats_before, self._server_stats_snapshot(client))
                 result = self._result_dict(
                     throughput, lat, errors, windows,
                     target_request_rate=rate,
                     request_distribution=distribution,
                     delayed_requests=driver.delayed,
+                    **({"server": server} if server else {}),
                 )
                 results.append(result)
                 if self.verbose:
