@@ -347,6 +347,9 @@ Error InferenceServerHttpClient::Create(
 // TLS plumbing (sync path). OpenSSL is initialized lazily per client.
 
 Error InferenceServerHttpClient::EnsureSslCtx() {
+  // sync callers and the async worker can both get here — serialize
+  // context creation (mu_ also guards new_transfers_; creation is rare)
+  std::lock_guard<std::mutex> ctx_lock(mu_);
   if (ssl_ctx_ != nullptr) return Error::Success;
   static std::once_flag ssl_init;
   // SIGPIPE: SSL_write lacks MSG_NOSIGNAL; a peer reset would kill
@@ -1223,33 +1226,36 @@ void InferenceServerHttpClient::AsyncWorker() {
 
   char chunk[kRecvChunk];
   while (!exiting_) {
-    // adopt new transfers
+    // adopt new transfers: drain the queue under mu_, then do the
+    // slow work (connect, TLS object setup) without holding it so
+    // AsyncInfer submitters never block behind a connect()
+    std::deque<std::unique_ptr<AsyncTransfer>> adopted;
     {
       std::lock_guard<std::mutex> lock(mu_);
-      while (!new_transfers_.empty()) {
-        auto t = std::move(new_transfers_.front());
-        new_transfers_.pop_front();
-        t->fd = ConnectTo(host_, port_, true);
-        if (t->fd < 0) {
-          finish(std::move(t), 0, "failed to connect");
+      adopted.swap(new_transfers_);
+    }
+    for (auto& tp : adopted) {
+      auto t = std::move(tp);
+      t->fd = ConnectTo(host_, port_, true);
+      if (t->fd < 0) {
+        finish(std::move(t), 0, "failed to connect");
+        continue;
+      }
+      if (use_ssl_) {
+        void* ssl_v = nullptr;
+        Error serr = NewSsl(t->fd, &ssl_v);
+        if (!serr.IsOk()) {
+          finish(std::move(t), 0, serr.Message());
           continue;
         }
-        if (use_ssl_) {
-          void* ssl_v = nullptr;
-          Error serr = NewSsl(t->fd, &ssl_v);
-          if (!serr.IsOk()) {
-            finish(std::move(t), 0, serr.Message());
-            continue;
-          }
-          t->ssl = (SSL*)ssl_v;
-        }
-        t->timer.CaptureTimestamp(RequestTimers::Kind::SEND_START);
-        struct epoll_event tev;
-        tev.events = EPOLLOUT | EPOLLIN;
-        tev.data.fd = t->fd;
-        epoll_ctl(epfd, EPOLL_CTL_ADD, t->fd, &tev);
-        active[t->fd] = std::move(t);
+        t->ssl = (SSL*)ssl_v;
       }
+      t->timer.CaptureTimestamp(RequestTimers::Kind::SEND_START);
+      struct epoll_event tev;
+      tev.events = EPOLLOUT | EPOLLIN;
+      tev.data.fd = t->fd;
+      epoll_ctl(epfd, EPOLL_CTL_ADD, t->fd, &tev);
+      active[t->fd] = std::move(t);
     }
 
     struct epoll_event events[64];
