@@ -121,6 +121,19 @@ class _HttpProtocol(asyncio.Protocol):
             resp_body = json.dumps({"error": str(e)}).encode()
             status = 500
             resp_headers = {"Content-Type": "application/json"}
+        # whole-body response compression when the client asked for it
+        # (tritonclient response_compression_algorithm sends
+        # Accept-Encoding; InferResult gunzips the whole body)
+        accept = headers.get("accept-encoding", "")
+        if status == 200 and len(resp_body) > 256 \
+                and "content-encoding" not in {k.lower()
+                                               for k in resp_headers}:
+            if "gzip" in accept:
+                resp_body = gzip.compress(resp_body)
+                resp_headers["Content-Encoding"] = "gzip"
+            elif "deflate" in accept:
+                resp_body = zlib.compress(resp_body)
+                resp_headers["Content-Encoding"] = "deflate"
         reason = {200: "OK", 400: "Bad Request", 404: "Not Found",
                   500: "Internal Server Error"}.get(status, "OK")
         out = [f"HTTP/1.1 {status} {reason}\r\n".encode()]

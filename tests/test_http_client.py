@@ -163,6 +163,20 @@ def test_compression(client):
         np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
 
 
+def test_response_compression(client):
+    """response_compression_algorithm: server gzips/deflates the whole
+    body (Accept-Encoding), InferResult transparently decompresses."""
+    x = np.random.rand(1, 1024).astype(np.float32)
+    inp = httpclient.InferInput("INPUT0", list(x.shape), "FP32")
+    inp.set_data_from_numpy(x)
+    for algo in ("gzip", "deflate"):
+        result = client.infer(
+            "identity_fp32", [inp], response_compression_algorithm=algo
+        )
+        assert result.get_response_header("content-encoding") == algo
+        np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
+
+
 def test_error_mapping(client):
     with pytest.raises(InferenceServerException) as exc:
         client.get_model_metadata("not_a_model")
