@@ -55,6 +55,10 @@ class InferenceServerClient(InferenceServerClientBase):
         self._session = aiohttp.ClientSession(
             connector=connector, timeout=aiohttp.ClientTimeout(total=conn_timeout),
             auto_decompress=False,
+            # aiohttp otherwise advertises gzip itself; compression is
+            # opt-in via response_compression_algorithm (we decompress
+            # manually in InferResult)
+            headers={"Accept-Encoding": "identity"},
         )
 
     async def __aenter__(self):

@@ -93,3 +93,32 @@ def test_grpc_aio_stream(grpc_fixture_server):
             assert vals == [1, 3, 6]
 
     _run(main())
+
+
+def test_http_aio_response_compression(http_fixture_server):
+    """aio response_compression_algorithm: opt-in Accept-Encoding with
+    manual whole-body decompression in InferResult."""
+    import asyncio
+
+    import numpy as np
+
+    import client_amd.http.aio as aiohttpclient
+
+    host, port, _ = http_fixture_server
+
+    async def run():
+        client = aiohttpclient.InferenceServerClient(f"{host}:{port}")
+        try:
+            x = np.random.rand(1, 1024).astype(np.float32)
+            inp = aiohttpclient.InferInput("INPUT0", list(x.shape), "FP32")
+            inp.set_data_from_numpy(x)
+            for algo in ("gzip", "deflate"):
+                result = await client.infer(
+                    "identity_fp32", [inp],
+                    response_compression_algorithm=algo,
+                )
+                np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
+        finally:
+            await client.close()
+
+    asyncio.run(run())
