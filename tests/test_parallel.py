@@ -69,3 +69,24 @@ def test_broadcaster_single_process_noop():
     bc = RegionBroadcaster(t)
     assert bc.broadcast() is None  # no dist group -> no-op
     assert aggregate_max(3.5, device="cpu") == 3.5
+
+
+def test_region_broadcast_gloo_world4():
+    """World=4 replica fan-out (the 8-GPU shape at CI scale): broadcast +
+    max-aggregation across two ranks per... four single-slot replicas."""
+    ctx = mp.get_context("spawn")
+    port = 29716
+    procs = []
+    conns = []
+    for rank in range(4):
+        parent, child = ctx.Pipe()
+        p = ctx.Process(target=_rank_main, args=(rank, 4, port, child))
+        p.start()
+        procs.append(p)
+        conns.append(parent)
+    for rank, (p, conn) in enumerate(zip(procs, conns)):
+        assert conn.poll(180), f"rank {rank} timed out"
+        msg = conn.recv()
+        assert msg == "ok", msg
+    for p in procs:
+        p.join(timeout=30)
