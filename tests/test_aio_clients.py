@@ -122,3 +122,43 @@ def test_http_aio_response_compression(http_fixture_server):
             await client.close()
 
     asyncio.run(run())
+
+
+def test_grpc_aio_stream_cancel(grpc_fixture_server):
+    """aio stream iterator .cancel() aborts a slow decoupled request."""
+    import asyncio
+    import time
+
+    import numpy as np
+
+    import client_amd.grpc.aio as aiogrpc
+
+    host, port, _ = grpc_fixture_server
+
+    async def run():
+        client = aiogrpc.InferenceServerClient(f"{host}:{port}")
+        try:
+            async def requests():
+                inputs = [
+                    aiogrpc.InferInput("IN", [4], "INT32"),
+                    aiogrpc.InferInput("DELAY", [4], "UINT32"),
+                ]
+                inputs[0].set_data_from_numpy(np.arange(4, dtype=np.int32))
+                inputs[1].set_data_from_numpy(
+                    np.full(4, 800, dtype=np.uint32))
+                yield {"model_name": "repeat_int32", "inputs": inputs}
+
+            it = client.stream_infer(requests())
+            got_one = False
+            t0 = time.monotonic()
+            async for result, error in it:
+                assert error is None
+                got_one = True
+                it.cancel()
+                break
+            assert got_one
+            assert time.monotonic() - t0 < 5.0
+        finally:
+            await client.close()
+
+    asyncio.run(run())

@@ -349,3 +349,70 @@ def test_model_config_full_schema_roundtrip():
     cfg3.ParseFromString(cfg2.SerializeToString())
     assert cfg3.sequence_batching.oldest.max_candidate_sequences == 4
     assert cfg3.sequence_batching.state[0].initial_state[0].zero_data is True
+
+
+def test_async_infer_cancel(grpc_fixture_server):
+    """CallContext.cancel() on an in-flight delayed request surfaces a
+    CANCELLED error to the callback (reference grpc/_client.py:101-117
+    semantics) and leaves the client usable."""
+    import queue
+    import numpy as np
+
+    import client_amd.grpc as grpcclient
+
+    host, port, _ = grpc_fixture_server
+    client = grpcclient.InferenceServerClient(f"{host}:{port}")
+    try:
+        inputs = [
+            grpcclient.InferInput("IN", [1], "INT32"),
+            grpcclient.InferInput("DELAY", [1], "UINT32"),
+        ]
+        inputs[0].set_data_from_numpy(np.array([1], dtype=np.int32))
+        inputs[1].set_data_from_numpy(np.array([1500], dtype=np.uint32))
+        events = queue.Queue()
+        ctx = client.async_infer(
+            "repeat_int32", inputs,
+            callback=lambda result, error: events.put((result, error)),
+        )
+        assert ctx.cancel() or True  # cancel may race completion
+        result, error = events.get(timeout=30)
+        # either cancelled (normal) or, if it raced, a completed result
+        if error is not None:
+            assert "CANCELLED" in str(error).upper() or "cancel" in str(
+                error).lower()
+        # client still works
+        assert client.is_server_live()
+    finally:
+        client.close()
+
+
+def test_stop_stream_cancel_requests(grpc_fixture_server):
+    """stop_stream(cancel_requests=True) returns promptly with a slow
+    decoupled request still in flight."""
+    import queue
+    import time
+    import numpy as np
+
+    import client_amd.grpc as grpcclient
+
+    host, port, _ = grpc_fixture_server
+    client = grpcclient.InferenceServerClient(f"{host}:{port}")
+    try:
+        events = queue.Queue()
+        client.start_stream(
+            callback=lambda result, error: events.put((result, error)))
+        inputs = [
+            grpcclient.InferInput("IN", [4], "INT32"),
+            grpcclient.InferInput("DELAY", [4], "UINT32"),
+        ]
+        inputs[0].set_data_from_numpy(np.arange(4, dtype=np.int32))
+        inputs[1].set_data_from_numpy(
+            np.full(4, 800, dtype=np.uint32))  # 4 x 800ms of responses
+        client.async_stream_infer("repeat_int32", inputs)
+        time.sleep(0.2)
+        t0 = time.monotonic()
+        client.stop_stream(cancel_requests=True)
+        assert time.monotonic() - t0 < 3.0, "cancel did not cut the stream"
+        assert client.is_server_live()
+    finally:
+        client.close()

@@ -313,3 +313,24 @@ def test_orca_load_metrics_header():
         client.close()
     finally:
         stop()
+
+
+def test_network_timeout(http_fixture_server):
+    """Client-side network_timeout on a response the server delays past
+    it maps to an InferenceServerException (timeout taxonomy)."""
+    import numpy as np
+
+    host, port, _ = http_fixture_server
+    slow = httpclient.InferenceServerClient(
+        f"{host}:{port}", network_timeout=0.3)
+    try:
+        inputs = [
+            httpclient.InferInput("IN", [1], "INT32"),
+            httpclient.InferInput("DELAY", [1], "UINT32"),
+        ]
+        inputs[0].set_data_from_numpy(np.array([1], dtype=np.int32))
+        inputs[1].set_data_from_numpy(np.array([2000], dtype=np.uint32))
+        with pytest.raises(InferenceServerException):
+            slow.infer("repeat_int32", inputs)
+    finally:
+        slow.close()
