@@ -416,3 +416,50 @@ def test_stop_stream_cancel_requests(grpc_fixture_server):
         assert client.is_server_live()
     finally:
         client.close()
+
+
+def test_string_sequence_id_and_query_params(grpc_fixture_server,
+                                             http_fixture_server):
+    """String sequence ids ride the InferParameter oneof (gRPC) and the
+    JSON parameters (HTTP); query_params must not break HTTP URLs."""
+    import numpy as np
+
+    import client_amd.grpc as grpcclient
+    import client_amd.http as httpclient
+
+    ghost, gport, _ = grpc_fixture_server
+    gc = grpcclient.InferenceServerClient(f"{ghost}:{gport}")
+    try:
+        total = 0
+        for i, v in enumerate([5, 3]):
+            inp = grpcclient.InferInput("INPUT", [1], "INT32")
+            inp.set_data_from_numpy(np.array([v], dtype=np.int32))
+            result = gc.infer(
+                "sequence_accumulate", [inp], sequence_id="stream-A",
+                sequence_start=(i == 0), sequence_end=(i == 1),
+            )
+            total += v
+            assert int(result.as_numpy("OUTPUT")[0]) == total
+    finally:
+        gc.close()
+
+    hhost, hport, _ = http_fixture_server
+    hc = httpclient.InferenceServerClient(f"{hhost}:{hport}")
+    try:
+        inp = httpclient.InferInput("INPUT0", [1, 16], "INT32")
+        inp.set_data_from_numpy(np.ones((1, 16), dtype=np.int32))
+        result = hc.infer(
+            "simple", [inp, _second_ones(httpclient)],
+            query_params={"trace": "1", "k": "v"},
+        )
+        assert result.as_numpy("OUTPUT0").sum() == 32
+    finally:
+        hc.close()
+
+
+def _second_ones(mod):
+    import numpy as np
+
+    inp = mod.InferInput("INPUT1", [1, 16], "INT32")
+    inp.set_data_from_numpy(np.ones((1, 16), dtype=np.int32))
+    return inp
