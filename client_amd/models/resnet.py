@@ -74,5 +74,29 @@ class ResNet50(nn.Module):
         return self.fc(x)
 
 
-def resnet50():
-    return ResNet50()
+def fold_batchnorm(module):
+    """Fold every Conv2d -> BatchNorm2d pair into the conv weights
+    (algebraically exact in eval mode: W' = W*g/sqrt(v+eps),
+    b' = b0*g/sqrt(v+eps) + beta - g*mean/sqrt(v+eps)). The serving
+    profile showed BN kernels at ~29% of ResNet50 kernel time
+    (profiles/serving_rocprof_r01.txt); folding removes them entirely.
+    Pairs are detected by _modules adjacency (conv immediately followed
+    by its BN, which matches this file and torchvision layouts); the BN
+    slot is replaced with Identity so forwards run unchanged."""
+    from torch.nn.utils.fusion import fuse_conv_bn_eval
+
+    for m in list(module.modules()):
+        names = list(m._modules.keys())
+        for a, b in zip(names, names[1:]):
+            conv, bn = m._modules[a], m._modules[b]
+            if isinstance(conv, nn.Conv2d) and isinstance(bn, nn.BatchNorm2d):
+                m._modules[a] = fuse_conv_bn_eval(conv.eval(), bn.eval())
+                m._modules[b] = nn.Identity()
+    return module
+
+
+def resnet50(fold_bn=True):
+    model = ResNet50()
+    if fold_bn:
+        model = fold_batchnorm(model.eval())
+    return model
