@@ -6,6 +6,8 @@ this image, so the standard architecture is defined here directly
 (random-init weights; the benchmark uses synthetic inputs).
 """
 
+import os
+
 import torch.nn as nn
 
 
@@ -74,6 +76,38 @@ class ResNet50(nn.Module):
         return self.fc(x)
 
 
+class BiasAct(nn.Module):
+    """Per-channel bias (+optional ReLU) as one CDNA4 kernel pass.
+
+    torch-on-ROCm emits a separate elementwise kernel for conv bias
+    (miopen_convolution has no bias epilogue), so after BN folding the
+    bias costs a full extra memory pass. Round-2 lever, gated by
+    CLIENT_AMD_FUSED_BIAS=1 (kernel numerics are GPU-tested; default
+    path keeps bias on the conv)."""
+
+    def __init__(self, bias, relu=False):
+        super().__init__()
+        self.register_buffer("bias", bias.detach().float())
+        self.relu_flag = relu
+
+    def forward(self, x):
+        import torch
+
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and x.is_contiguous()):
+            from ..ops import hip_runtime as hr
+
+            n, c = x.shape[0], x.shape[1]
+            plane = x.numel() // (n * c)
+            hr.bias_act_bf16(
+                x.data_ptr(), self.bias.data_ptr(), x.data_ptr(),
+                n * c, plane, c, self.relu_flag,
+                torch.cuda.current_stream().cuda_stream)
+            return x
+        out = x + self.bias.view(1, -1, *([1] * (x.dim() - 2))).to(x.dtype)
+        return out.relu_() if self.relu_flag else out
+
+
 def fold_batchnorm(module):
     """Fold every Conv2d -> BatchNorm2d pair into the conv weights
     (algebraically exact in eval mode: W' = W*g/sqrt(v+eps),
@@ -90,8 +124,17 @@ def fold_batchnorm(module):
         for a, b in zip(names, names[1:]):
             conv, bn = m._modules[a], m._modules[b]
             if isinstance(conv, nn.Conv2d) and isinstance(bn, nn.BatchNorm2d):
-                m._modules[a] = fuse_conv_bn_eval(conv.eval(), bn.eval())
-                m._modules[b] = nn.Identity()
+                fused = fuse_conv_bn_eval(conv.eval(), bn.eval())
+                if os.environ.get("CLIENT_AMD_FUSED_BIAS") == "1":
+                    # bias as one fused kernel pass instead of torch's
+                    # separate elementwise add (see BiasAct)
+                    bias = fused.bias.detach().clone()
+                    fused.bias = None
+                    m._modules[a] = fused
+                    m._modules[b] = BiasAct(bias, relu=False)
+                else:
+                    m._modules[a] = fused
+                    m._modules[b] = nn.Identity()
     return module
 
 

@@ -272,6 +272,16 @@ static void gather_pack(uintptr_t src, uintptr_t dst, int elem_size,
   }
 }
 
+static void bias_act_bf16(uintptr_t x, uintptr_t bias, uintptr_t out,
+                          long n_planes, long plane, int channels,
+                          bool relu, uintptr_t stream_handle) {
+  HIP_CHECK(ca_bias_act_bf16(reinterpret_cast<const void*>(x),
+                             reinterpret_cast<const void*>(bias),
+                             reinterpret_cast<void*>(out), n_planes, plane,
+                             channels, relu ? 1 : 0,
+                             reinterpret_cast<hipStream_t>(stream_handle)));
+}
+
 static void rmsnorm_bf16(uintptr_t x, uintptr_t w, uintptr_t out,
                          long rows, int dim, double eps,
                          uintptr_t stream_handle) {
@@ -345,6 +355,9 @@ PYBIND11_MODULE(_hip_c, m) {
   m.def("cast_fp8e4m3_fp32", &cast_fp8e4m3_fp32, py::arg("src"), py::arg("dst"),
         py::arg("n"), py::arg("device") = 0, py::arg("sync") = true,
         py::arg("stream_idx") = 0);
+  m.def("bias_act_bf16", &bias_act_bf16, py::arg("x"), py::arg("bias"),
+        py::arg("out"), py::arg("n_planes"), py::arg("plane"),
+        py::arg("channels"), py::arg("relu"), py::arg("stream_handle"));
   m.def("rmsnorm_bf16", &rmsnorm_bf16, py::arg("x"), py::arg("w"),
         py::arg("out"), py::arg("rows"), py::arg("dim"), py::arg("eps"),
         py::arg("stream_handle"));

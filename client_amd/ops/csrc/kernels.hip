@@ -554,6 +554,62 @@ extern "C" hipError_t ca_image_preprocess(const void* src, void* dst, int ih,
   return hipGetLastError();
 }
 
+// Per-channel bias add over an NCHW bf16 tensor, in-place capable —
+// absorbs the separate elementwise bias pass torch-on-ROCm emits for
+// conv biases created by BatchNorm folding (55 ms / 5247 calls in the
+// ResNet50 serving profile). One block per (n,c) plane: the fp32 bias
+// loads once, elements stream 8-wide with a scalar tail so any plane
+// size (incl. 7x7=49) is handled. Optional fused ReLU.
+extern "C" __global__ void bias_act_bf16_kernel(
+    const uint16_t* __restrict__ x, const float* __restrict__ bias,
+    uint16_t* __restrict__ out, long plane, int channels, int do_relu) {
+  const long pidx = blockIdx.x;            // n*C + c
+  const int c = (int)(pidx % channels);
+  const float b = bias[c];
+  const uint16_t* xp = x + pidx * plane;
+  uint16_t* op = out + pidx * plane;
+  const long vec_n = plane / 8;
+
+  auto apply = [&](uint16_t v) -> uint16_t {
+    float f = __uint_as_float(((uint32_t)v) << 16) + b;
+    if (do_relu && f < 0.f) f = 0.f;
+    // round-to-nearest-even bf16 (matches torch float->bf16 casts)
+    uint32_t u = __float_as_uint(f);
+    u += 0x7FFF + ((u >> 16) & 1);
+    return (uint16_t)(u >> 16);
+  };
+
+  const uint4* xv = (const uint4*)xp;
+  uint4* ov = (uint4*)op;
+  for (long i = threadIdx.x; i < vec_n; i += blockDim.x) {
+    uint4 v = xv[i];
+    uint32_t* w = (uint32_t*)&v;
+    for (int j = 0; j < 4; ++j) {
+      uint16_t lo = apply((uint16_t)(w[j] & 0xFFFF));
+      uint16_t hi = apply((uint16_t)(w[j] >> 16));
+      w[j] = (uint32_t)lo | ((uint32_t)hi << 16);
+    }
+    ov[i] = v;
+  }
+  for (long i = vec_n * 8 + threadIdx.x; i < plane; i += blockDim.x) {
+    op[i] = apply(xp[i]);
+  }
+}
+
+extern "C" hipError_t ca_bias_act_bf16(const void* x, const void* bias,
+                                       void* out, long n_planes, long plane,
+                                       int channels, int do_relu,
+                                       hipStream_t stream) {
+  if (((uintptr_t)x & 15) || ((uintptr_t)out & 15)) {
+    return hipErrorInvalidValue;  // 16B alignment for the uint4 path
+  }
+  hipLaunchKernelGGL(bias_act_bf16_kernel, dim3((uint32_t)n_planes),
+                     dim3(256), 0, stream, (const uint16_t*)x,
+                     (const float*)bias, (uint16_t*)out, plane, channels,
+                     do_relu);
+  return hipGetLastError();
+}
+
 extern "C" hipError_t ca_rmsnorm_bf16(const void* x, const void* w, void* out,
                                       long rows, int dim, float eps,
                                       hipStream_t stream) {

@@ -429,3 +429,29 @@ def test_region_lifecycle_no_leak(hipshm):
     free_after, _ = hr.mem_info(0)
     leaked = free_before - free_after
     assert leaked < 64 * 2**20, f"leaked {leaked/2**20:.1f} MiB over 200 cycles"
+
+
+@pytest.mark.gpu
+def test_bias_act_kernel_numerics(hipshm):
+    """bias_act_bf16 vs torch reference across plane sizes incl. the
+    odd 7x7=49 tail path, with and without ReLU, in-place."""
+    import torch
+
+    from client_amd.ops import hip_runtime as hr
+
+    torch.manual_seed(3)
+    for (c, h, w) in [(64, 56, 56), (256, 7, 7), (32, 14, 14), (8, 3, 5)]:
+        x = torch.randn(4, c, h, w, device="cuda", dtype=torch.bfloat16)
+        bias = torch.randn(c, device="cuda", dtype=torch.float32)
+        for relu in (False, True):
+            ref = x.float() + bias.view(1, -1, 1, 1)
+            if relu:
+                ref = ref.relu()
+            ref = ref.to(torch.bfloat16)
+            y = x.clone()
+            hr.bias_act_bf16(
+                y.data_ptr(), bias.data_ptr(), y.data_ptr(),
+                4 * c, h * w, c, relu,
+                torch.cuda.current_stream().cuda_stream)
+            torch.cuda.synchronize()
+            assert torch.equal(y, ref), (c, h, w, relu)
