@@ -97,6 +97,12 @@ class BiasAct(nn.Module):
                 and x.is_contiguous()):
             from ..ops import hip_runtime as hr
 
+            if self.bias.dtype != torch.float32:
+                # TorchModel's .to(bf16) sweeps buffers too; the kernel
+                # reads fp32 bias — normalize once (pre-capture, the
+                # eager warmup hits this before any hipGraph capture)
+                self.bias.data = self.bias.data.float()
+
             n, c = x.shape[0], x.shape[1]
             plane = x.numel() // (n * c)
             hr.bias_act_bf16(
