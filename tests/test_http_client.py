@@ -334,3 +334,36 @@ def test_network_timeout(http_fixture_server):
             slow.infer("repeat_int32", inputs)
     finally:
         slow.close()
+
+
+def test_parse_response_body_roundtrip(client, http_fixture_server):
+    """generate_request_body -> raw POST -> parse_response_body: the
+    stateless out-of-band pair (the perf_analyzer hook surface)."""
+    import http.client as stdhttp
+
+    import numpy as np
+
+    host, port, _ = http_fixture_server
+    x = np.arange(16, dtype=np.float32).reshape(1, 16)
+    inp = httpclient.InferInput("INPUT0", [1, 16], "FP32")
+    inp.set_data_from_numpy(x)
+    body, json_size = httpclient.InferenceServerClient.generate_request_body(
+        [inp])
+
+    conn = stdhttp.HTTPConnection(host, port, timeout=30)
+    try:
+        conn.request(
+            "POST", "/v2/models/identity_fp32/infer", body=body,
+            headers={"Inference-Header-Content-Length": str(json_size),
+                     "Content-Type": "application/octet-stream"},
+        )
+        resp = conn.getresponse()
+        assert resp.status == 200
+        header_len = resp.headers.get("Inference-Header-Content-Length")
+        raw = resp.read()
+    finally:
+        conn.close()
+
+    result = httpclient.InferenceServerClient.parse_response_body(
+        raw, header_length=int(header_len) if header_len else None)
+    np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
