@@ -367,3 +367,16 @@ def test_parse_response_body_roundtrip(client, http_fixture_server):
     result = httpclient.InferenceServerClient.parse_response_body(
         raw, header_length=int(header_len) if header_len else None)
     np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
+
+
+def test_bf16_json_rejected():
+    """FP16/BF16 must ride binary framing — JSON path raises (reference
+    _infer_input.py rule)."""
+    import numpy as np
+
+    inp = httpclient.InferInput("X", [2], "BF16")
+    with pytest.raises(InferenceServerException):
+        inp.set_data_from_numpy(np.zeros(2, np.float32), binary_data=False)
+    inp16 = httpclient.InferInput("X", [2], "FP16")
+    with pytest.raises(InferenceServerException):
+        inp16.set_data_from_numpy(np.zeros(2, np.float16), binary_data=False)
