@@ -79,9 +79,10 @@ class InferInput:
         self._parameters.pop("shared_memory_offset", None)
 
         if not binary_data:
-            if self._datatype == "BF16":
+            if self._datatype in ("BF16", "FP16"):
                 raise_error(
-                    "BF16 inputs must be sent as binary data over HTTP. Please "
+                    f"{self._datatype} inputs must be sent as binary data "
+                    "over HTTP. Please "
                     "set binary_data=True"
                 )
             self._parameters.pop("binary_data_size", None)
