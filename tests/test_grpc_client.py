@@ -463,3 +463,24 @@ def _second_ones(mod):
     inp = mod.InferInput("INPUT1", [1, 16], "INT32")
     inp.set_data_from_numpy(np.ones((1, 16), dtype=np.int32))
     return inp
+
+
+def test_grpc_compression(grpc_fixture_server):
+    """gRPC channel compression (gzip/deflate enums) roundtrips."""
+    import numpy as np
+
+    import client_amd.grpc as grpcclient
+
+    host, port, _ = grpc_fixture_server
+    client = grpcclient.InferenceServerClient(f"{host}:{port}")
+    try:
+        x = np.random.rand(1, 1024).astype(np.float32)
+        inp = grpcclient.InferInput("INPUT0", list(x.shape), "FP32")
+        inp.set_data_from_numpy(x)
+        for algo in ("gzip", "deflate"):
+            result = client.infer(
+                "identity_fp32", [inp], compression_algorithm=algo
+            )
+            np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
+    finally:
+        client.close()
