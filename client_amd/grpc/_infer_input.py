@@ -39,47 +39,52 @@ class InferInput:
         self._input.shape.extend(shape)
         return self
 
-    def set_data_from_numpy(self, input_tensor):
-        if not isinstance(input_tensor, (np.ndarray,)):
+    def _validate_array(self, input_tensor):
+        """Dtype/shape admission; error strings are compat contract,
+        pinned in tests/test_infer_input_compat.py (reference
+        grpc/_infer_input.py wording)."""
+        if not isinstance(input_tensor, np.ndarray):
             raise_error("input_tensor must be a numpy array")
-        dtype = np_to_triton_dtype(input_tensor.dtype)
-        if self._input.datatype != dtype:
-            if self._input.datatype == "BF16":
-                if input_tensor.dtype not in (np.float16, np.float32):
-                    raise_error(
-                        "got unexpected datatype {} from numpy array, expected "
-                        "float16/float32 for BF16 input".format(dtype)
-                    )
+        got = np_to_triton_dtype(input_tensor.dtype)
+        if got != self._input.datatype:
+            # BF16 has no numpy dtype: fp32 (or fp16) in, truncated on
+            # serialization
+            if self._input.datatype == "BF16" and input_tensor.dtype in (
+                np.float16, np.float32,
+            ):
+                pass
+            elif self._input.datatype == "BF16":
+                raise_error(
+                    "got unexpected datatype {} from numpy array, expected "
+                    "float16/float32 for BF16 input".format(got)
+                )
             else:
                 raise_error(
-                    "got unexpected datatype {} from numpy array, expected {}".format(
-                        dtype, self._input.datatype
-                    )
+                    "got unexpected datatype {} from numpy array, "
+                    "expected {}".format(got, self._input.datatype)
                 )
-        valid_shape = True
-        if len(self._input.shape) != len(input_tensor.shape):
-            valid_shape = False
-        else:
-            for i in range(len(self._input.shape)):
-                if self._input.shape[i] != input_tensor.shape[i]:
-                    valid_shape = False
-        if not valid_shape:
+        if tuple(input_tensor.shape) != tuple(self._input.shape):
             raise_error(
                 "got unexpected numpy array shape [{}], expected [{}]".format(
-                    str(input_tensor.shape)[1:-1], str(list(self._input.shape))[1:-1]
+                    str(input_tensor.shape)[1:-1],
+                    str(list(self._input.shape))[1:-1],
                 )
             )
-        self._input.parameters.pop("shared_memory_region", None)
-        self._input.parameters.pop("shared_memory_byte_size", None)
-        self._input.parameters.pop("shared_memory_offset", None)
-        if self._input.datatype == "BYTES":
-            serialized_output = serialize_byte_tensor(input_tensor)
-            if serialized_output.size > 0:
-                self._raw_content = serialized_output.item()
-            else:
-                self._raw_content = b""
-        elif self._input.datatype == "BF16":
-            self._raw_content = serialize_bf16_tensor(input_tensor).tobytes()
+
+    def set_data_from_numpy(self, input_tensor):
+        """Serialize a numpy array into this input's raw wire bytes
+        (rides in ModelInferRequest.raw_input_contents)."""
+        self._validate_array(input_tensor)
+        # numpy data replaces any shared-memory binding
+        for key in ("shared_memory_region", "shared_memory_byte_size",
+                    "shared_memory_offset"):
+            self._input.parameters.pop(key, None)
+        dt = self._input.datatype
+        if dt in ("BYTES", "BF16"):
+            serializer = (serialize_byte_tensor if dt == "BYTES"
+                          else serialize_bf16_tensor)
+            packed = serializer(input_tensor)
+            self._raw_content = packed.item() if packed.size else b""
         else:
             self._raw_content = np.ascontiguousarray(input_tensor).tobytes()
         return self

@@ -36,96 +36,98 @@ class InferInput:
         self._shape = list(shape)
         return self
 
-    def set_data_from_numpy(self, input_tensor, binary_data=True):
-        """Attach tensor data from a numpy array.
-
-        binary_data=True sends the raw little-endian bytes after the JSON
-        header; False inlines the values into the JSON ``data`` field
-        (BF16 must be binary — reference _infer_input.py:106-242).
-        """
-        if not isinstance(input_tensor, (np.ndarray,)):
+    def _validate_array(self, input_tensor):
+        """Dtype/shape admission for set_data_from_numpy. The error
+        strings are part of the compat contract (user code matches on
+        them); tests/test_infer_input_compat.py pins them against the
+        reference wording (reference http/_infer_input.py:130-160)."""
+        if not isinstance(input_tensor, np.ndarray):
             raise_error("input_tensor must be a numpy array")
-
-        dtype = np_to_triton_dtype(input_tensor.dtype)
-        if self._datatype != dtype:
-            if self._datatype == "BF16":
-                if input_tensor.dtype not in (np.float16, np.float32):
-                    raise_error(
-                        "got unexpected datatype {} from numpy array, expected "
-                        "{} for BF16 input".format(dtype, np.float32)
-                    )
+        got = np_to_triton_dtype(input_tensor.dtype)
+        if got != self._datatype:
+            # a BF16 input has no numpy dtype; it is fed as fp32 (or
+            # fp16) and truncated on serialization
+            if self._datatype == "BF16" and input_tensor.dtype in (
+                np.float16, np.float32,
+            ):
+                pass
+            elif self._datatype == "BF16":
+                raise_error(
+                    "got unexpected datatype {} from numpy array, expected "
+                    "{} for BF16 input".format(got, np.float32)
+                )
             else:
                 raise_error(
-                    "got unexpected datatype {} from numpy array, expected {}".format(
-                        dtype, self._datatype
-                    )
+                    "got unexpected datatype {} from numpy array, "
+                    "expected {}".format(got, self._datatype)
                 )
-        valid_shape = True
-        if len(self._shape) != len(input_tensor.shape):
-            valid_shape = False
-        else:
-            for i in range(len(self._shape)):
-                if self._shape[i] != input_tensor.shape[i]:
-                    valid_shape = False
-        if not valid_shape:
+        if tuple(input_tensor.shape) != tuple(self._shape):
             raise_error(
                 "got unexpected numpy array shape [{}], expected [{}]".format(
                     str(input_tensor.shape)[1:-1], str(self._shape)[1:-1]
                 )
             )
 
-        self._parameters.pop("shared_memory_region", None)
-        self._parameters.pop("shared_memory_byte_size", None)
-        self._parameters.pop("shared_memory_offset", None)
-
-        if not binary_data:
-            if self._datatype in ("BF16", "FP16"):
-                raise_error(
-                    f"{self._datatype} inputs must be sent as binary data "
-                    "over HTTP. Please "
-                    "set binary_data=True"
-                )
-            self._parameters.pop("binary_data_size", None)
-            self._raw_data = None
-            if self._datatype == "BYTES":
-                self._data = []
+    @staticmethod
+    def _bytes_elements_as_json(input_tensor):
+        """Flatten a BYTES tensor to a list of JSON-safe strings:
+        bytes elements are UTF-8 decoded, anything else is str()'d."""
+        if input_tensor.size == 0:
+            return []
+        out = []
+        for elem in input_tensor.reshape(-1, order="C").tolist():
+            if isinstance(elem, bytes):
                 try:
-                    if input_tensor.size > 0:
-                        for obj in np.nditer(
-                            input_tensor, flags=["refs_ok"], order="C"
-                        ):
-                            # We need to convert the object to string using
-                            # utf-8 codec.
-                            if input_tensor.dtype == np.object_:
-                                if type(obj.item()) == bytes:
-                                    self._data.append(str(obj.item(), encoding="utf-8"))
-                                else:
-                                    self._data.append(str(obj.item()))
-                            else:
-                                self._data.append(str(obj.item(), encoding="utf-8"))
+                    out.append(elem.decode("utf-8"))
                 except UnicodeDecodeError:
                     raise_error(
                         'Failed to encode "{}" using UTF-8. Please use '
-                        "binary_data=True, if you want to pass a byte array.".format(
-                            obj.item()
-                        )
+                        "binary_data=True, if you want to pass a byte "
+                        "array.".format(elem)
                     )
             else:
-                self._data = [val.item() for val in input_tensor.flatten()]
-        else:
+                out.append(str(elem))
+        return out
+
+    def set_data_from_numpy(self, input_tensor, binary_data=True):
+        """Attach tensor data from a numpy array.
+
+        binary_data=True sends the raw little-endian bytes after the
+        JSON header; False inlines the values into the JSON ``data``
+        field. Only BF16 is binary-only (it has no JSON representation;
+        FP16 inlines fine — reference _infer_input.py:169-171 restricts
+        exactly BF16).
+        """
+        self._validate_array(input_tensor)
+
+        # numpy data replaces any shared-memory binding
+        for key in ("shared_memory_region", "shared_memory_byte_size",
+                    "shared_memory_offset"):
+            self._parameters.pop(key, None)
+
+        if binary_data:
             self._data = None
             if self._datatype == "BYTES":
-                serialized_output = serialize_byte_tensor(input_tensor)
-                if serialized_output.size > 0:
-                    self._raw_data = serialized_output.item()
-                else:
-                    self._raw_data = b""
+                packed = serialize_byte_tensor(input_tensor)
+                self._raw_data = packed.item() if packed.size else b""
             elif self._datatype == "BF16":
-                serialized_output = serialize_bf16_tensor(input_tensor)
-                self._raw_data = serialized_output.tobytes()
+                packed = serialize_bf16_tensor(input_tensor)
+                self._raw_data = packed.item() if packed.size else b""
             else:
                 self._raw_data = input_tensor.tobytes()
             self._parameters["binary_data_size"] = len(self._raw_data)
+        else:
+            if self._datatype == "BF16":
+                raise_error(
+                    "BF16 inputs must be sent as binary data over HTTP. "
+                    "Please set binary_data=True"
+                )
+            self._raw_data = None
+            self._parameters.pop("binary_data_size", None)
+            if self._datatype == "BYTES":
+                self._data = self._bytes_elements_as_json(input_tensor)
+            else:
+                self._data = input_tensor.reshape(-1, order="C").tolist()
         return self
 
     def set_shared_memory(self, region_name, byte_size, offset=0):

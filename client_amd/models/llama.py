@@ -158,19 +158,34 @@ class LlamaModel(nn.Module):
 
     def _get_rope(self, device):
         if self._rope is None or self._rope[0].device != device:
+            # +1 row so scratch_pos is a valid RoPE index for inactive
+            # decode rows (their output is discarded; only indexability
+            # matters)
             self._rope = precompute_rope(
-                self.cfg.dim // self.cfg.n_heads, self.cfg.max_seq,
+                self.cfg.dim // self.cfg.n_heads, self.cfg.max_seq + 1,
                 self.cfg.rope_theta, device,
             )
         return self._rope
+
+    # position index reserved for rows that must not write real cache
+    # state during a batched decode step (FREE / mid-prefill slots):
+    # make_kv_cache allocates one position past max_seq and
+    # forward_decode_batch only ever reads [:max_len <= max_seq], so a
+    # scatter at scratch_pos is invisible — and shapes stay static for
+    # hipGraph capture (the fix for the mid-prefill cache-corruption
+    # bug: a decode step used to overwrite position 1 of rows still in
+    # chunked prefill).
+    @property
+    def scratch_pos(self):
+        return self.cfg.max_seq
 
     def make_kv_cache(self, batch, device, dtype):
         head_dim = self.cfg.dim // self.cfg.n_heads
         return [
             (
-                torch.zeros(batch, self.cfg.n_kv_heads, self.cfg.max_seq,
+                torch.zeros(batch, self.cfg.n_kv_heads, self.cfg.max_seq + 1,
                             head_dim, device=device, dtype=dtype),
-                torch.zeros(batch, self.cfg.n_kv_heads, self.cfg.max_seq,
+                torch.zeros(batch, self.cfg.n_kv_heads, self.cfg.max_seq + 1,
                             head_dim, device=device, dtype=dtype),
             )
             for _ in range(self.cfg.n_layers)

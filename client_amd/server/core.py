@@ -169,6 +169,22 @@ class InferenceError(Exception):
         self.status = status
 
 
+def _check_shm_bounds(region, offset, byte_size):
+    """Reject client-supplied (offset, byte_size) windows that fall
+    outside the registered region before ANY copy touches the mapping —
+    the offsets come off the wire and otherwise index arbitrary server
+    memory (HIP regions share the address space with model weights).
+    Mirrors the reference server's rejection of out-of-range shm access.
+    """
+    if (offset < 0 or byte_size < 0
+            or offset + byte_size > region["byte_size"]):
+        raise InferenceError(
+            "Invalid offset + byte size for shared memory region: "
+            f"'{region['name']}' (offset {offset} + byte size {byte_size} "
+            f"exceeds registered size {region['byte_size']})"
+        )
+
+
 class InferenceCore:
     """Executes KServe-v2 inference requests against the model repository."""
 
@@ -255,6 +271,7 @@ class InferenceCore:
             offset = params.get("shared_memory_offset", 0)
             region = self.shm.get_system(shm_name)
             if region is not None:
+                _check_shm_bounds(region, offset, byte_size)
                 base = region["offset"] + offset
                 raw = bytes(region["mmap"][base : base + byte_size])
                 return self._decode_raw(raw, datatype, shape), binary_cursor
@@ -262,6 +279,7 @@ class InferenceCore:
             if hip_region is not None:
                 from ..ops import hip_runtime as hr
 
+                _check_shm_bounds(hip_region, offset, byte_size)
                 raw = hr.memcpy_d2h(
                     hip_region["ptr"] + offset, byte_size, hip_region["device_id"]
                 )
@@ -315,7 +333,8 @@ class InferenceCore:
             s = serialize_byte_tensor(arr)
             return s.item() if s.size > 0 else b""
         if datatype == "BF16":
-            return serialize_bf16_tensor(arr).tobytes()
+            s = serialize_bf16_tensor(arr)
+            return s.item() if s.size > 0 else b""
         return np.ascontiguousarray(arr).tobytes()
 
     # ---- inference ----
@@ -397,6 +416,16 @@ class InferenceCore:
             params = io["parameters"]
             region = self.shm.get_hip(params["shared_memory_region"])
             offset = params.get("shared_memory_offset", 0)
+            # bound BOTH the declared window and the actual tensor extent
+            # the DLPack view will touch
+            itemsize = 2 if datatype == "BF16" else np.dtype(
+                triton_to_np_dtype(datatype)
+            ).itemsize  # BF16 maps to fp32 in the numpy table; wire is 2B
+            nbytes = int(np.prod(shape)) * itemsize
+            _check_shm_bounds(
+                region, offset,
+                max(nbytes, params.get("shared_memory_byte_size", 0)),
+            )
             smt = SharedMemoryTensor(
                 datatype=datatype,
                 shape=shape,
@@ -511,6 +540,8 @@ class InferenceCore:
                     )
                 region = self.shm.get_system(shm_name)
                 if region is not None:
+                    _check_shm_bounds(region, offset, len(raw))
+                    _check_shm_bounds(region, offset, byte_size)
                     base = region["offset"] + offset
                     region["mmap"][base : base + len(raw)] = raw
                 else:
@@ -521,6 +552,8 @@ class InferenceCore:
                         )
                     from ..ops import hip_runtime as hr
 
+                    _check_shm_bounds(hip_region, offset, len(raw))
+                    _check_shm_bounds(hip_region, offset, byte_size)
                     hr.memcpy_h2d(
                         hip_region["ptr"] + offset, raw, len(raw),
                         hip_region["device_id"],

@@ -65,8 +65,12 @@ class DecodeScheduler:
         self.prefill_chunk = prefill_chunk
         self._tokens_dev = torch.zeros(max_batch, 1, dtype=torch.int64,
                                        device=device)
-        self._pos_dev = torch.ones(max_batch, dtype=torch.int64,
-                                   device=device)
+        # inactive rows scatter their (garbage) K/V at the model's
+        # reserved scratch position, which decode never reads — NOT at
+        # position 1, which would corrupt rows still in chunked prefill
+        self._scratch = model.scratch_pos
+        self._pos_dev = torch.full((max_batch,), self._scratch,
+                                   dtype=torch.int64, device=device)
         self._graphs = {}  # bucket -> (graph, next_tokens_out)
         self._pending = queue.Queue()
         self._cv = threading.Condition()
@@ -213,7 +217,8 @@ class DecodeScheduler:
             device=self.device,
         )[:, None]
         pos_rows = torch.tensor(
-            [max(s.pos, 1) if s.active else 1 for s in self.slots],
+            [max(s.pos, 1) if s.active else self._scratch
+             for s in self.slots],
             dtype=torch.int64, device=self.device,
         )
         if self.use_graph:

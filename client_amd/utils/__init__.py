@@ -196,16 +196,19 @@ def serialize_bf16_tensor(input_tensor):
     Semantics match the reference (utils/__init__.py:294-330): each fp32
     element's upper 16 bits are kept (truncation, no rounding). The
     reference loops per element in Python; here it is one vectorized
-    numpy view+shift. Returns a 1-D np.uint8 array.
+    numpy view+shift. Return type matches the reference for drop-in
+    compatibility: a 0-d np.object_ array holding the wire bytes
+    (callers do ``.item()``), np.empty([0], object_) when empty.
+    fp16 input is accepted as a superset (upcast to fp32 first).
     """
     if input_tensor.size == 0:
-        return np.empty([0], dtype=np.uint8)
+        return np.empty([0], dtype=np.object_)
     if input_tensor.dtype not in (np.float16, np.float32):
         raise_error("cannot serialize bf16 tensor: invalid datatype")
     f32 = np.ascontiguousarray(input_tensor, dtype="<f4")
     u32 = f32.view("<u4").reshape(-1)
     u16 = (u32 >> np.uint32(16)).astype("<u2")
-    return u16.view(np.uint8)
+    return np.asarray(u16.tobytes(), dtype=np.object_)
 
 
 def deserialize_bf16_tensor(encoded_tensor):

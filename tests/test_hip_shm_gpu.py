@@ -81,11 +81,11 @@ def test_cast_bf16_wire_exact(hipshm):
         nbytes = hipshm.set_shared_memory_region_cast(h, x, "BF16")
         assert nbytes == n * 2
         raw = hipshm.get_contents_as_numpy(h, np.uint8, [n * 2])
-        expected = serialize_bf16_tensor(x)
-        np.testing.assert_array_equal(raw, expected)
+        wire = serialize_bf16_tensor(x).item()
+        np.testing.assert_array_equal(raw, np.frombuffer(wire, np.uint8))
         # device unpack matches CPU deserialize
         back = hipshm.get_contents_cast(h, "BF16", [n])
-        np.testing.assert_array_equal(back, deserialize_bf16_tensor(expected))
+        np.testing.assert_array_equal(back, deserialize_bf16_tensor(wire))
     finally:
         hipshm.destroy_shared_memory_region(h)
 
@@ -99,8 +99,11 @@ def test_cast_bf16_odd_sizes(hipshm):
         try:
             hipshm.set_shared_memory_region_cast(h, x, "BF16")
             raw = hipshm.get_contents_as_numpy(h, np.uint8, [n * 2])
-            np.testing.assert_array_equal(raw, serialize_bf16_tensor(x),
-                                          err_msg=f"n={n}")
+            np.testing.assert_array_equal(
+                raw,
+                np.frombuffer(serialize_bf16_tensor(x).item(), np.uint8),
+                err_msg=f"n={n}",
+            )
         finally:
             hipshm.destroy_shared_memory_region(h)
 
@@ -306,7 +309,7 @@ def test_rccl_broadcast_region_tensor(hipshm):
             )
 
             np.testing.assert_array_equal(
-                back, deserialize_bf16_tensor(serialize_bf16_tensor(x).tobytes())
+                back, deserialize_bf16_tensor(serialize_bf16_tensor(x).item())
             )
         finally:
             hipshm.destroy_shared_memory_region(h)

@@ -369,14 +369,18 @@ def test_parse_response_body_roundtrip(client, http_fixture_server):
     np.testing.assert_array_equal(result.as_numpy("OUTPUT0"), x)
 
 
-def test_bf16_json_rejected():
-    """FP16/BF16 must ride binary framing — JSON path raises (reference
-    _infer_input.py rule)."""
+def test_bf16_json_rejected_fp16_allowed():
+    """Only BF16 is binary-only over HTTP (it has no JSON number form);
+    FP16 inlines into JSON like any float — the reference restricts
+    exactly BF16 (reference http/_infer_input.py:169-171)."""
     import numpy as np
 
     inp = httpclient.InferInput("X", [2], "BF16")
     with pytest.raises(InferenceServerException):
         inp.set_data_from_numpy(np.zeros(2, np.float32), binary_data=False)
     inp16 = httpclient.InferInput("X", [2], "FP16")
-    with pytest.raises(InferenceServerException):
-        inp16.set_data_from_numpy(np.zeros(2, np.float16), binary_data=False)
+    inp16.set_data_from_numpy(
+        np.array([1.5, -2.0], np.float16), binary_data=False
+    )
+    assert inp16._get_tensor()["data"] == [1.5, -2.0]
+    assert inp16._get_binary_data() is None

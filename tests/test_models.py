@@ -312,6 +312,87 @@ def test_chunked_prefill_matches_full():
         sched.shutdown()
 
 
+def test_mid_prefill_rows_not_corrupted_by_decode():
+    """Regression: a decode step scatters K/V for EVERY batch row
+    (forward_decode_batch), and inactive rows used to be pointed at
+    position 1 — overwriting cache position 1 of rows still in chunked
+    prefill, silently corrupting their generations (diverged in 2/5
+    seeds with this exact shape before the scratch-position fix). A
+    short prompt becomes ACTIVE and decodes while long prompts are
+    mid-prefill; every stream must still match sequential generate,
+    across multiple seeds."""
+    from client_amd.models.llama import LlamaModel, llama_tiny_config
+    from client_amd.server.decode_scheduler import DecodeScheduler
+
+    for seed in range(5):
+        torch.manual_seed(seed)
+        cfg = llama_tiny_config()
+        m = LlamaModel(cfg).eval()
+        # short prompt finishes prefill in one chunk and starts
+        # decoding; 30/45-token prompts stay in PREFILL for several
+        # decode steps (chunk=8)
+        prompts = [torch.randint(0, cfg.vocab_size, (1, n))
+                   for n in (4, 30, 45)]
+        expected = [[int(t[0]) for t in m.generate(p, 8)] for p in prompts]
+
+        sched = DecodeScheduler(m, max_batch=4, device="cpu",
+                                dtype=torch.float32, prefill_chunk=8)
+        try:
+            queues = [sched.submit(p[0].numpy(), 8) for p in prompts]
+            got = []
+            for q in queues:
+                toks = []
+                while True:
+                    t = q.get(timeout=60)
+                    if t is sched.END:
+                        break
+                    toks.append(t)
+                got.append(toks)
+            assert got == expected, f"divergence at seed {seed}"
+        finally:
+            sched.shutdown()
+
+
+def test_shm_offset_bounds_rejected():
+    """Client-controlled shared_memory_offset/byte_size windows outside
+    the registered region must be rejected before any copy (the
+    reference server rejects out-of-range shm access; an unchecked
+    offset would index arbitrary server memory)."""
+    import multiprocessing.shared_memory as mpshm
+
+    from client_amd.server.core import InferenceCore, InferenceError
+
+    core = InferenceCore()
+    seg = mpshm.SharedMemory(create=True, size=64)
+    try:
+        key = "/" + seg.name.lstrip("/")
+        core.shm.register_system("r0", key, 0, 64)
+
+        def infer(offset=0, byte_size=32):
+            req = {
+                "inputs": [{
+                    "name": "IN", "datatype": "FP32", "shape": [8],
+                    "parameters": {
+                        "shared_memory_region": "r0",
+                        "shared_memory_byte_size": byte_size,
+                        "shared_memory_offset": offset,
+                    },
+                }],
+            }
+            core._input_array(req["inputs"][0], b"", 0)
+
+        infer(offset=0, byte_size=32)       # in-bounds: fine
+        infer(offset=32, byte_size=32)      # exactly fits: fine
+        for off, size in ((33, 32), (0, 65), (-1, 32), (0, -1),
+                          (2**40, 32)):
+            with pytest.raises(InferenceError):
+                infer(offset=off, byte_size=size)
+    finally:
+        core.shm.unregister_system()
+        seg.close()
+        seg.unlink()
+
+
 def test_torchmodel_warmup_cpu_noop_and_cli():
     """warmup() is a no-op off-GPU and the --model-warmup CLI path wires
     through build_core without error."""

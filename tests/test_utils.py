@@ -62,27 +62,36 @@ def test_bf16_roundtrip_exact():
     # Values exactly representable in bf16 survive the round trip.
     vals = np.array([1.0, -2.5, 0.0, 0.5, -0.375, 128.0], dtype=np.float32)
     raw = serialize_bf16_tensor(vals)
-    assert raw.nbytes == vals.size * 2
-    back = deserialize_bf16_tensor(raw.tobytes())
+    # reference-compatible return type: 0-d object_ array, .item() gives
+    # the wire bytes (reference utils/__init__.py:294-330 callers)
+    assert raw.dtype == np.object_
+    wire = raw.item()
+    assert isinstance(wire, bytes) and len(wire) == vals.size * 2
+    back = deserialize_bf16_tensor(wire)
     np.testing.assert_array_equal(back, vals)
+
+
+def test_bf16_empty_returns_empty_object_array():
+    raw = serialize_bf16_tensor(np.empty((0,), dtype=np.float32))
+    assert raw.dtype == np.object_ and raw.size == 0
 
 
 def test_bf16_truncation_semantics():
     # Reference semantics: plain truncation of the fp32 low 16 bits
     # (no round-to-nearest-even) — utils/__init__.py:294-330.
     x = np.array([1.0000001], dtype=np.float32)
-    raw = serialize_bf16_tensor(x)
+    wire = serialize_bf16_tensor(x).item()
     expected = x.view(np.uint32) >> 16
-    assert raw.view(np.uint16)[0] == expected[0]
-    back = deserialize_bf16_tensor(raw.tobytes())
+    assert np.frombuffer(wire, dtype=np.uint16)[0] == expected[0]
+    back = deserialize_bf16_tensor(wire)
     # truncation error is bounded by 1 ulp of bf16
     assert abs(back[0] - x[0]) < 2 ** -7
 
 
 def test_bf16_from_fp16():
     x = np.array([1.5, -3.25], dtype=np.float16)
-    raw = serialize_bf16_tensor(x)
-    back = deserialize_bf16_tensor(raw.tobytes())
+    wire = serialize_bf16_tensor(x).item()
+    back = deserialize_bf16_tensor(wire)
     np.testing.assert_allclose(back, x.astype(np.float32), rtol=2 ** -7)
 
 
