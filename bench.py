@@ -36,6 +36,13 @@ OUT_SHAPE = (BATCH, 1000)
 
 def start_server(device_index, port, extra_args=()):
     env = dict(os.environ)
+    # one MIOpen find-db/cache per rank: 8 server processes sharing one
+    # sqlite user-db serialize on its lock during warmup and can blow
+    # the ready window (ROUND2_NOTES item 1)
+    miopen_dir = f"/tmp/miopen_rank{device_index}"
+    os.makedirs(miopen_dir, exist_ok=True)
+    env.setdefault("MIOPEN_USER_DB_PATH", miopen_dir)
+    env.setdefault("MIOPEN_CUSTOM_CACHE_DIR", miopen_dir)
     repo = os.path.dirname(os.path.abspath(__file__))
     log_dir = os.path.join(repo, "gpurun_out")
     os.makedirs(log_dir, exist_ok=True)
@@ -80,6 +87,13 @@ def main():
     ap.add_argument("--reqs-per-step", type=int, default=48)
     ap.add_argument("--no-dynamic-batching", action="store_true")
     ap.add_argument("--preferred-batch-size", type=int, default=32)
+    ap.add_argument("--stagger-s", type=float, default=1.5,
+                    help="per-rank server-spawn stagger (world>1)")
+    ap.add_argument("--fanout", choices=("rccl", "p2p"),
+                    default=os.environ.get("CLIENT_AMD_FANOUT", "rccl"),
+                    help="8-way input replication: RCCL broadcast (tree "
+                         "over xGMI) or 7-link hipMemcpyPeerAsync direct "
+                         "scatter")
     args = ap.parse_args()
 
     import torch
@@ -95,7 +109,8 @@ def main():
 
     torch.cuda.set_device(local_rank)
     from client_amd.parallel import (
-        RegionBroadcaster,
+        OverlappedBroadcaster,
+        PeerScatterBroadcaster,
         aggregate_max,
         init_distributed,
     )
@@ -112,6 +127,10 @@ def main():
         "--preferred-batch-size", str(args.preferred_batch_size),
         "--max-queue-delay-us", "400",
     ]
+    # stagger the 8 server spawns a little so torch/MIOpen cold-start
+    # page-ins don't all land at once (ready window is 300 s)
+    if distributed and args.stagger_s > 0:
+        time.sleep(local_rank * args.stagger_s)
     server_proc, port = start_server(local_rank, 8101 + local_rank,
                                      server_args)
     try:
@@ -155,27 +174,69 @@ def main():
         for slot in slots:
             hipshm.set_shared_memory_region_cast(slot["in"], host_x, "BF16")
 
-        # zero-copy view of the slot-0 input region for the RCCL
-        # broadcast fan-out (client_amd.parallel)
-        broadcaster = (
-            RegionBroadcaster(slots[0]["in"], IN_SHAPE, "BF16", src=0)
-            if distributed else None
-        )
+        # world>1 fan-out: ping-pong staging regions so pack(n+1) into
+        # one buffer overlaps the broadcast of the other on a side
+        # stream (event-ordered; request issue gated on the bcast event
+        # only — no per-step global sync). Step n's requests read
+        # bench_bc{n%2}; step n+1's pack+bcast run during step n's
+        # serving.
+        broadcaster = None
+        pp_inputs = None
+        if distributed:
+            pp_regions = []
+            pp_inputs = []
+            for b in range(2):
+                r = hipshm.create_shared_memory_region(
+                    f"bench_bc{b}", in_elems * 2, local_rank
+                )
+                client.register_cuda_shared_memory(
+                    f"bench_bc{b}", hipshm.get_raw_handle_bytes(r),
+                    local_rank, in_elems * 2,
+                )
+                hipshm.set_shared_memory_region_cast(r, host_x, "BF16")
+                inp = grpcclient.InferInput("INPUT0", list(IN_SHAPE), "BF16")
+                inp.set_shared_memory(f"bench_bc{b}", in_elems * 2)
+                pp_regions.append(r)
+                pp_inputs.append([inp])
+
+            def pack_fn(buf_idx):
+                # enqueues h2d + CDNA4 cast on hip_runtime stream 0;
+                # the broadcaster orders the collective after it by
+                # event
+                hipshm.set_shared_memory_region_cast(
+                    pp_regions[buf_idx], host_x, "BF16", sync=False
+                )
+
+            if args.fanout == "p2p":
+                broadcaster = PeerScatterBroadcaster(
+                    pp_regions,
+                    in_elems * 2,  # per-buffer scatter size
+                    src=0,
+                )
+            else:
+                broadcaster = OverlappedBroadcaster(
+                    pp_regions, IN_SHAPE, "BF16", src=0
+                )
 
         latencies = []
         lat_lock = threading.Lock()
+        step_no = [0]
 
         def run_step(record=False):
-            """pack -> (broadcast) -> reqs_per_step requests at the
-            configured concurrency, closed loop."""
-            hipshm.set_shared_memory_region_cast(
-                slots[0]["in"], host_x, "BF16", sync=not distributed
-            )
+            """(gate on bcast n) -> enqueue pack+bcast n+1 -> serve step
+            n (closed loop, reqs_per_step requests at the configured
+            concurrency)."""
+            n = step_no[0]
+            step_no[0] += 1
             if distributed:
-                # fan the staged input out to every replica over xGMI
-                torch.cuda.synchronize()
-                broadcaster.broadcast()
-                torch.cuda.synchronize()
+                broadcaster.wait_ready()
+                broadcaster.stage_and_broadcast((n + 1) % 2, pack_fn)
+                step_inputs = pp_inputs[n % 2]
+            else:
+                hipshm.set_shared_memory_region_cast(
+                    slots[0]["in"], host_x, "BF16", sync=True
+                )
+                step_inputs = None
             # dispatcher model: the main thread issues every request,
             # bounded by a concurrency semaphore; gRPC completion
             # callbacks only record + release (never issue RPCs from
@@ -196,7 +257,9 @@ def main():
                     sem.release()
 
                 client.async_infer(
-                    "resnet50", slot["inputs"], callback=cb,
+                    "resnet50",
+                    step_inputs if step_inputs is not None else slot["inputs"],
+                    callback=cb,
                     outputs=slot["outputs"],
                 )
 
@@ -214,6 +277,10 @@ def main():
                 sem.release()
             if errors:
                 raise SystemExit(f"infer error: {errors[0]}")
+
+        # prologue: stage+broadcast buffer 0 so step 0 has data
+        if distributed:
+            broadcaster.stage_and_broadcast(0, pack_fn)
 
         # warmup
         for _ in range(args.warmup):
@@ -255,6 +322,26 @@ def main():
                                           list(OUT_SHAPE))
         assert np.all(np.isfinite(logits)), "non-finite output"
 
+        if distributed:
+            fanout_desc = (
+                "RCCL tree bcast over xGMI, side-stream overlapped with "
+                "next step's pack (hipEvent-ordered, no per-step global "
+                "sync)" if args.fanout == "rccl"
+                else "7-link hipMemcpyPeerAsync direct scatter over xGMI, "
+                     "overlapped with serving"
+            )
+            bc = broadcaster.bcast_ms[-args.steps:]
+            bcast_stats = {
+                "mean_ms": round(float(np.mean(bc)), 4) if bc else None,
+                "max_ms": round(float(np.max(bc)), 4) if bc else None,
+                "timed_from": ("hipEvents on the side stream"
+                               if args.fanout == "rccl"
+                               else "host clock around scatter+signal"),
+            }
+        else:
+            fanout_desc = "single GPU (no fan-out executed)"
+            bcast_stats = None
+
         if rank == 0:
             print(json.dumps({
                 "metric": "perf_analyzer inferences/sec + p99 latency, "
@@ -277,7 +364,8 @@ def main():
                     "seq_len": None,
                     "parallelism": f"replicated-serving x{world}, one gRPC "
                                    f"client+server pair per GPU, HIP-IPC shm "
-                                   f"I/O, RCCL bcast fan-out",
+                                   f"I/O, fan-out: {fanout_desc}",
+                    "bcast": bcast_stats,
                     "concurrency": args.concurrency,
                     "reqs_per_step": args.reqs_per_step,
                     "requests_per_sec": round(total_requests / elapsed, 2),
@@ -292,6 +380,9 @@ def main():
         for slot in slots:
             hipshm.destroy_shared_memory_region(slot["in"])
             hipshm.destroy_shared_memory_region(slot["out"])
+        if distributed:
+            for r in pp_regions:
+                hipshm.destroy_shared_memory_region(r)
         client.close()
         if distributed:
             dist.destroy_process_group()

@@ -26,11 +26,27 @@ def _pybind11_includes():
     return [pybind11.get_include()]
 
 
+HASH_SIDECAR = OPS_DIR / "_hip_c.so.srchash"
+
+
+def _source_hash():
+    import hashlib
+
+    h = hashlib.sha256()
+    for p in (SRC, KERNELS):
+        h.update(p.read_bytes())
+    h.update(GFX_ARCH.encode())
+    return h.hexdigest()
+
+
 def needs_build():
-    if not OUT.exists():
+    """True unless the existing .so was provably compiled from the
+    current sources: a sidecar file records the source hash at build
+    time (an mtime check could silently run a stale committed binary
+    on a fresh checkout)."""
+    if not OUT.exists() or not HASH_SIDECAR.exists():
         return True
-    newest = max(SRC.stat().st_mtime, KERNELS.stat().st_mtime)
-    return newest > OUT.stat().st_mtime
+    return HASH_SIDECAR.read_text().strip() != _source_hash()
 
 
 def build(force=False, verbose=True):
@@ -57,6 +73,7 @@ def build(force=False, verbose=True):
     if verbose:
         print("[client_amd.ops.build]", " ".join(cmd), file=sys.stderr)
     subprocess.run(cmd, check=True)
+    HASH_SIDECAR.write_text(_source_hash() + "\n")
     return str(OUT)
 
 

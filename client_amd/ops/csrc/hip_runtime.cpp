@@ -198,6 +198,55 @@ static void stream_sync(int device, int idx) {
   HIP_CHECK(hipStreamSynchronize(s));
 }
 
+static uintptr_t stream_handle(int device, int idx) {
+  // raw hipStream_t of a cached stream, for torch.cuda.ExternalStream
+  // interop (event-ordered overlap of pack kernels with RCCL
+  // collectives — SURVEY.md §2.8 final row)
+  return reinterpret_cast<uintptr_t>(get_stream(device, idx));
+}
+
+static void memcpy_peer_async(uintptr_t dst, int dst_device, uintptr_t src,
+                              int src_device, size_t byte_size,
+                              int stream_idx) {
+  // direct xGMI point-to-point copy; with one call per destination GPU
+  // on its own stream, the 7 copies of an 8-way scatter ride 7 distinct
+  // xGMI links concurrently (ring bcast is per-link
+  // bound; direct scatter is not)
+  hipStream_t s = get_stream(src_device, stream_idx);
+  py::gil_scoped_release release;
+  HIP_CHECK(hipMemcpyPeerAsync(reinterpret_cast<void*>(dst), dst_device,
+                               reinterpret_cast<void*>(src), src_device,
+                               byte_size, s));
+}
+
+static void stream_fence(int device, int from_idx, std::vector<int> to_idxs) {
+  // device-side ordering: record an event on stream from_idx and make
+  // every stream in to_idxs wait on it — no host block. Used to order
+  // the 7-link peer scatter after the pack kernel.
+  hipStream_t from = get_stream(device, from_idx);
+  hipEvent_t ev;
+  HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  HIP_CHECK(hipEventRecord(ev, from));
+  for (int idx : to_idxs) {
+    HIP_CHECK(hipStreamWaitEvent(get_stream(device, idx), ev, 0));
+  }
+  // destruction is deferred by the runtime until the event completes
+  HIP_CHECK(hipEventDestroy(ev));
+}
+
+static void device_enable_peer_access(int device, int peer) {
+  int prev;
+  HIP_CHECK(hipGetDevice(&prev));
+  HIP_CHECK(hipSetDevice(device));
+  hipError_t e = hipDeviceEnablePeerAccess(peer, 0);
+  if (e != hipSuccess && e != hipErrorPeerAccessAlreadyEnabled) {
+    HIP_CHECK(hipSetDevice(prev));
+    throw std::runtime_error(std::string("hipDeviceEnablePeerAccess: ") +
+                             hipGetErrorString(e));
+  }
+  HIP_CHECK(hipSetDevice(prev));
+}
+
 static void device_sync() {
   py::gil_scoped_release release;
   HIP_CHECK(hipDeviceSynchronize());
@@ -342,6 +391,15 @@ PYBIND11_MODULE(_hip_c, m) {
         py::arg("byte_size"), py::arg("device") = 0, py::arg("sync") = true);
   m.def("stream_sync", &stream_sync, py::arg("device") = 0,
         py::arg("stream_idx") = 0);
+  m.def("stream_handle", &stream_handle, py::arg("device") = 0,
+        py::arg("stream_idx") = 0);
+  m.def("memcpy_peer_async", &memcpy_peer_async, py::arg("dst"),
+        py::arg("dst_device"), py::arg("src"), py::arg("src_device"),
+        py::arg("byte_size"), py::arg("stream_idx") = 0);
+  m.def("device_enable_peer_access", &device_enable_peer_access,
+        py::arg("device"), py::arg("peer"));
+  m.def("stream_fence", &stream_fence, py::arg("device"),
+        py::arg("from_idx"), py::arg("to_idxs"));
   m.def("device_sync", &device_sync);
   m.def("cast_fp32_bf16", &cast_fp32_bf16, py::arg("src"), py::arg("dst"),
         py::arg("n"), py::arg("device") = 0, py::arg("sync") = true,
