@@ -331,6 +331,17 @@ static void bias_act_bf16(uintptr_t x, uintptr_t bias, uintptr_t out,
                              reinterpret_cast<hipStream_t>(stream_handle)));
 }
 
+static void bias_res_act_bf16(uintptr_t x, uintptr_t res, uintptr_t bias,
+                              uintptr_t out, long n_planes, long plane,
+                              int channels, bool relu,
+                              uintptr_t stream_handle) {
+  HIP_CHECK(ca_bias_res_act_bf16(
+      reinterpret_cast<const void*>(x), reinterpret_cast<const void*>(res),
+      reinterpret_cast<const void*>(bias), reinterpret_cast<void*>(out),
+      n_planes, plane, channels, relu ? 1 : 0,
+      reinterpret_cast<hipStream_t>(stream_handle)));
+}
+
 static void rmsnorm_bf16(uintptr_t x, uintptr_t w, uintptr_t out,
                          long rows, int dim, double eps,
                          uintptr_t stream_handle) {
@@ -366,6 +377,25 @@ static void image_preprocess(uintptr_t src, uintptr_t dst, int ih, int iw,
                                 reinterpret_cast<void*>(dst), ih, iw, oh, ow,
                                 mode, out_bf16 ? 1 : 0, mean.data(),
                                 stdev.data(), s));
+  if (sync) {
+    py::gil_scoped_release release;
+    HIP_CHECK(hipStreamSynchronize(s));
+  }
+}
+
+static void image_preprocess_batched(uintptr_t src, uintptr_t dst,
+                                     int n_images, int ih, int iw, int oh,
+                                     int ow, int mode, bool out_bf16,
+                                     std::vector<float> mean,
+                                     std::vector<float> stdev, int device,
+                                     bool sync) {
+  if (mean.size() != 3 || stdev.size() != 3)
+    throw std::runtime_error("mean/std must have 3 channels");
+  hipStream_t s = get_stream(device);
+  HIP_CHECK(ca_image_preprocess_batched(
+      reinterpret_cast<const void*>(src), reinterpret_cast<void*>(dst),
+      n_images, ih, iw, oh, ow, mode, out_bf16 ? 1 : 0, mean.data(),
+      stdev.data(), s));
   if (sync) {
     py::gil_scoped_release release;
     HIP_CHECK(hipStreamSynchronize(s));
@@ -416,6 +446,9 @@ PYBIND11_MODULE(_hip_c, m) {
   m.def("bias_act_bf16", &bias_act_bf16, py::arg("x"), py::arg("bias"),
         py::arg("out"), py::arg("n_planes"), py::arg("plane"),
         py::arg("channels"), py::arg("relu"), py::arg("stream_handle"));
+  m.def("bias_res_act_bf16", &bias_res_act_bf16, py::arg("x"), py::arg("res"),
+        py::arg("bias"), py::arg("out"), py::arg("n_planes"), py::arg("plane"),
+        py::arg("channels"), py::arg("relu"), py::arg("stream_handle"));
   m.def("rmsnorm_bf16", &rmsnorm_bf16, py::arg("x"), py::arg("w"),
         py::arg("out"), py::arg("rows"), py::arg("dim"), py::arg("eps"),
         py::arg("stream_handle"));
@@ -429,6 +462,13 @@ PYBIND11_MODULE(_hip_c, m) {
   m.def("image_preprocess", &image_preprocess, py::arg("src"), py::arg("dst"),
         py::arg("ih"), py::arg("iw"), py::arg("oh"), py::arg("ow"),
         py::arg("mode") = 0, py::arg("out_bf16") = false,
+        py::arg("mean") = std::vector<float>{0.f, 0.f, 0.f},
+        py::arg("std") = std::vector<float>{1.f, 1.f, 1.f},
+        py::arg("device") = 0, py::arg("sync") = true);
+  m.def("image_preprocess_batched", &image_preprocess_batched, py::arg("src"),
+        py::arg("dst"), py::arg("n_images"), py::arg("ih"), py::arg("iw"),
+        py::arg("oh"), py::arg("ow"), py::arg("mode") = 0,
+        py::arg("out_bf16") = false,
         py::arg("mean") = std::vector<float>{0.f, 0.f, 0.f},
         py::arg("std") = std::vector<float>{1.f, 1.f, 1.f},
         py::arg("device") = 0, py::arg("sync") = true);

@@ -376,6 +376,52 @@ __global__ void gather_pack_kernel(const char* __restrict__ src,
   }
 }
 
+// LDS-tiled batched transpose: the gather_pack fast path for the
+// pattern "innermost output dim is strided in the source, the dim
+// before it is contiguous" (a transpose view, the worst case for the
+// naive per-element kernel: 729 GB/s vs ~5 TB/s for contiguous casts —
+// profiles/kernels_rocprof_r01.txt:18). 64x64 tile staged through LDS:
+// loads coalesce along the source's contiguous dim, stores along the
+// destination's, both full 64-lane wavefronts. +1 column pad keeps the
+// LDS column reads conflict-free.
+//
+//   dst[b0][b1][r][c] = src[b0*s0 + b1*s1 + r + c*col_stride]
+//   (r = contiguous source dim of extent rows; c of extent cols;
+//    blockIdx.z enumerates the flattened (b0, b1) outer batch)
+#define CA_TILE 64
+template <typename T>
+__global__ void transpose_tiled_kernel(const T* __restrict__ src,
+                                       T* __restrict__ dst, long rows,
+                                       long cols, long col_stride, long s0,
+                                       long s1, long d1) {
+  __shared__ T tile[CA_TILE][CA_TILE + 1];
+  const long batch = blockIdx.z;
+  const T* sb = src + (batch / d1) * s0 + (batch % d1) * s1;
+  T* db = dst + batch * rows * cols;
+  const long r0 = (long)blockIdx.x * CA_TILE;  // along rows (src-contig)
+  const long c0 = (long)blockIdx.y * CA_TILE;  // along cols (dst-contig)
+  const int tx = threadIdx.x % CA_TILE;        // fast lane index
+  const int ty0 = threadIdx.x / CA_TILE;       // 4 rows/pass with 256 thr
+  // load: lanes sweep the contiguous source dim (r), rows of the tile
+  // are different c values
+  for (int ty = ty0; ty < CA_TILE; ty += blockDim.x / CA_TILE) {
+    const long c = c0 + ty;
+    const long r = r0 + tx;
+    if (r < rows && c < cols) {
+      tile[ty][tx] = sb[r + c * col_stride];
+    }
+  }
+  __syncthreads();
+  // store: lanes sweep the contiguous destination dim (c)
+  for (int ty = ty0; ty < CA_TILE; ty += blockDim.x / CA_TILE) {
+    const long r = r0 + ty;
+    const long c = c0 + tx;
+    if (r < rows && c < cols) {
+      db[r * cols + c] = tile[tx][ty];
+    }
+  }
+}
+
 // Image preprocess: u8 HWC (ih,iw,3) -> bilinear resize (oh,ow) ->
 // normalize -> planar CHW fp32 (or bf16). One thread per output pixel
 // computes all 3 channels (reads coalesce along ow; the 4 source pixels
@@ -502,12 +548,54 @@ extern "C" hipError_t ca_cast_fp8e4m3_fp32(const void* src, void* dst, long n,
   return hipGetLastError();
 }
 
+namespace {
+
+template <typename T>
+hipError_t ca_transpose_tiled_launch(const void* src, void* dst,
+                                     const long* shape, const long* strides,
+                                     hipStream_t stream) {
+  const long rows = shape[2], cols = shape[3];
+  dim3 grid((uint32_t)((rows + CA_TILE - 1) / CA_TILE),
+            (uint32_t)((cols + CA_TILE - 1) / CA_TILE),
+            (uint32_t)(shape[0] * shape[1]));
+  hipLaunchKernelGGL((transpose_tiled_kernel<T>), grid, dim3(256), 0, stream,
+                     (const T*)src, (T*)dst, rows, cols, strides[3],
+                     strides[0], strides[1], shape[1]);
+  return hipGetLastError();
+}
+
+}  // namespace
+
 extern "C" hipError_t ca_gather_pack(const void* src, void* dst,
                                      int elem_size, const long* shape,
                                      const long* strides,
                                      hipStream_t stream) {
   // shape/strides are 4-element host arrays (leading dims padded)
   long n = shape[0] * shape[1] * shape[2] * shape[3];
+  // transpose-like fast path: dim 2 contiguous in the source, innermost
+  // output dim strided -> LDS-tiled transpose (coalesced both ways;
+  // the naive per-element path was 7x off peak on this pattern,
+  // profiles/kernels_rocprof_r01.txt:18)
+  if (strides[2] == 1 && strides[3] != 1 && shape[2] >= 16 &&
+      shape[3] >= 16 && shape[0] * shape[1] <= 65535 &&
+      (shape[3] + CA_TILE - 1) / CA_TILE <= 65535) {
+    switch (elem_size) {
+      case 1:
+        return ca_transpose_tiled_launch<uint8_t>(src, dst, shape, strides,
+                                                  stream);
+      case 2:
+        return ca_transpose_tiled_launch<uint16_t>(src, dst, shape, strides,
+                                                   stream);
+      case 4:
+        return ca_transpose_tiled_launch<uint32_t>(src, dst, shape, strides,
+                                                   stream);
+      case 8:
+        return ca_transpose_tiled_launch<uint64_t>(src, dst, shape, strides,
+                                                   stream);
+      default:
+        break;  // fall through to the scalar path
+    }
+  }
   int grid = ca_grid_for(n);
   const char* sp = (const char*)src;
   switch (elem_size) {
@@ -551,6 +639,81 @@ extern "C" hipError_t ca_image_preprocess(const void* src, void* dst, int ih,
                      (const uint8_t*)src, (float*)dst, ih, iw, oh, ow, mode,
                      out_bf16, mean[0], mean[1], mean[2], stdev[0], stdev[1],
                      stdev[2]);
+  return hipGetLastError();
+}
+
+// Batched variant: N same-sized images in one launch (u8 HWC stacked
+// contiguously -> NCHW). The single-image kernel is launch-bound at
+// high request rates (~27 us/image vs ~2-3 us of kernel time for
+// 224x224 — profiles/kernels_rocprof_r01.txt:19); the batch amortizes
+// one launch over the whole request. blockIdx.y = image index (grid.y
+// caps at 65535 — far above any sane batch).
+extern "C" __global__ void image_preprocess_batched_kernel(
+    const uint8_t* __restrict__ src, float* __restrict__ dst, int ih, int iw,
+    int oh, int ow, int mode, int out_bf16, float m0, float m1, float m2,
+    float s0, float s1, float s2) {
+  const long in_img = (long)ih * iw * 3;
+  const long n = (long)oh * ow;
+  const uint8_t* simg = src + (long)blockIdx.y * in_img;
+  float* dimg = dst + (long)blockIdx.y * 3 * n;
+  uint16_t* dimg16 =
+      reinterpret_cast<uint16_t*>(dst) + (long)blockIdx.y * 3 * n;
+  long i = (long)(blockIdx.x * blockDim.x + threadIdx.x);
+  const long stride = (long)gridDim.x * blockDim.x;
+  const float scale_h = (float)ih / oh;
+  const float scale_w = (float)iw / ow;
+  for (; i < n; i += stride) {
+    int oy = i / ow;
+    int ox = i % ow;
+    float fy = (oy + 0.5f) * scale_h - 0.5f;
+    float fx = (ox + 0.5f) * scale_w - 0.5f;
+    int y0 = max(0, (int)floorf(fy));
+    int x0 = max(0, (int)floorf(fx));
+    int y1 = min(ih - 1, y0 + 1);
+    int x1 = min(iw - 1, x0 + 1);
+    y0 = min(y0, ih - 1);
+    x0 = min(x0, iw - 1);
+    float wy = fy - floorf(fy);
+    float wx = fx - floorf(fx);
+    if (fy < 0) wy = 0.f;
+    if (fx < 0) wx = 0.f;
+    const uint8_t* p00 = simg + ((long)y0 * iw + x0) * 3;
+    const uint8_t* p01 = simg + ((long)y0 * iw + x1) * 3;
+    const uint8_t* p10 = simg + ((long)y1 * iw + x0) * 3;
+    const uint8_t* p11 = simg + ((long)y1 * iw + x1) * 3;
+#pragma unroll
+    for (int c = 0; c < 3; ++c) {
+      float v = (1 - wy) * ((1 - wx) * p00[c] + wx * p01[c]) +
+                wy * ((1 - wx) * p10[c] + wx * p11[c]);
+      float mean = c == 0 ? m0 : (c == 1 ? m1 : m2);
+      float sc = c == 0 ? s0 : (c == 1 ? s1 : s2);
+      if (mode == 1) {
+        v = v / 127.5f - 1.0f;
+      } else if (mode == 2) {
+        v = v - mean;
+      } else {
+        v = (v - mean) * sc;
+      }
+      long out_idx = (long)c * n + i;  // CHW within the image
+      if (out_bf16) {
+        dimg16[out_idx] = (uint16_t)(__float_as_uint(v) >> 16);
+      } else {
+        dimg[out_idx] = v;
+      }
+    }
+  }
+}
+
+extern "C" hipError_t ca_image_preprocess_batched(
+    const void* src, void* dst, int n_images, int ih, int iw, int oh, int ow,
+    int mode, int out_bf16, const float* mean, const float* stdev,
+    hipStream_t stream) {
+  if (n_images < 1 || n_images > 65535) return hipErrorInvalidValue;
+  dim3 grid((uint32_t)ca_grid_for((long)oh * ow), (uint32_t)n_images);
+  hipLaunchKernelGGL(image_preprocess_batched_kernel, grid, dim3(256), 0,
+                     stream, (const uint8_t*)src, (float*)dst, ih, iw, oh,
+                     ow, mode, out_bf16, mean[0], mean[1], mean[2], stdev[0],
+                     stdev[1], stdev[2]);
   return hipGetLastError();
 }
 
@@ -607,6 +770,74 @@ extern "C" hipError_t ca_bias_act_bf16(const void* x, const void* bias,
                      dim3(256), 0, stream, (const uint16_t*)x,
                      (const float*)bias, (uint16_t*)out, plane, channels,
                      do_relu);
+  return hipGetLastError();
+}
+
+// Per-channel bias + residual add + ReLU over NCHW bf16 in ONE pass —
+// the bottleneck-exit fusion (out = relu(x + bias[c] + residual)).
+// After BN folding, torch-on-ROCm runs this as three elementwise
+// kernels (bias add, residual add, relu), i.e. 3 extra full memory
+// passes over the activation per block exit; this kernel is one
+// read-x + read-residual + write pass. fp32 math, RNE back to bf16.
+// residual may be null (degenerates to bias[+relu]).
+extern "C" __global__ void bias_res_act_bf16_kernel(
+    const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
+    const float* __restrict__ bias, uint16_t* __restrict__ out, long plane,
+    int channels, int do_relu) {
+  const long pidx = blockIdx.x;  // n*C + c
+  const int c = (int)(pidx % channels);
+  const float b = bias[c];
+  const uint16_t* xp = x + pidx * plane;
+  const uint16_t* rp = res ? res + pidx * plane : nullptr;
+  uint16_t* op = out + pidx * plane;
+  const long vec_n = plane / 8;
+
+  auto fin = [&](float f) -> uint16_t {
+    if (do_relu && f < 0.f) f = 0.f;
+    uint32_t u = __float_as_uint(f);
+    u += 0x7FFF + ((u >> 16) & 1);  // RNE to bf16
+    return (uint16_t)(u >> 16);
+  };
+
+  const uint4* xv = (const uint4*)xp;
+  const uint4* rv = (const uint4*)rp;
+  uint4* ov = (uint4*)op;
+  for (long i = threadIdx.x; i < vec_n; i += blockDim.x) {
+    uint4 v = xv[i];
+    uint4 r = rp ? rv[i] : uint4{0, 0, 0, 0};
+    uint32_t* w = (uint32_t*)&v;
+    const uint32_t* rw = (const uint32_t*)&r;
+    for (int j = 0; j < 4; ++j) {
+      float lo = __uint_as_float((w[j] & 0xFFFFu) << 16) + b;
+      float hi = __uint_as_float(w[j] & 0xFFFF0000u) + b;
+      if (rp) {
+        lo += __uint_as_float((rw[j] & 0xFFFFu) << 16);
+        hi += __uint_as_float(rw[j] & 0xFFFF0000u);
+      }
+      w[j] = (uint32_t)fin(lo) | ((uint32_t)fin(hi) << 16);
+    }
+    ov[i] = v;
+  }
+  for (long i = vec_n * 8 + threadIdx.x; i < plane; i += blockDim.x) {
+    float f = __uint_as_float(((uint32_t)xp[i]) << 16) + b;
+    if (rp) f += __uint_as_float(((uint32_t)rp[i]) << 16);
+    op[i] = fin(f);
+  }
+}
+
+extern "C" hipError_t ca_bias_res_act_bf16(const void* x, const void* res,
+                                           const void* bias, void* out,
+                                           long n_planes, long plane,
+                                           int channels, int do_relu,
+                                           hipStream_t stream) {
+  if (((uintptr_t)x & 15) || ((uintptr_t)out & 15) ||
+      ((uintptr_t)res & 15)) {
+    return hipErrorInvalidValue;  // 16B alignment for the uint4 path
+  }
+  hipLaunchKernelGGL(bias_res_act_bf16_kernel, dim3((uint32_t)n_planes),
+                     dim3(256), 0, stream, (const uint16_t*)x,
+                     (const uint16_t*)res, (const float*)bias,
+                     (uint16_t*)out, plane, channels, do_relu);
   return hipGetLastError();
 }
 
