@@ -114,33 +114,136 @@ class BiasAct(nn.Module):
         return out.relu_() if self.relu_flag else out
 
 
-def fold_batchnorm(module):
+class BiasResAct(nn.Module):
+    """Bottleneck-exit fusion: out = relu(x + bias[c] + residual) in ONE
+    CDNA4 kernel pass. After BN folding torch-on-ROCm runs this as three
+    separate elementwise kernels (bias add, residual add, relu) — three
+    full memory passes over the activation per block
+    (profiles/serving_rocprof_postfold_r01.txt). fp32 math in-kernel,
+    RNE back to bf16."""
+
+    def __init__(self, bias):
+        super().__init__()
+        self.register_buffer("bias", bias.detach().float())
+
+    def forward(self, x, residual):
+        import torch
+
+        if (x.is_cuda and x.dtype == torch.bfloat16 and x.is_contiguous()
+                and residual.dtype == torch.bfloat16
+                and residual.is_contiguous()):
+            from ..ops import hip_runtime as hr
+
+            if self.bias.dtype != torch.float32:
+                # TorchModel's .to(bf16) sweeps buffers; kernel reads fp32
+                self.bias.data = self.bias.data.float()
+            n, c = x.shape[0], x.shape[1]
+            plane = x.numel() // (n * c)
+            hr.bias_res_act_bf16(
+                x.data_ptr(), residual.data_ptr(), self.bias.data_ptr(),
+                x.data_ptr(), n * c, plane, c, True,
+                torch.cuda.current_stream().cuda_stream)
+            return x
+        out = x + residual + self.bias.view(
+            1, -1, *([1] * (x.dim() - 2))
+        ).to(x.dtype)
+        return out.relu_()
+
+
+class FusedBottleneck(nn.Module):
+    """Bottleneck with BN folded into the convs and every elementwise
+    epilogue fused into one-pass CDNA4 kernels:
+
+        conv1 -> BiasAct(relu)     1 pass  (was bias add + relu = 2)
+        conv2 -> BiasAct(relu)     1 pass
+        conv3 (biasless), identity = downsample conv (biasless) or x
+              -> BiasResAct        1 pass  (was bias [+ds bias] + add
+                                            + relu = 3-4 passes)
+
+    The downsample conv's folded bias is per-channel on the same C as
+    conv3's, so it is pre-summed into the exit bias."""
+
+    def __init__(self, conv1, ba1, conv2, ba2, conv3, ds_conv, exit_bias):
+        super().__init__()
+        self.conv1, self.ba1 = conv1, ba1
+        self.conv2, self.ba2 = conv2, ba2
+        self.conv3 = conv3
+        self.ds_conv = ds_conv
+        self.exit = BiasResAct(exit_bias)
+
+    def forward(self, x):
+        identity = self.ds_conv(x) if self.ds_conv is not None else x
+        out = self.ba1(self.conv1(x))
+        out = self.ba2(self.conv2(out))
+        out = self.conv3(out)
+        return self.exit(out, identity)
+
+
+def _strip_bias(conv):
+    bias = conv.bias.detach().clone()
+    conv.bias = None
+    return bias
+
+
+def fold_batchnorm(module, fuse_eltwise=None):
     """Fold every Conv2d -> BatchNorm2d pair into the conv weights
     (algebraically exact in eval mode: W' = W*g/sqrt(v+eps),
     b' = b0*g/sqrt(v+eps) + beta - g*mean/sqrt(v+eps)). The serving
     profile showed BN kernels at ~29% of ResNet50 kernel time
     (profiles/serving_rocprof_r01.txt); folding removes them entirely.
-    Pairs are detected by _modules adjacency (conv immediately followed
-    by its BN, which matches this file and torchvision layouts); the BN
-    slot is replaced with Identity so forwards run unchanged."""
+
+    fuse_eltwise=True additionally rewrites every Bottleneck as a
+    FusedBottleneck and the stem bias+relu as one BiasAct pass, so the
+    folded conv biases never run as separate torch elementwise kernels
+    (default; CLIENT_AMD_FUSED_BIAS=0 disables). Pairs are detected by
+    _modules adjacency (conv immediately followed by its BN, matching
+    this file and torchvision layouts)."""
     from torch.nn.utils.fusion import fuse_conv_bn_eval
 
+    if fuse_eltwise is None:
+        fuse_eltwise = os.environ.get("CLIENT_AMD_FUSED_BIAS", "1") != "0"
+
+    def fuse_pair(conv, bn):
+        return fuse_conv_bn_eval(conv.eval(), bn.eval())
+
+    if fuse_eltwise:
+        # bottlenecks -> FusedBottleneck
+        for m in list(module.modules()):
+            for name, block in list(m._modules.items()):
+                if not isinstance(block, Bottleneck):
+                    continue
+                c1 = fuse_pair(block.conv1, block.bn1)
+                c2 = fuse_pair(block.conv2, block.bn2)
+                c3 = fuse_pair(block.conv3, block.bn3)
+                ba1 = BiasAct(_strip_bias(c1), relu=True)
+                ba2 = BiasAct(_strip_bias(c2), relu=True)
+                exit_bias = _strip_bias(c3)
+                ds_conv = None
+                if block.downsample is not None:
+                    ds_conv = fuse_pair(block.downsample[0],
+                                        block.downsample[1])
+                    exit_bias = exit_bias + _strip_bias(ds_conv)
+                m._modules[name] = FusedBottleneck(
+                    c1, ba1, c2, ba2, c3, ds_conv, exit_bias
+                )
+        # stem: conv1+bn1+relu -> biasless conv + one BiasAct(relu) pass
+        if (isinstance(getattr(module, "conv1", None), nn.Conv2d)
+                and isinstance(getattr(module, "bn1", None),
+                               nn.BatchNorm2d)):
+            fused = fuse_pair(module.conv1, module.bn1)
+            module.conv1 = fused
+            module.bn1 = BiasAct(_strip_bias(fused), relu=True)
+            module.relu = nn.Identity()
+
+    # generic pass folds any remaining conv->bn adjacency (stem when
+    # fuse_eltwise is off, custom models)
     for m in list(module.modules()):
         names = list(m._modules.keys())
         for a, b in zip(names, names[1:]):
             conv, bn = m._modules[a], m._modules[b]
             if isinstance(conv, nn.Conv2d) and isinstance(bn, nn.BatchNorm2d):
-                fused = fuse_conv_bn_eval(conv.eval(), bn.eval())
-                if os.environ.get("CLIENT_AMD_FUSED_BIAS") == "1":
-                    # bias as one fused kernel pass instead of torch's
-                    # separate elementwise add (see BiasAct)
-                    bias = fused.bias.detach().clone()
-                    fused.bias = None
-                    m._modules[a] = fused
-                    m._modules[b] = BiasAct(bias, relu=False)
-                else:
-                    m._modules[a] = fused
-                    m._modules[b] = nn.Identity()
+                m._modules[a] = fuse_pair(conv, bn)
+                m._modules[b] = nn.Identity()
     return module
 
 

@@ -412,6 +412,44 @@ def test_torchmodel_warmup_cpu_noop_and_cli():
     assert out["OUTPUT0"].shape == (2, 16)
 
 
+def test_resnet_fused_eltwise_matches_plain_fold():
+    """The FusedBottleneck rewrite (bias+relu and bias+residual+relu as
+    one-pass kernels, CPU fallback here) must match the plain BN-fold
+    numerics, with no conv left carrying a bias for torch to run as a
+    separate elementwise pass."""
+    import copy
+
+    import torch
+
+    from client_amd.models.resnet import (
+        FusedBottleneck,
+        ResNet50,
+        fold_batchnorm,
+    )
+
+    torch.manual_seed(1)
+    m = ResNet50().eval()
+    for mod in m.modules():
+        if isinstance(mod, torch.nn.BatchNorm2d):
+            mod.running_mean.uniform_(-0.5, 0.5)
+            mod.running_var.uniform_(0.5, 2.0)
+            mod.weight.data.uniform_(0.5, 1.5)
+            mod.bias.data.uniform_(-0.3, 0.3)
+    x = torch.randn(2, 3, 64, 64)
+    with torch.inference_mode():
+        plain = fold_batchnorm(copy.deepcopy(m), fuse_eltwise=False)
+        fused = fold_batchnorm(copy.deepcopy(m), fuse_eltwise=True)
+        assert torch.allclose(plain(x), fused(x), atol=1e-4, rtol=1e-4)
+    blocks = [b for b in fused.modules() if isinstance(b, FusedBottleneck)]
+    assert len(blocks) == 16  # 3 + 4 + 6 + 3
+    for b in blocks:
+        assert b.conv1.bias is None and b.conv2.bias is None
+        assert b.conv3.bias is None
+        if b.ds_conv is not None:
+            assert b.ds_conv.bias is None
+    assert fused.conv1.bias is None  # stem bias fused into BiasAct
+
+
 def test_resnet_batchnorm_folding_exact():
     """fold_batchnorm is algebraically exact in eval mode and removes
     every BatchNorm2d (serving profile showed BN at ~29% of ResNet50
