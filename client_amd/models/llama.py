@@ -288,6 +288,95 @@ class LlamaModel(nn.Module):
         x = self.norm(x)
         return self.lm_head(x)[:, 0]
 
+    def forward_prefill_chunk(self, tokens, pos_rows, row_lens, last_idx,
+                              kv_cache, bucket):
+        """One prefill chunk for a batch whose rows are at DIFFERENT
+        positions, with static shapes so the step can be hipGraph
+        captured (the eager per-slot chunk loop was the 86 ms ITL-stall
+        source — every admission blocked one decode iteration for a
+        full eager 32-layer forward).
+
+        tokens   [B, C] int64 — each row's chunk, padded arbitrarily
+        pos_rows [B]    int64 — absolute start position per row
+        row_lens [B]    int64 — real tokens in this row's chunk (0 for
+                                rows not prefilling: ALL their writes go
+                                to the scratch position)
+        last_idx [B]    int64 — index of the last real token (>= 0)
+        bucket          int   — static attention extent (>= every
+                                row's pos+len; padded extents are
+                                masked off per row)
+
+        Returns logits [B, vocab] taken at last_idx per row (garbage
+        for rows with row_lens == 0 — callers ignore them).
+
+        Row independence mirrors forward_decode_batch: per-row RoPE
+        offsets, per-row causal masks against each row's own history,
+        and padded/inactive positions scatter K/V to the reserved
+        scratch cache slot that reads never touch."""
+        cos, sin = self._get_rope(tokens.device)
+        b, c = tokens.shape
+        ar_b = torch.arange(b, device=tokens.device)
+        ar_c = torch.arange(c, device=tokens.device)
+        abs_pos = pos_rows[:, None] + ar_c[None, :]          # [B, C]
+        real = ar_c[None, :] < row_lens[:, None]             # [B, C]
+        # cache write index: real tokens at their absolute position,
+        # padding/inactive rows at the scratch slot (never read)
+        write_idx = torch.where(
+            real, abs_pos.clamp(max=self.scratch_pos),
+            torch.full_like(abs_pos, self.scratch_pos),
+        )
+        rope_idx = abs_pos.clamp(max=self.scratch_pos)       # [B, C]
+        c_rows = cos[rope_idx]                               # [B, C, hd/2]
+        s_rows = sin[rope_idx]
+        # causal mask vs each row's own history: query at abs pos p
+        # sees keys j <= p
+        key_idx = torch.arange(bucket, device=tokens.device)
+        mask = torch.where(
+            key_idx[None, None, :] <= abs_pos[:, :, None],
+            torch.zeros((), device=tokens.device, dtype=torch.float32),
+            torch.full((), float("-inf"), device=tokens.device,
+                       dtype=torch.float32),
+        )[:, None]                                           # [B,1,C,bucket]
+
+        def rope_rows(t):
+            # t: [B, h, C, d]; tables broadcast over heads
+            cc = c_rows[:, None]
+            ss = s_rows[:, None]
+            t1, t2 = t[..., 0::2], t[..., 1::2]
+            out = torch.empty_like(t)
+            out[..., 0::2] = t1 * cc - t2 * ss
+            out[..., 1::2] = t1 * ss + t2 * cc
+            return out
+
+        x = self.tok(tokens)                                 # [B, C, dim]
+        rep = self.cfg.n_heads // self.cfg.n_kv_heads
+        for block, (ck, cv) in zip(self.blocks, kv_cache):
+            h = block.attn_norm(x)
+            q = block.wq(h).view(b, c, self.cfg.n_heads, block.head_dim
+                                 ).transpose(1, 2)
+            k = block.wk(h).view(b, c, self.cfg.n_kv_heads, block.head_dim
+                                 ).transpose(1, 2)
+            v = block.wv(h).view(b, c, self.cfg.n_kv_heads, block.head_dim
+                                 ).transpose(1, 2)
+            q = rope_rows(q)
+            k = rope_rows(k)
+            # scatter the chunk's K/V at per-row positions (padding ->
+            # scratch); advanced-index dims land in front: [B, C, kv, d]
+            ck[ar_b[:, None], :, write_idx] = k.permute(0, 2, 1, 3)
+            cv[ar_b[:, None], :, write_idx] = v.permute(0, 2, 1, 3)
+            k_all = ck[:, :, :bucket].repeat_interleave(rep, dim=1)
+            v_all = cv[:, :, :bucket].repeat_interleave(rep, dim=1)
+            attn = F.scaled_dot_product_attention(
+                q, k_all, v_all, attn_mask=mask.to(q.dtype)
+            )
+            attn = attn.transpose(1, 2).reshape(b, c, -1)
+            x = x + block.wo(attn)
+            h = block.ffn_norm(x)
+            x = x + block.w2(F.silu(block.w1(h)) * block.w3(h))
+        x_last = x[ar_b, last_idx]                           # [B, dim]
+        x_last = self.norm(x_last)
+        return self.lm_head(x_last)
+
     @torch.inference_mode()
     def generate(self, input_ids, max_new_tokens):
         """Greedy decode; yields one token id tensor [b] per step."""

@@ -808,20 +808,23 @@ extern "C" __global__ void bias_res_act_bf16_kernel(
     uint32_t* w = (uint32_t*)&v;
     const uint32_t* rw = (const uint32_t*)&r;
     for (int j = 0; j < 4; ++j) {
-      float lo = __uint_as_float((w[j] & 0xFFFFu) << 16) + b;
-      float hi = __uint_as_float(w[j] & 0xFFFF0000u) + b;
+      // addition order (x + residual) + bias matches the torch
+      // reference sequence exactly (fp32 add is non-associative;
+      // a different order flips RNE boundary cases)
+      float lo = __uint_as_float((w[j] & 0xFFFFu) << 16);
+      float hi = __uint_as_float(w[j] & 0xFFFF0000u);
       if (rp) {
         lo += __uint_as_float((rw[j] & 0xFFFFu) << 16);
         hi += __uint_as_float(rw[j] & 0xFFFF0000u);
       }
-      w[j] = (uint32_t)fin(lo) | ((uint32_t)fin(hi) << 16);
+      w[j] = (uint32_t)fin(lo + b) | ((uint32_t)fin(hi + b) << 16);
     }
     ov[i] = v;
   }
   for (long i = vec_n * 8 + threadIdx.x; i < plane; i += blockDim.x) {
-    float f = __uint_as_float(((uint32_t)xp[i]) << 16) + b;
+    float f = __uint_as_float(((uint32_t)xp[i]) << 16);
     if (rp) f += __uint_as_float(((uint32_t)rp[i]) << 16);
-    op[i] = fin(f);
+    op[i] = fin(f + b);
   }
 }
 

@@ -72,6 +72,19 @@ class DecodeScheduler:
         self._pos_dev = torch.full((max_batch,), self._scratch,
                                    dtype=torch.int64, device=device)
         self._graphs = {}  # bucket -> (graph, next_tokens_out)
+        # batched prefill: one static-shape forward advances EVERY
+        # mid-prefill slot by one chunk (replacing the per-slot eager
+        # loop whose 32-layer launch overhead stalled decode ~86 ms per
+        # admission); hipGraph-captured per attention bucket like decode
+        self._pf_tokens = torch.zeros(max_batch, prefill_chunk,
+                                      dtype=torch.int64, device=device)
+        self._pf_pos = torch.zeros(max_batch, dtype=torch.int64,
+                                   device=device)
+        self._pf_lens = torch.zeros(max_batch, dtype=torch.int64,
+                                    device=device)
+        self._pf_last = torch.zeros(max_batch, dtype=torch.int64,
+                                    device=device)
+        self._pf_graphs = {}  # bucket -> (graph, first_tokens_out)
         self._pending = queue.Queue()
         self._cv = threading.Condition()
         self._alive = True

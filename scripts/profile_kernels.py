@@ -67,6 +67,12 @@ def main():
         lambda: hr.gather_pack(a, b, 4, [rows, cols], [1, rows], 0, True),
         rows * cols * 8,
     )
+    # bf16 transpose through the same tiled path (2B elements)
+    results["gather_pack_T_bf16"] = bench(
+        "gather_pack bf16 16k^2 T",
+        lambda: hr.gather_pack(a, b, 2, [rows, cols], [1, rows], 0, True),
+        rows * cols * 4,
+    )
     # image preprocess: 720p -> 224x224
     ih, iw = 720, 1280
     img = hr.malloc(0, ih * iw * 3)
@@ -78,6 +84,19 @@ def main():
         ih * iw * 3 + 3 * 224 * 224 * 4,
     )
     hr.free(img)
+    # batched preprocess: 64 images per launch (launch-bound case)
+    nimg = 64
+    imgb = hr.malloc(0, nimg * ih * iw * 3)
+    outb = hr.malloc(0, nimg * 3 * 224 * 224 * 4)
+    results["image_preprocess_batched"] = bench(
+        f"image_preprocess x{nimg}/launch",
+        lambda: hr.image_preprocess_batched(imgb, outb, nimg, ih, iw, 224,
+                                            224, 1, False, [0, 0, 0],
+                                            [1, 1, 1], 0, True),
+        nimg * (ih * iw * 3 + 3 * 224 * 224 * 4),
+    )
+    hr.free(imgb)
+    hr.free(outb)
     hr.free(out)
     hr.free(a)
     hr.free(b)

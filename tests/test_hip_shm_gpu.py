@@ -547,7 +547,10 @@ def test_gather_pack_tiled_transpose(hipshm):
                            list(xt.shape), list(xt.stride()), 0, True)
             out = np.empty(n * esz, dtype=np.uint8)
             hr.memcpy_d2h_into(dst, out, n * esz, 0)
-            ref = xt.contiguous().cpu().numpy().view(np.uint8).reshape(-1)
+            refc = xt.contiguous().cpu()
+            if dt == torch.bfloat16:
+                refc = refc.view(torch.uint16)
+            ref = refc.numpy().view(np.uint8).reshape(-1)
             np.testing.assert_array_equal(out, ref, err_msg=str((shape, dt)))
         finally:
             hr.free(dst)
