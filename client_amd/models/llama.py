@@ -289,8 +289,8 @@ class LlamaModel(nn.Module):
         return self.lm_head(x)[:, 0]
 
     def forward_prefill_chunk(self, tokens, pos_rows, row_lens, last_idx,
-                              kv_cache, bucket):
-        """One prefill chunk for a batch whose rows are at DIFFERENT
+                              kv_cache, bucket, row_map=None):
+        """One prefill chunk for a (small) batch of rows at DIFFERENT
         positions, with static shapes so the step can be hipGraph
         captured (the eager per-slot chunk loop was the 86 ms ITL-stall
         source — every admission blocked one decode iteration for a
@@ -299,23 +299,35 @@ class LlamaModel(nn.Module):
         tokens   [B, C] int64 — each row's chunk, padded arbitrarily
         pos_rows [B]    int64 — absolute start position per row
         row_lens [B]    int64 — real tokens in this row's chunk (0 for
-                                rows not prefilling: ALL their writes go
-                                to the scratch position)
+                                padding rows: ALL their writes go to
+                                the scratch position)
         last_idx [B]    int64 — index of the last real token (>= 0)
         bucket          int   — static attention extent (>= every
                                 row's pos+len; padded extents are
                                 masked off per row)
+        row_map  [B]    int64 — KV-cache row index each batch row maps
+                                to (defaults to arange). This is what
+                                keeps the captured compute proportional
+                                to the rows actually prefilling: B is
+                                the GROUP size (usually 1), not the
+                                full decode batch — a [full_batch, C]
+                                static forward costs ~B*C positions of
+                                GEMM per replay no matter how few rows
+                                are real, which measured as a 60+ ms
+                                replay that defeated the point.
 
         Returns logits [B, vocab] taken at last_idx per row (garbage
         for rows with row_lens == 0 — callers ignore them).
 
         Row independence mirrors forward_decode_batch: per-row RoPE
         offsets, per-row causal masks against each row's own history,
-        and padded/inactive positions scatter K/V to the reserved
-        scratch cache slot that reads never touch."""
+        and padded positions scatter K/V to the reserved scratch cache
+        slot that reads never touch."""
         cos, sin = self._get_rope(tokens.device)
         b, c = tokens.shape
         ar_b = torch.arange(b, device=tokens.device)
+        if row_map is None:
+            row_map = ar_b
         ar_c = torch.arange(c, device=tokens.device)
         abs_pos = pos_rows[:, None] + ar_c[None, :]          # [B, C]
         real = ar_c[None, :] < row_lens[:, None]             # [B, C]
@@ -360,12 +372,15 @@ class LlamaModel(nn.Module):
                                  ).transpose(1, 2)
             q = rope_rows(q)
             k = rope_rows(k)
-            # scatter the chunk's K/V at per-row positions (padding ->
-            # scratch); advanced-index dims land in front: [B, C, kv, d]
-            ck[ar_b[:, None], :, write_idx] = k.permute(0, 2, 1, 3)
-            cv[ar_b[:, None], :, write_idx] = v.permute(0, 2, 1, 3)
-            k_all = ck[:, :, :bucket].repeat_interleave(rep, dim=1)
-            v_all = cv[:, :, :bucket].repeat_interleave(rep, dim=1)
+            # scatter the chunk's K/V at per-row positions in the
+            # MAPPED cache rows (padding -> scratch); advanced-index
+            # dims land in front: [B, C, kv, d]
+            ck[row_map[:, None], :, write_idx] = k.permute(0, 2, 1, 3)
+            cv[row_map[:, None], :, write_idx] = v.permute(0, 2, 1, 3)
+            k_all = ck.index_select(0, row_map)[:, :, :bucket]
+            v_all = cv.index_select(0, row_map)[:, :, :bucket]
+            k_all = k_all.repeat_interleave(rep, dim=1)
+            v_all = v_all.repeat_interleave(rep, dim=1)
             attn = F.scaled_dot_product_attention(
                 q, k_all, v_all, attn_mask=mask.to(q.dtype)
             )

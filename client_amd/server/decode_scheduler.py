@@ -43,7 +43,7 @@ class DecodeScheduler:
     END = object()
 
     def __init__(self, model, max_batch=8, device="cuda:0", dtype=None,
-                 use_graph=None, len_bucket=256, prefill_chunk=512):
+                 use_graph=None, len_bucket=256, prefill_chunk=256):
         self.model = model
         self.device = device
         self.dtype = dtype if dtype is not None else next(
@@ -72,19 +72,16 @@ class DecodeScheduler:
         self._pos_dev = torch.full((max_batch,), self._scratch,
                                    dtype=torch.int64, device=device)
         self._graphs = {}  # bucket -> (graph, next_tokens_out)
-        # batched prefill: one static-shape forward advances EVERY
-        # mid-prefill slot by one chunk (replacing the per-slot eager
+        # batched prefill: one static-shape forward advances the
+        # mid-prefill slots by one chunk (replacing the per-slot eager
         # loop whose 32-layer launch overhead stalled decode ~86 ms per
-        # admission); hipGraph-captured per attention bucket like decode
-        self._pf_tokens = torch.zeros(max_batch, prefill_chunk,
-                                      dtype=torch.int64, device=device)
-        self._pf_pos = torch.zeros(max_batch, dtype=torch.int64,
-                                   device=device)
-        self._pf_lens = torch.zeros(max_batch, dtype=torch.int64,
-                                    device=device)
-        self._pf_last = torch.zeros(max_batch, dtype=torch.int64,
-                                    device=device)
-        self._pf_graphs = {}  # bucket -> (graph, first_tokens_out)
+        # admission). Graphs are captured per (group_size, bucket) with
+        # a row_map into the cache, so replay compute is proportional
+        # to the rows actually prefilling — a single full-batch static
+        # shape measured as a 60+ ms replay (8x512 positions of GEMM)
+        # and made the stall WORSE.
+        self._pf_bufs = {}    # group_size -> dict of persistent tensors
+        self._pf_graphs = {}  # (group_size, bucket) -> (graph, out)
         self._pending = queue.Queue()
         self._cv = threading.Condition()
         self._alive = True
@@ -108,9 +105,16 @@ class DecodeScheduler:
         for i in range(1, n_buckets + 1):
             bucket = min(i * self.len_bucket, self.model.cfg.max_seq)
             self._get_graph(bucket)
-            self._get_prefill_graph(bucket)
+            # group=1 covers the common case (one admission at a time);
+            # other group sizes capture lazily on first use
+            self._get_prefill_graph(1, bucket)
             if bucket >= self.model.cfg.max_seq:
                 break
+        # a full-concurrency start burst admits max_batch prompts in one
+        # iteration (first bucket) — capture that shape up front too
+        self._get_prefill_graph(self._group_size(self.max_batch),
+                                min(self.len_bucket,
+                                    self.model.cfg.max_seq))
 
     def shutdown(self):
         with self._cv:
@@ -153,20 +157,35 @@ class DecodeScheduler:
             slot.remaining = max_new
             slot.out_queue = out
 
-    def _get_prefill_graph(self, bucket):
-        entry = self._pf_graphs.get(bucket)
+    def _pf_buffers(self, group):
+        bufs = self._pf_bufs.get(group)
+        if bufs is None:
+            dev = self.device
+            bufs = {
+                "tokens": torch.zeros(group, self.prefill_chunk,
+                                      dtype=torch.int64, device=dev),
+                "pos": torch.zeros(group, dtype=torch.int64, device=dev),
+                "lens": torch.zeros(group, dtype=torch.int64, device=dev),
+                "last": torch.zeros(group, dtype=torch.int64, device=dev),
+                "rows": torch.zeros(group, dtype=torch.int64, device=dev),
+            }
+            self._pf_bufs[group] = bufs
+        return bufs
+
+    def _get_prefill_graph(self, group, bucket):
+        entry = self._pf_graphs.get((group, bucket))
         if entry is not None:
             return entry
+        bufs = self._pf_buffers(group)
+        args = (bufs["tokens"], bufs["pos"], bufs["lens"], bufs["last"],
+                self.kv_cache, bucket, bufs["rows"])
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
             with torch.inference_mode():
                 for _ in range(2):  # warmup (idempotent cache writes)
-                    logits = self.model.forward_prefill_chunk(
-                        self._pf_tokens, self._pf_pos, self._pf_lens,
-                        self._pf_last, self.kv_cache, bucket,
-                    )
-                    warm = logits.argmax(-1)
+                    warm = self.model.forward_prefill_chunk(
+                        *args).argmax(-1)
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
         from .models import GRAPH_CAPTURE_LOCK
@@ -176,79 +195,95 @@ class DecodeScheduler:
             with torch.inference_mode():
                 with torch.cuda.graph(graph,
                                       capture_error_mode="thread_local"):
-                    logits = self.model.forward_prefill_chunk(
-                        self._pf_tokens, self._pf_pos, self._pf_lens,
-                        self._pf_last, self.kv_cache, bucket,
-                    )
-                    first_out = logits.argmax(-1)
+                    first_out = self.model.forward_prefill_chunk(
+                        *args).argmax(-1)
         entry = (graph, first_out)
-        self._pf_graphs[bucket] = entry
+        self._pf_graphs[(group, bucket)] = entry
         return entry
 
+    @staticmethod
+    def _group_size(n):
+        g = 1
+        while g < n:
+            g *= 2
+        return g
+
     def _prefill_step(self):
-        """Advance EVERY mid-prefill slot by one chunk with a single
-        static-shape forward (hipGraph-replayed on GPU). The previous
-        per-slot eager loop paid a full 32-layer launch-bound forward
-        per admission, stalling the decode loop ~86 ms; the batched
-        captured step costs roughly one decode step."""
+        """Advance every mid-prefill slot by one chunk. Slots are
+        processed in power-of-two groups whose hipGraphs are captured
+        per (group_size, bucket): replay compute stays proportional to
+        the rows actually prefilling (usually one), so the decode-loop
+        stall per admission is a few ms instead of a full eager
+        32-layer forward (~86 ms)."""
         pf = [i for i, s in enumerate(self.slots)
               if s.state == _Slot.PREFILL]
         if not pf:
             return
         c = self.prefill_chunk
-        b = self.max_batch
-        tokens = torch.zeros(b, c, dtype=torch.int64)
-        pos = torch.zeros(b, dtype=torch.int64)
-        lens = torch.zeros(b, dtype=torch.int64)
-        last = torch.zeros(b, dtype=torch.int64)
-        ends = {}
-        max_end = 1
-        for i in pf:
-            slot = self.slots[i]
-            ids = slot.prefill_ids
-            total = ids.shape[1]
-            start = slot.prefill_pos
-            end = min(start + c, total)
-            n_real = end - start
-            tokens[i, :n_real] = ids[0, start:end]
-            pos[i] = start
-            lens[i] = n_real
-            last[i] = n_real - 1
-            ends[i] = (end, total)
-            max_end = max(max_end, end)
-        self._pf_tokens.copy_(tokens)
-        self._pf_pos.copy_(pos)
-        self._pf_lens.copy_(lens)
-        self._pf_last.copy_(last)
-        if self.use_graph:
-            graph, first_out = self._get_prefill_graph(self._bucket(max_end))
-            graph.replay()
-            firsts = first_out.tolist()
-        else:
-            with torch.inference_mode():
-                logits = self.model.forward_prefill_chunk(
-                    self._pf_tokens, self._pf_pos, self._pf_lens,
-                    self._pf_last, self.kv_cache,
-                    self._bucket(max_end),
-                )
-                firsts = logits.argmax(-1).tolist()
-        for i in pf:
-            slot = self.slots[i]
-            end, total = ends[i]
-            slot.prefill_pos = end
-            if end < total:
-                continue
-            first = int(firsts[i])
-            slot.out_queue.put(first)
-            slot.prefill_ids = None
-            slot.pos = total  # position the NEXT token is written at
-            slot.last_token = first
-            slot.remaining -= 1
-            if slot.remaining <= 0:
-                slot.state = _Slot.FREE
-                slot.out_queue.put(self.END)
+        done = 0
+        while done < len(pf):
+            group_slots = pf[done:done + self.max_batch]
+            group = min(self._group_size(len(group_slots)),
+                        self.max_batch)
+            group_slots = group_slots[:group]
+            done += len(group_slots)
+            tokens = torch.zeros(group, c, dtype=torch.int64)
+            pos = torch.zeros(group, dtype=torch.int64)
+            lens = torch.zeros(group, dtype=torch.int64)
+            last = torch.zeros(group, dtype=torch.int64)
+            rows = torch.zeros(group, dtype=torch.int64)
+            ends = {}
+            max_end = 1
+            for g, i in enumerate(group_slots):
+                slot = self.slots[i]
+                ids = slot.prefill_ids
+                total = ids.shape[1]
+                start = slot.prefill_pos
+                end = min(start + c, total)
+                n_real = end - start
+                tokens[g, :n_real] = ids[0, start:end]
+                pos[g] = start
+                lens[g] = n_real
+                last[g] = n_real - 1
+                rows[g] = i
+                ends[i] = (end, total)
+                max_end = max(max_end, end)
+            bucket = self._bucket(max_end)
+            bufs = self._pf_buffers(group)
+            bufs["tokens"].copy_(tokens)
+            bufs["pos"].copy_(pos)
+            bufs["lens"].copy_(lens)
+            bufs["last"].copy_(last)
+            bufs["rows"].copy_(rows)
+            if self.use_graph:
+                graph, first_out = self._get_prefill_graph(group, bucket)
+                graph.replay()
+                firsts = first_out.tolist()
             else:
-                slot.state = _Slot.ACTIVE
+                with torch.inference_mode():
+                    logits = self.model.forward_prefill_chunk(
+                        bufs["tokens"], bufs["pos"], bufs["lens"],
+                        bufs["last"], self.kv_cache, bucket,
+                        bufs["rows"],
+                    )
+                    firsts = logits.argmax(-1).tolist()
+            for g, i in enumerate(group_slots):
+                slot = self.slots[i]
+                end, total = ends[i]
+                slot.prefill_pos = end
+                if end < total:
+                    continue
+                first = int(firsts[g])
+                slot.out_queue.put(first)
+                slot.prefill_ids = None
+                slot.pos = total  # position the NEXT token writes at
+                slot.last_token = first
+                slot.remaining -= 1
+                if slot.remaining <= 0:
+                    slot.state = _Slot.FREE
+                    slot.out_queue.put(self.END)
+                else:
+                    slot.state = _Slot.ACTIVE
 
     def _bucket(self, max_len):
         b = ((max_len + self.len_bucket - 1) // self.len_bucket
