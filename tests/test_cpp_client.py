@@ -451,3 +451,256 @@ def test_cc_memory_leak_suite(http_fixture_server, grpc_fixture_server):
     )
     assert proc.returncode == 0, proc.stdout + proc.stderr
     assert "ALL PASSED" in proc.stdout
+
+
+def test_modelconfig_cross_decode():
+    """C++ full-tree ModelConfig decoder vs the Python runtime schema
+    as oracle: a maximal config touching EVERY message of
+    model_config.proto (reference proto/model_config.proto:86-2180) is
+    serialized by Python and decoded by the pb_config_dump tool; every
+    field must round-trip."""
+    import json
+
+    from client_amd.grpc._proto import service_pb2
+
+    try:
+        dump_bin = _compile("pb_config_dump",
+                            CPP / "tests" / "pb_config_dump.cc")
+    except subprocess.CalledProcessError as e:
+        pytest.fail(f"C++ compile failed:\n{e.stderr}")
+
+    c = service_pb2.ModelConfig()
+    c.name = "maximal"
+    c.platform = "client_amd"
+    c.backend = "hip"
+    c.runtime = "mi355x"
+    c.max_batch_size = 64
+    c.version_policy.specific.versions.extend([3, 7, 11])
+
+    i0 = c.input.add()
+    i0.name = "IN0"
+    i0.data_type = 11  # TYPE_FP32
+    i0.dims.extend([3, 224, 224])
+    i0.format = 2      # FORMAT_NCHW
+    i0.reshape.shape.extend([1, 3, 224, 224])
+    i0.is_shape_tensor = False
+    i0.allow_ragged_batch = True
+    i0.optional = True
+    i0.is_non_linear_format_io = True
+
+    o0 = c.output.add()
+    o0.name = "OUT0"
+    o0.data_type = 13  # TYPE_BF16-ish enum slot
+    o0.dims.extend([1000])
+    o0.label_filename = "labels.txt"
+    o0.reshape.shape.extend([10, 100])
+    o0.is_shape_tensor = True
+    o0.is_non_linear_format_io = True
+
+    g = c.instance_group.add()
+    g.name = "ig0"
+    g.kind = 1
+    g.count = 4
+    g.gpus.extend([0, 1, 2, 3])
+    g.profile.extend(["p0", "p1"])
+    g.passive = True
+    g.host_policy = "numa0"
+    res = g.rate_limiter.resources.add()
+    res.name = "R0"
+    setattr(res, "global", True)
+    res.count = 5
+    g.rate_limiter.priority = 9
+    sd = g.secondary_devices.add()
+    sd.kind = 0
+    sd.device_id = 42
+
+    c.default_model_filename = "model.bin"
+    c.cc_model_filenames["linux"] = "model_linux.so"
+    c.metric_tags["team"] = "infra"
+    c.parameters["key"].string_value = "value"
+
+    opt = c.optimization
+    opt.graph.level = 2
+    opt.priority = 1
+    opt.cuda.graphs = True
+    opt.cuda.busy_wait_events = True
+    opt.cuda.output_copy_stream = True
+    spec = opt.cuda.graph_spec.add()
+    spec.batch_size = 8
+    spec.input["IN0"].dim.extend([3, 224, 224])
+    spec.graph_lower_bound.batch_size = 1
+    spec.graph_lower_bound.input["IN0"].dim.extend([3, 64, 64])
+    acc = opt.execution_accelerators.gpu_execution_accelerator.add()
+    acc.name = "mfma"
+    acc.parameters["tile"] = "64"
+    acc2 = opt.execution_accelerators.cpu_execution_accelerator.add()
+    acc2.name = "cpu0"
+    opt.input_pinned_memory.enable = True
+    opt.output_pinned_memory.enable = False
+    opt.gather_kernel_buffer_threshold = 7
+    opt.eager_batching = True
+
+    sb = c.sequence_batching
+    sb.max_sequence_idle_microseconds = 1000000
+    sb.iterative_sequence = True
+    sb.oldest.max_candidate_sequences = 12
+    sb.oldest.preferred_batch_size.extend([4, 8])
+    sb.oldest.max_queue_delay_microseconds = 500
+    sb.oldest.preserve_ordering = True
+    ci = sb.control_input.add()
+    ci.name = "START"
+    ctl = ci.control.add()
+    ctl.kind = 0
+    ctl.int32_false_true.extend([0, 1])
+    ctl.data_type = 8
+    ctl2 = ci.control.add()
+    ctl2.kind = 2
+    ctl2.fp32_false_true.extend([0.0, 1.0])
+    ctl3 = ci.control.add()
+    ctl3.kind = 1
+    ctl3.bool_false_true.extend([False, True])
+    st = sb.state.add()
+    st.input_name = "S_IN"
+    st.output_name = "S_OUT"
+    st.data_type = 11
+    st.dims.extend([256])
+    st.use_same_buffer_for_input_output = True
+    st.use_growable_memory = True
+    ist = st.initial_state.add()
+    ist.name = "init0"
+    ist.data_type = 11
+    ist.dims.extend([256])
+    ist.zero_data = True
+
+    w = c.model_warmup.add()
+    w.name = "warm0"
+    w.batch_size = 8
+    w.count = 3
+    wi = w.inputs["IN0"]
+    wi.data_type = 11
+    wi.dims.extend([3, 224, 224])
+    wi.random_data = True
+
+    bi = c.batch_input.add()
+    bi.kind = 3
+    bi.target_name.append("RAGGED_SHAPE")
+    bi.data_type = 8
+    bi.source_input.append("IN0")
+    bo = c.batch_output.add()
+    bo.kind = 0
+    bo.target_name.append("OUT0")
+    bo.source_input.append("IN0")
+
+    c.model_operations.op_library_filename.append("libops.so")
+    c.model_transaction_policy.decoupled = True
+    ag = c.model_repository_agents.agents.add()
+    ag.name = "checksum"
+    ag.parameters["algo"] = "sha256"
+    c.response_cache.enable = True
+    mc = c.model_metrics.metric_control.add()
+    mc.metric_identifier.family = "nv_inference_count"
+    mc.histogram_options.buckets.extend([0.5, 1.0, 2.5])
+
+    resp = service_pb2.ModelConfigResponse()
+    resp.config.CopyFrom(c)
+    wire = resp.SerializeToString()
+
+    proc = subprocess.run([str(dump_bin)], input=wire,
+                          capture_output=True, timeout=30)
+    assert proc.returncode == 0, proc.stderr.decode()
+    d = json.loads(proc.stdout.decode())
+
+    assert d["name"] == "maximal"
+    assert d["platform"] == "client_amd"
+    assert d["backend"] == "hip"
+    assert d["runtime"] == "mi355x"
+    assert d["max_batch_size"] == 64
+    assert d["version_policy_choice"] == 3  # SPECIFIC
+    assert d["version_policy_specific"] == [3, 7, 11]
+
+    di = d["input"][0]
+    assert di["name"] == "IN0" and di["data_type"] == 11
+    assert di["dims"] == [3, 224, 224] and di["format"] == 2
+    assert di["has_reshape"] and di["reshape"] == [1, 3, 224, 224]
+    assert di["allow_ragged_batch"] and di["optional"]
+    assert di["is_non_linear_format_io"]
+
+    do = d["output"][0]
+    assert do["name"] == "OUT0" and do["label_filename"] == "labels.txt"
+    assert do["dims"] == [1000] and do["reshape"] == [10, 100]
+    assert do["is_shape_tensor"] and do["is_non_linear_format_io"]
+
+    dg = d["instance_group"][0]
+    assert dg["name"] == "ig0" and dg["kind"] == 1 and dg["count"] == 4
+    assert dg["gpus"] == [0, 1, 2, 3] and dg["profile"] == ["p0", "p1"]
+    assert dg["passive"] and dg["host_policy"] == "numa0"
+    assert dg["has_rate_limiter"] and dg["rate_limiter_priority"] == 9
+    assert dg["rate_limiter_resources"][0]["name"] == "R0"
+    assert dg["rate_limiter_resources"][0]["count"] == 5
+    assert dg["secondary_devices"][0]["device_id"] == 42
+
+    assert d["default_model_filename"] == "model.bin"
+    assert d["cc_model_filenames"] == {"linux": "model_linux.so"}
+    assert d["metric_tags"] == {"team": "infra"}
+    assert d["parameters"] == {"key": "value"}
+
+    od = d["optimization"]
+    assert d["has_optimization"]
+    assert od["has_graph"] and od["graph_level"] == 2
+    assert od["priority"] == 1
+    assert od["has_cuda"] and od["cuda_graphs"]
+    assert od["cuda_busy_wait_events"] and od["cuda_output_copy_stream"]
+    gs = od["cuda_graph_spec"][0]
+    assert gs["batch_size"] == 8 and gs["input"]["IN0"] == [3, 224, 224]
+    assert gs["has_lower_bound"] and gs["lower_bound_batch_size"] == 1
+    assert od["gpu_execution_accelerator"][0]["name"] == "mfma"
+    assert od["gpu_execution_accelerator"][0]["parameters"] == {"tile": "64"}
+    assert od["cpu_execution_accelerator"][0]["name"] == "cpu0"
+    assert od["has_input_pinned_memory"] and od["input_pinned_memory"]
+    assert od["has_output_pinned_memory"] and not od["output_pinned_memory"]
+    assert od["gather_kernel_buffer_threshold"] == 7
+    assert od["eager_batching"]
+
+    sd2 = d["sequence_batching"]
+    assert d["has_sequence_batching"]
+    assert sd2["strategy"] == 2  # OLDEST
+    assert sd2["oldest_max_candidate_sequences"] == 12
+    assert sd2["oldest_preferred_batch_size"] == [4, 8]
+    assert sd2["oldest_max_queue_delay_microseconds"] == 500
+    assert sd2["oldest_preserve_ordering"]
+    assert sd2["max_sequence_idle_microseconds"] == 1000000
+    assert sd2["iterative_sequence"]
+    dci = sd2["control_input"][0]
+    assert dci["name"] == "START"
+    assert dci["control"][0]["int32_false_true"] == [0, 1]
+    assert dci["control"][0]["data_type"] == 8
+    assert dci["control"][1]["fp32_false_true"] == [0.0, 1.0]
+    assert dci["control"][2]["bool_false_true"] == [False, True]
+    dst = sd2["state"][0]
+    assert dst["input_name"] == "S_IN" and dst["output_name"] == "S_OUT"
+    assert dst["dims"] == [256]
+    assert dst["use_same_buffer_for_input_output"]
+    assert dst["use_growable_memory"]
+    assert dst["initial_state"][0]["name"] == "init0"
+    assert dst["initial_state"][0]["data_choice"] == 1  # ZERO
+    assert dst["initial_state"][0]["zero_data"]
+
+    dw = d["model_warmup"][0]
+    assert dw["name"] == "warm0" and dw["batch_size"] == 8
+    assert dw["count"] == 3
+    assert dw["inputs"]["IN0"]["dims"] == [3, 224, 224]
+    assert dw["inputs"]["IN0"]["data_choice"] == 2  # RANDOM
+    assert dw["inputs"]["IN0"]["random_data"]
+
+    assert d["batch_input"][0]["kind"] == 3
+    assert d["batch_input"][0]["target_name"] == ["RAGGED_SHAPE"]
+    assert d["batch_input"][0]["source_input"] == ["IN0"]
+    assert d["batch_output"][0]["target_name"] == ["OUT0"]
+
+    assert d["op_library_filename"] == ["libops.so"]
+    assert d["decoupled"]
+    assert d["repository_agents"][0]["name"] == "checksum"
+    assert d["repository_agents"][0]["parameters"] == {"algo": "sha256"}
+    assert d["response_cache_enable"]
+    assert d["metric_control"][0]["family"] == "nv_inference_count"
+    assert d["metric_control"][0]["histogram_buckets"] == [0.5, 1.0, 2.5]
