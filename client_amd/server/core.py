@@ -212,6 +212,16 @@ class InferenceCore:
         }
         self.live = True
         self.ready = True
+        self._copy_stream = None
+        self._copy_stream_lock = threading.Lock()
+
+    def _get_copy_stream(self, torch):
+        """Dedicated stream for output-region copies (shared; the copies
+        are tiny and serialize harmlessly)."""
+        with self._copy_stream_lock:
+            if self._copy_stream is None:
+                self._copy_stream = torch.cuda.Stream()
+            return self._copy_stream
 
     # ---- repository ----
     def add_model(self, model, ready=True):
@@ -441,7 +451,10 @@ class InferenceCore:
         for io in req_inputs:
             tensors.append(region_tensor(io, io["datatype"], io["shape"]))
         t1 = time.monotonic_ns()
-        results = model.execute_torch(tensors)
+        if hasattr(model, "execute_torch_async"):
+            results, done_ev = model.execute_torch_async(tensors)
+        else:
+            results, done_ev = model.execute_torch(tensors), None
         t2 = time.monotonic_ns()
 
         model_dtypes = {n: d for n, d, _ in model.outputs}
@@ -458,32 +471,58 @@ class InferenceCore:
             "UINT8": torch.uint8, "FP64": torch.float64, "BOOL": torch.bool,
         }
         by_name = {n: r for (n, _, _), r in zip(model.outputs, results)}
-        for io in req_outputs:
-            name = io["name"]
-            if name not in by_name:
-                raise InferenceError(
-                    f"unexpected inference output '{name}' for model "
-                    f"'{model.name}'"
+        # output copies ride a dedicated copy stream ordered on the
+        # model's completion event, and the request host-waits only its
+        # OWN copy event — a full torch.cuda.synchronize() here queued
+        # every request's ~16 KB output copy behind the next batches'
+        # forwards on the shared default stream (5-6 ms of
+        # avg_compute_output at c8, DenseNet — ROUND2_NOTES item 7)
+        copy_stream = None
+        if done_ev is not None:
+            copy_stream = self._get_copy_stream(torch)
+        import contextlib
+
+        ctx = (torch.cuda.stream(copy_stream) if copy_stream is not None
+               else contextlib.nullcontext())
+        with ctx:
+            if copy_stream is not None:
+                copy_stream.wait_event(done_ev)
+            for io in req_outputs:
+                name = io["name"]
+                if name not in by_name:
+                    raise InferenceError(
+                        f"unexpected inference output '{name}' for model "
+                        f"'{model.name}'"
+                    )
+                result = by_name[name]
+                datatype = model_dtypes[name]
+                out_view = region_tensor(
+                    {"parameters": io["parameters"]}, datatype,
+                    list(result.shape)
                 )
-            result = by_name[name]
-            datatype = model_dtypes[name]
-            out_view = region_tensor(
-                {"parameters": io["parameters"]}, datatype, list(result.shape)
-            )
-            out_view.copy_(result.to(torch_dt[datatype]))
-            params = io["parameters"]
-            response["outputs"].append({
-                "name": name,
-                "datatype": datatype,
-                "shape": list(result.shape),
-                "parameters": {
-                    "shared_memory_region": params["shared_memory_region"],
-                    "shared_memory_byte_size": params["shared_memory_byte_size"],
-                    **({"shared_memory_offset": params["shared_memory_offset"]}
-                       if params.get("shared_memory_offset") else {}),
-                },
-            })
-        torch.cuda.synchronize()
+                out_view.copy_(result.to(torch_dt[datatype]))
+                params = io["parameters"]
+                response["outputs"].append({
+                    "name": name,
+                    "datatype": datatype,
+                    "shape": list(result.shape),
+                    "parameters": {
+                        "shared_memory_region":
+                            params["shared_memory_region"],
+                        "shared_memory_byte_size":
+                            params["shared_memory_byte_size"],
+                        **({"shared_memory_offset":
+                            params["shared_memory_offset"]}
+                           if params.get("shared_memory_offset") else {}),
+                    },
+                })
+            if copy_stream is not None:
+                copied = torch.cuda.Event()
+                copied.record(copy_stream)
+        if copy_stream is not None:
+            copied.synchronize()
+        else:
+            torch.cuda.synchronize()
         return response, [], t1 - t0, t2 - t1
 
     def _build_response(self, model, request, result, parameters):

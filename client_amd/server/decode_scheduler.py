@@ -43,7 +43,8 @@ class DecodeScheduler:
     END = object()
 
     def __init__(self, model, max_batch=8, device="cuda:0", dtype=None,
-                 use_graph=None, len_bucket=256, prefill_chunk=256):
+                 use_graph=None, len_bucket=256, prefill_chunk=256,
+                 prefill_rows_per_step=2):
         self.model = model
         self.device = device
         self.dtype = dtype if dtype is not None else next(
@@ -82,6 +83,12 @@ class DecodeScheduler:
         # and made the stall WORSE.
         self._pf_bufs = {}    # group_size -> dict of persistent tensors
         self._pf_graphs = {}  # (group_size, bucket) -> (graph, out)
+        # cap prefill rows advanced per loop iteration: a start burst
+        # admitting max_batch prompts at once would otherwise run one
+        # [8, C] replay (~45 ms measured) in a single decode gap;
+        # spreading the rows keeps every ITL near one decode step while
+        # the burst drains over a few iterations
+        self.prefill_rows_per_step = prefill_rows_per_step
         self._pending = queue.Queue()
         self._cv = threading.Condition()
         self._alive = True
@@ -105,16 +112,15 @@ class DecodeScheduler:
         for i in range(1, n_buckets + 1):
             bucket = min(i * self.len_bucket, self.model.cfg.max_seq)
             self._get_graph(bucket)
-            # group=1 covers the common case (one admission at a time);
-            # other group sizes capture lazily on first use
-            self._get_prefill_graph(1, bucket)
+            # capture every group size the scheduler can actually issue
+            # (1..prefill_rows_per_step, powers of two) — an uncaptured
+            # (group, bucket) pair costs ~0.5 s in some stream's TTFT
+            g = 1
+            while g <= max(1, self.prefill_rows_per_step):
+                self._get_prefill_graph(g, bucket)
+                g *= 2
             if bucket >= self.model.cfg.max_seq:
                 break
-        # a full-concurrency start burst admits max_batch prompts in one
-        # iteration (first bucket) — capture that shape up front too
-        self._get_prefill_graph(self._group_size(self.max_batch),
-                                min(self.len_bucket,
-                                    self.model.cfg.max_seq))
 
     def shutdown(self):
         with self._cv:
@@ -220,9 +226,13 @@ class DecodeScheduler:
         if not pf:
             return
         c = self.prefill_chunk
+        # advance at most prefill_rows_per_step rows THIS iteration;
+        # the rest progress on subsequent loop iterations (each
+        # interleaved with a decode step)
+        todo = pf[:max(1, self.prefill_rows_per_step)]
         done = 0
-        while done < len(pf):
-            group_slots = pf[done:done + self.max_batch]
+        while done < len(todo):
+            group_slots = todo[done:done + self.max_batch]
             group = min(self._group_size(len(group_slots)),
                         self.max_batch)
             group_slots = group_slots[:group]
