@@ -43,7 +43,7 @@ class DecodeScheduler:
     END = object()
 
     def __init__(self, model, max_batch=8, device="cuda:0", dtype=None,
-                 use_graph=None, len_bucket=256, prefill_chunk=256,
+                 use_graph=None, len_bucket=256, prefill_chunk=128,
                  prefill_rows_per_step=2):
         self.model = model
         self.device = device

@@ -208,6 +208,15 @@ class GrpcServer:
             o.data_type = _DT_ENUM.get(d, 0)
             o.dims.extend(s)
         cfg.model_transaction_policy.decoupled = model.decoupled
+        # load-time config override is visible in the served config
+        # (same semantics as the HTTP config route)
+        override = self.core.config_overrides.get(model.name, {})
+        if "max_batch_size" in override:
+            cfg.max_batch_size = int(override["max_batch_size"])
+        if "backend" in override:
+            cfg.backend = override["backend"]
+        if "platform" in override:
+            cfg.platform = override["platform"]
         return resp
 
     def ModelInfer(self, request, context):
@@ -303,7 +312,15 @@ class GrpcServer:
         return resp
 
     def RepositoryModelLoad(self, request, context):
-        self.core.load_model(request.model_name)
+        config = None
+        files = {}
+        for key, param in request.parameters.items():
+            if key == "config":
+                config = param.string_param
+            else:
+                files[key] = param.bytes_param
+        self.core.load_model(request.model_name, config=config,
+                             files=files or None)
         return service_pb2.RepositoryModelLoadResponse()
 
     def RepositoryModelUnload(self, request, context):

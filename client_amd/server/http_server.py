@@ -9,6 +9,7 @@ trace/log settings, and system/cuda(HIP) shared-memory registration
 """
 
 import asyncio
+import base64
 import gzip
 import json
 import re
@@ -186,7 +187,7 @@ class HttpServer:
         m = _ROUTE_CONFIG.match(path)
         if m:
             model = core.get_model(m.group(1), must_be_ready=False)
-            return 200, {}, json.dumps(model.config()).encode()
+            return 200, {}, json.dumps(core.model_config_dict(model)).encode()
         m = _ROUTE_STATS.match(path)
         if m:
             return 200, {}, json.dumps(core.statistics(m.group(1))).encode()
@@ -200,7 +201,20 @@ class HttpServer:
             return 200, {}, json.dumps(core.repository_index()).encode()
         m = _ROUTE_LOAD.match(path)
         if m and method == "POST":
-            core.load_model(m.group(1))
+            config = None
+            files = None
+            if body:
+                try:
+                    params = json.loads(body.decode()).get("parameters", {})
+                except Exception:
+                    params = {}
+                config = params.get("config")
+                files = {
+                    k: base64.b64decode(v)
+                    for k, v in params.items()
+                    if k != "config" and isinstance(v, str)
+                } or None
+            core.load_model(m.group(1), config=config, files=files)
             return 200, {}, b"{}"
         m = _ROUTE_UNLOAD.match(path)
         if m and method == "POST":

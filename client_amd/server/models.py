@@ -81,6 +81,10 @@ class Model:
 class IdentityModel(Model):
     """Echo each INPUTi to OUTPUTi. identity_fp32 fixture analog."""
 
+    # the fixture identity accepts any rank (its [-1] dims are a
+    # placeholder, not a 1-D contract) — skip server shape validation
+    lax_shapes = True
+
     def __init__(self, name="identity_fp32", datatype="FP32", n_io=1):
         ios = [("INPUT" + str(i) if n_io > 1 else "INPUT0", datatype, [-1])
                for i in range(n_io)]

@@ -480,10 +480,19 @@ Error InferenceServerGrpcClient::ModelRepositoryIndex(
   return Error::Success;
 }
 
-Error InferenceServerGrpcClient::LoadModel(const std::string& model_name) {
+Error InferenceServerGrpcClient::LoadModel(
+    const std::string& model_name, const std::string& config,
+    const std::map<std::string, std::vector<char>>& files) {
+  std::map<std::string, std::string> file_blobs;
+  for (const auto& [path, content] : files) {
+    file_blobs.emplace(path,
+                       std::string(content.data(), content.size()));
+  }
   std::string resp;
   return UnaryCall(
-      "RepositoryModelLoad", kserve::EncodeRepositoryModelRequest(model_name),
+      "RepositoryModelLoad",
+      kserve::EncodeRepositoryModelLoadRequest(model_name, config,
+                                               file_blobs),
       &resp);
 }
 

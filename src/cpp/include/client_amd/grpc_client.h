@@ -66,7 +66,11 @@ class InferenceServerGrpcClient : public InferenceServerClient {
       const std::string& model_version = "");
   Error ModelRepositoryIndex(
       std::vector<kserve::RepositoryIndexEntryPb>* index);
-  Error LoadModel(const std::string& model_name);
+  // optional config-override JSON + file-override blobs (reference
+  // grpc_client.h LoadModel signature)
+  Error LoadModel(
+      const std::string& model_name, const std::string& config = "",
+      const std::map<std::string, std::vector<char>>& files = {});
   Error UnloadModel(const std::string& model_name);
   // Reference grpc_client.h:330-352 trace-settings RPCs (the reference
   // has no gRPC log-settings; ours adds none either — Python covers it).

@@ -384,6 +384,12 @@ std::vector<RepositoryIndexEntryPb> DecodeRepositoryIndex(
 // model_name = field 2 for load/unload requests
 std::string EncodeRepositoryModelRequest(const std::string& model_name);
 
+// Load with optional config-override JSON and file-override blobs
+// (ModelRepositoryParameter string_param / bytes_param map)
+std::string EncodeRepositoryModelLoadRequest(
+    const std::string& model_name, const std::string& config,
+    const std::map<std::string, std::string>& files);
+
 // shm management
 std::string EncodeSystemShmRegister(const std::string& name,
                                     const std::string& key, uint64_t offset,

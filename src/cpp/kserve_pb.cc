@@ -299,6 +299,28 @@ std::string EncodeRepositoryModelRequest(const std::string& model_name) {
   return w.out;
 }
 
+std::string EncodeRepositoryModelLoadRequest(
+    const std::string& model_name, const std::string& config,
+    const std::map<std::string, std::string>& files) {
+  // RepositoryModelLoadRequest { model_name = 2;
+  //   map<string, ModelRepositoryParameter> parameters = 3; }
+  // ModelRepositoryParameter oneof: string_param = 3, bytes_param = 4
+  Writer w;
+  w.put_str(2, model_name);
+  auto put_param = [&w](const std::string& key, int oneof_field,
+                        const std::string& val) {
+    Writer param;
+    param.put_str(oneof_field, val);
+    Writer entry;
+    entry.put_str(1, key);
+    entry.put_msg(2, param.out);
+    w.put_msg(3, entry.out);
+  };
+  if (!config.empty()) put_param("config", 3, config);
+  for (const auto& [path, content] : files) put_param(path, 4, content);
+  return w.out;
+}
+
 std::string EncodeSystemShmRegister(const std::string& name,
                                     const std::string& key, uint64_t offset,
                                     uint64_t byte_size) {

@@ -1,0 +1,405 @@
+// Dual-protocol parameterized suite — the reference's typed
+// ClientTest<ClientType> matrix (reference cc_client_test.cc:42-129,
+// 300-1350) rebuilt as one template over BOTH clients: the InferMulti /
+// AsyncInferMulti option+output matrices, the mismatch error cases, the
+// option/output validation errors, and load-with-config/file-override.
+//
+// Usage: cc_dual_test <http_host:port> <grpc_host:port>
+// (tests/test_cpp_client.py launches both fixture servers.)
+#include <atomic>
+#include <chrono>
+#include <condition_variable>
+#include <cstring>
+#include <iostream>
+#include <map>
+#include <mutex>
+#include <vector>
+
+#include "client_amd/common.h"
+#include "client_amd/grpc_client.h"
+#include "client_amd/http_client.h"
+#include "client_amd/json.h"
+
+using namespace client_amd;
+
+static int g_failures = 0;
+
+#define CHECK(cond)                                                       \
+  do {                                                                    \
+    if (!(cond)) {                                                        \
+      std::cerr << "FAILED at " << __LINE__ << ": " #cond << std::endl;   \
+      ++g_failures;                                                       \
+    }                                                                     \
+  } while (0)
+
+#define CHECK_OK(expr)                                                 \
+  do {                                                                 \
+    Error _e = (expr);                                                 \
+    if (!_e.IsOk()) {                                                  \
+      std::cerr << "FAILED at " << __LINE__ << ": " << _e.Message()    \
+                << std::endl;                                          \
+      ++g_failures;                                                    \
+    }                                                                  \
+  } while (0)
+
+#define CHECK_ERR(expr)                                                  \
+  do {                                                                   \
+    Error _e = (expr);                                                   \
+    if (_e.IsOk()) {                                                     \
+      std::cerr << "FAILED at " << __LINE__                              \
+                << ": expected error from " #expr << std::endl;          \
+      ++g_failures;                                                      \
+    }                                                                    \
+  } while (0)
+
+// ---- small per-protocol traits (only where the APIs differ) ----
+
+static int GetMaxBatchSize(InferenceServerHttpClient* c,
+                           const std::string& model) {
+  std::string cfg;
+  Error e = c->ModelConfig(&cfg, model);
+  if (!e.IsOk()) return -1;
+  return (int)Json::Parse(cfg)["max_batch_size"].AsInt();
+}
+
+static int GetMaxBatchSize(InferenceServerGrpcClient* c,
+                           const std::string& model) {
+  kserve::ModelConfigPb cfg;
+  Error e = c->ModelConfig(&cfg, model);
+  if (!e.IsOk()) return -1;
+  return cfg.max_batch_size;
+}
+
+static Error LoadOverride(InferenceServerHttpClient* c,
+                          const std::string& model,
+                          const std::string& config,
+                          const std::map<std::string, std::vector<char>>&
+                              files) {
+  return c->LoadModel(model, {}, config, files);
+}
+
+static Error LoadOverride(InferenceServerGrpcClient* c,
+                          const std::string& model,
+                          const std::string& config,
+                          const std::map<std::string, std::vector<char>>&
+                              files) {
+  return c->LoadModel(model, config, files);
+}
+
+// ---- shared request fixture (the "simple" addsub model) ----
+
+struct AddSubRequest {
+  std::vector<int32_t> in0, in1;
+  InferInput* input0 = nullptr;
+  InferInput* input1 = nullptr;
+  InferRequestedOutput* out0 = nullptr;
+  InferRequestedOutput* out1 = nullptr;
+
+  explicit AddSubRequest(int seed = 0) : in0(16), in1(16) {
+    for (int i = 0; i < 16; ++i) {
+      in0[i] = i + seed;
+      in1[i] = 2 * i + seed;
+    }
+    InferInput::Create(&input0, "INPUT0", {1, 16}, "INT32");
+    InferInput::Create(&input1, "INPUT1", {1, 16}, "INT32");
+    input0->AppendRaw(reinterpret_cast<uint8_t*>(in0.data()),
+                      in0.size() * 4);
+    input1->AppendRaw(reinterpret_cast<uint8_t*>(in1.data()),
+                      in1.size() * 4);
+    InferRequestedOutput::Create(&out0, "OUTPUT0");
+    InferRequestedOutput::Create(&out1, "OUTPUT1");
+  }
+  ~AddSubRequest() {
+    delete input0;
+    delete input1;
+    delete out0;
+    delete out1;
+  }
+  std::vector<InferInput*> inputs() const { return {input0, input1}; }
+  std::vector<const InferRequestedOutput*> outputs() const {
+    return {out0, out1};
+  }
+
+  // OUTPUT0 = INPUT0 + INPUT1; OUTPUT1 = INPUT0 - INPUT1
+  bool CheckResult(InferResult* result, bool expect_out0 = true,
+                   bool expect_out1 = true) const {
+    if (result == nullptr || !result->RequestStatus().IsOk()) return false;
+    if (expect_out0) {
+      const uint8_t* buf;
+      size_t n;
+      if (!result->RawData("OUTPUT0", &buf, &n).IsOk() || n != 64)
+        return false;
+      const int32_t* v = reinterpret_cast<const int32_t*>(buf);
+      for (int i = 0; i < 16; ++i)
+        if (v[i] != in0[i] + in1[i]) return false;
+    }
+    if (expect_out1) {
+      const uint8_t* buf;
+      size_t n;
+      if (!result->RawData("OUTPUT1", &buf, &n).IsOk() || n != 64)
+        return false;
+      const int32_t* v = reinterpret_cast<const int32_t*>(buf);
+      for (int i = 0; i < 16; ++i)
+        if (v[i] != in0[i] - in1[i]) return false;
+    }
+    return true;
+  }
+};
+
+template <typename Client>
+static void WaitMulti(Client* client,
+                      const std::vector<InferOptions>& options,
+                      const std::vector<std::vector<InferInput*>>& inputs,
+                      const std::vector<std::vector<
+                          const InferRequestedOutput*>>& outputs,
+                      std::vector<InferResult*>* results, Error* err) {
+  std::mutex mu;
+  std::condition_variable cv;
+  bool done = false;
+  Error async_err = client->AsyncInferMulti(
+      [&](std::vector<InferResult*> res) {
+        std::lock_guard<std::mutex> lk(mu);
+        *results = res;
+        done = true;
+        cv.notify_one();
+      },
+      options, inputs, outputs);
+  if (!async_err.IsOk()) {
+    *err = async_err;
+    return;
+  }
+  std::unique_lock<std::mutex> lk(mu);
+  if (!cv.wait_for(lk, std::chrono::seconds(30), [&] { return done; })) {
+    *err = Error("AsyncInferMulti timed out");
+    return;
+  }
+  *err = Error::Success;
+}
+
+template <typename Client>
+static void RunSuite(Client* client, const char* proto) {
+  std::cerr << "=== dual suite: " << proto << " ===" << std::endl;
+  InferOptions options("simple");
+
+  // -- InferMulti: shared option/output set over 3 requests
+  {
+    AddSubRequest r0(0), r1(5), r2(11);
+    std::vector<InferResult*> results;
+    CHECK_OK(client->InferMulti(
+        &results, {options},
+        {r0.inputs(), r1.inputs(), r2.inputs()},
+        {r0.outputs(), r1.outputs(), r2.outputs()}));
+    CHECK(results.size() == 3);
+    CHECK(r0.CheckResult(results[0]));
+    CHECK(r1.CheckResult(results[1]));
+    CHECK(r2.CheckResult(results[2]));
+    for (auto* r : results) delete r;
+  }
+
+  // -- InferMulti: per-request options (distinct request ids)
+  {
+    AddSubRequest r0(1), r1(2);
+    InferOptions o1("simple"), o2("simple");
+    o1.request_id_ = "multi-1";
+    o2.request_id_ = "multi-2";
+    std::vector<InferResult*> results;
+    CHECK_OK(client->InferMulti(&results, {o1, o2},
+                                {r0.inputs(), r1.inputs()},
+                                {r0.outputs(), r1.outputs()}));
+    CHECK(results.size() == 2);
+    std::string id;
+    CHECK(results[0]->Id(&id).IsOk() && id == "multi-1");
+    CHECK(results[1]->Id(&id).IsOk() && id == "multi-2");
+    CHECK(r0.CheckResult(results[0]));
+    CHECK(r1.CheckResult(results[1]));
+    for (auto* r : results) delete r;
+  }
+
+  // -- InferMulti: different outputs per request (OUTPUT0-only vs
+  //    OUTPUT1-only)
+  {
+    AddSubRequest r0(3), r1(4);
+    std::vector<InferResult*> results;
+    CHECK_OK(client->InferMulti(
+        &results, {options}, {r0.inputs(), r1.inputs()},
+        {{r0.out0}, {r1.out1}}));
+    CHECK(results.size() == 2);
+    CHECK(r0.CheckResult(results[0], true, false));
+    CHECK(r1.CheckResult(results[1], false, true));
+    // the un-requested output must be absent
+    const uint8_t* buf;
+    size_t n;
+    CHECK(!results[0]->RawData("OUTPUT1", &buf, &n).IsOk() || n == 0);
+    for (auto* r : results) delete r;
+  }
+
+  // -- InferMulti: ONE shared output set for N requests
+  {
+    AddSubRequest r0(6), r1(7);
+    std::vector<InferResult*> results;
+    CHECK_OK(client->InferMulti(&results, {options},
+                                {r0.inputs(), r1.inputs()},
+                                {r0.outputs()}));
+    CHECK(results.size() == 2);
+    CHECK(r0.CheckResult(results[0]));
+    for (auto* r : results) delete r;
+  }
+
+  // -- InferMulti: no outputs requested -> server returns all
+  {
+    AddSubRequest r0(8);
+    std::vector<InferResult*> results;
+    CHECK_OK(client->InferMulti(&results, {options}, {r0.inputs()}, {}));
+    CHECK(results.size() == 1);
+    CHECK(r0.CheckResult(results[0]));
+    for (auto* r : results) delete r;
+  }
+
+  // -- InferMulti: mismatched option count -> client-side error
+  {
+    AddSubRequest r0(9), r1(10), r2(12);
+    InferOptions o1("simple"), o2("simple");
+    std::vector<InferResult*> results;
+    CHECK_ERR(client->InferMulti(
+        &results, {o1, o2},
+        {r0.inputs(), r1.inputs(), r2.inputs()}, {}));
+  }
+
+  // -- InferMulti: mismatched output count -> client-side error
+  {
+    AddSubRequest r0(13), r1(14), r2(15);
+    std::vector<InferResult*> results;
+    CHECK_ERR(client->InferMulti(
+        &results, {options},
+        {r0.inputs(), r1.inputs(), r2.inputs()},
+        {r0.outputs(), r1.outputs()}));
+  }
+
+  // -- AsyncInferMulti: shared options, countdown join
+  {
+    AddSubRequest r0(16), r1(17), r2(18);
+    std::vector<InferResult*> results;
+    Error err;
+    WaitMulti(client, {options},
+              {r0.inputs(), r1.inputs(), r2.inputs()},
+              {r0.outputs(), r1.outputs(), r2.outputs()}, &results, &err);
+    CHECK_OK(err);
+    CHECK(results.size() == 3);
+    CHECK(r0.CheckResult(results[0]));
+    CHECK(r1.CheckResult(results[1]));
+    CHECK(r2.CheckResult(results[2]));
+    for (auto* r : results) delete r;
+  }
+
+  // -- AsyncInferMulti: per-request outputs
+  {
+    AddSubRequest r0(19), r1(20);
+    std::vector<InferResult*> results;
+    Error err;
+    WaitMulti(client, {options}, {r0.inputs(), r1.inputs()},
+              {{r0.out0}, {r1.out1}}, &results, &err);
+    CHECK_OK(err);
+    CHECK(results.size() == 2);
+    CHECK(r0.CheckResult(results[0], true, false));
+    CHECK(r1.CheckResult(results[1], false, true));
+    for (auto* r : results) delete r;
+  }
+
+  // -- AsyncInferMulti: mismatch -> immediate client-side error
+  {
+    AddSubRequest r0(21), r1(22);
+    InferOptions o1("simple"), o2("simple"), o3("simple");
+    CHECK_ERR(client->AsyncInferMulti(
+        [](std::vector<InferResult*>) {}, {o1, o2, o3},
+        {r0.inputs(), r1.inputs()}, {}));
+  }
+
+  // -- Infer with an unknown requested output -> server error
+  {
+    AddSubRequest r0(23);
+    InferRequestedOutput* bad = nullptr;
+    InferRequestedOutput::Create(&bad, "NO_SUCH_OUTPUT");
+    InferResult* result = nullptr;
+    Error err = client->Infer(&result, options, r0.inputs(), {bad});
+    bool errored = !err.IsOk() ||
+                   (result && !result->RequestStatus().IsOk());
+    CHECK(errored);
+    delete bad;
+    delete result;
+  }
+
+  // -- Infer with a wrong-shape input -> server error
+  {
+    std::vector<int32_t> data(8, 1);
+    InferInput* short0 = nullptr;
+    InferInput* short1 = nullptr;
+    InferInput::Create(&short0, "INPUT0", {1, 8}, "INT32");
+    InferInput::Create(&short1, "INPUT1", {1, 8}, "INT32");
+    short0->AppendRaw(reinterpret_cast<uint8_t*>(data.data()), 32);
+    short1->AppendRaw(reinterpret_cast<uint8_t*>(data.data()), 32);
+    InferResult* result = nullptr;
+    Error err = client->Infer(&result, options, {short0, short1});
+    bool errored = !err.IsOk() ||
+                   (result && !result->RequestStatus().IsOk());
+    CHECK(errored);
+    delete short0;
+    delete short1;
+    delete result;
+  }
+
+  // -- LoadWithConfigOverride: served config reflects the override,
+  //    plain reload clears it (reference cc_client_test.cc:1306-1350)
+  {
+    int orig = GetMaxBatchSize(client, "simple");
+    CHECK(orig >= 0);
+    CHECK_OK(LoadOverride(client, "simple",
+                          "{\"max_batch_size\": 57}", {}));
+    CHECK(GetMaxBatchSize(client, "simple") == 57);
+    CHECK_OK(client->LoadModel("simple"));
+    CHECK(GetMaxBatchSize(client, "simple") == orig);
+  }
+
+  // -- LoadWithFileOverride: file without config rejected; with config
+  //    accepted (reference cc_client_test.cc:1202-1305)
+  {
+    std::vector<char> blob = {'\x01', '\x02', '\x03'};
+    std::map<std::string, std::vector<char>> files{
+        {"file:1/model.bin", blob}};
+    CHECK_ERR(LoadOverride(client, "simple", "", files));
+    CHECK_OK(LoadOverride(client, "simple",
+                          "{\"max_batch_size\": 9}", files));
+    CHECK(GetMaxBatchSize(client, "simple") == 9);
+    CHECK_OK(client->LoadModel("simple"));
+  }
+}
+
+int main(int argc, char** argv) {
+  std::string http_url = argc > 1 ? argv[1] : "127.0.0.1:8000";
+  std::string grpc_url = argc > 2 ? argv[2] : "127.0.0.1:8001";
+
+  {
+    std::unique_ptr<InferenceServerHttpClient> http_client;
+    Error e = InferenceServerHttpClient::Create(&http_client, http_url);
+    if (!e.IsOk()) {
+      std::cerr << "http create failed: " << e.Message() << std::endl;
+      return 1;
+    }
+    RunSuite(http_client.get(), "http");
+  }
+  {
+    std::unique_ptr<InferenceServerGrpcClient> grpc_client;
+    Error e = InferenceServerGrpcClient::Create(&grpc_client, grpc_url);
+    if (!e.IsOk()) {
+      std::cerr << "grpc create failed: " << e.Message() << std::endl;
+      return 1;
+    }
+    RunSuite(grpc_client.get(), "grpc");
+  }
+
+  if (g_failures) {
+    std::cerr << g_failures << " dual-suite check(s) failed" << std::endl;
+    return 1;
+  }
+  std::cout << "ALL PASSED" << std::endl;
+  return 0;
+}

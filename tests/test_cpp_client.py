@@ -704,3 +704,29 @@ def test_modelconfig_cross_decode():
     assert d["response_cache_enable"]
     assert d["metric_control"][0]["family"] == "nv_inference_count"
     assert d["metric_control"][0]["histogram_buckets"] == [0.5, 1.0, 2.5]
+
+
+@pytest.fixture(scope="module")
+def cc_dual_binary():
+    try:
+        return _compile("cc_dual_test", CPP / "tests" / "cc_dual_test.cc")
+    except subprocess.CalledProcessError as e:
+        pytest.fail(f"C++ compile failed:\n{e.stderr}")
+
+
+def test_cc_dual_suite(cc_dual_binary, http_fixture_server,
+                       grpc_fixture_server):
+    """The reference's typed dual-protocol test matrix (InferMulti /
+    AsyncInferMulti option+output combinations, mismatch errors,
+    unknown-output and wrong-shape server errors, load with
+    config/file override) run against BOTH clients from one
+    parameterized C++ suite (reference cc_client_test.cc:42-129,
+    300-1350)."""
+    hhost, hport, _ = http_fixture_server
+    ghost, gport, _ = grpc_fixture_server
+    proc = subprocess.run(
+        [str(cc_dual_binary), f"{hhost}:{hport}", f"{ghost}:{gport}"],
+        capture_output=True, text=True, timeout=180,
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "ALL PASSED" in proc.stdout
