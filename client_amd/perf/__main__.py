@@ -17,6 +17,12 @@ def main(argv=None):
     p.add_argument("-b", "--batch-size", type=int, default=1)
     p.add_argument("--concurrency-range", default=None,
                    help="start:end:step (closed loop; default 1:4:1)")
+    p.add_argument("--periodic-concurrency-range", default=None,
+                   help="start:end:step — ONE run whose concurrency "
+                        "ramps by step every --request-period seconds "
+                        "(per-stage stats, connections kept up)")
+    p.add_argument("--request-period", type=float, default=2.0,
+                   help="seconds per stage of the periodic ramp")
     p.add_argument("--request-rate-range", default=None,
                    help="start:end:step req/s (open loop; overrides "
                         "--concurrency-range)")
@@ -79,7 +85,20 @@ def main(argv=None):
         input_data=args.input_data,
         percentile_q=args.percentile,
     )
-    if rate_list is not None:
+    if args.periodic_concurrency_range is not None:
+        lo, _, rest = args.periodic_concurrency_range.partition(":")
+        hi, _, st = rest.partition(":")
+        results = pa.run_periodic(
+            int(lo), int(hi or lo), int(st or 1),
+            period_s=args.request_period, warmup_s=args.warmup,
+        )
+        for r in results:
+            print(
+                f"Ramp concurrency: {r['concurrency']}, throughput: "
+                f"{r['inferences_per_sec']} infer/sec, latency p99: "
+                f"{r['latency_us']['p99']} usec"
+            )
+    elif rate_list is not None:
         results = pa.run_request_rate(
             rate_list,
             warmup_s=args.warmup,

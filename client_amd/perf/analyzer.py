@@ -117,6 +117,77 @@ class ConcurrencyDriver:
         return throughput, all_lat, total_err, len(window_results)
 
 
+class PeriodicConcurrencyDriver:
+    """Ramped closed-loop load (perf_analyzer
+    --periodic-concurrency-range): ONE run during which the number of
+    concurrent workers grows from ``start`` to ``end`` by ``step``
+    every ``period_s`` seconds, recording a separate measurement stage
+    per concurrency level. Shows how the server's latency/throughput
+    curve bends as load rises without tearing the connections down
+    between levels (the ordinary sweep reconnects per level)."""
+
+    def __init__(self, issue_fn, start, end, step, period_s):
+        self._issue_fn = issue_fn
+        self._start = start
+        self._end = end
+        self._step = max(1, step)
+        self._period_s = period_s
+        self._recorder = LatencyRecorder()
+        self._stop = threading.Event()
+        self._lock = threading.Lock()
+        self._active = 0
+        self._done = threading.Event()
+
+    def _worker(self, slot):
+        while not self._stop.is_set():
+            start = time.monotonic_ns()
+            try:
+                self._issue_fn(slot)
+                self._recorder.record(start, time.monotonic_ns())
+            except Exception as e:
+                self._recorder.record(start, time.monotonic_ns(), e)
+                time.sleep(0.01)
+        with self._lock:
+            self._active -= 1
+            if self._active == 0:
+                self._done.set()
+
+    def run(self, warmup_s=1.0):
+        """Returns a list of per-stage dicts:
+        (concurrency, throughput_req_s, latencies_ns_sorted, errors)."""
+        stages = []
+        level = self._start
+        slot = 0
+        with self._lock:
+            self._active = 0
+        while True:
+            # bring up the delta workers for this level
+            while slot < level:
+                with self._lock:
+                    self._active += 1
+                threading.Thread(target=self._worker, args=(slot,),
+                                 daemon=True).start()
+                slot += 1
+            if level == self._start and warmup_s > 0:
+                time.sleep(warmup_s)
+            self._recorder.snapshot_and_reset()
+            time.sleep(self._period_s)
+            lat, err = self._recorder.snapshot_and_reset()
+            lat.sort()
+            stages.append({
+                "concurrency": level,
+                "throughput": len(lat) / self._period_s,
+                "latencies_ns": lat,
+                "errors": err,
+            })
+            if level >= self._end:
+                break
+            level = min(level + self._step, self._end)
+        self._stop.set()
+        self._done.wait(timeout=60)
+        return stages
+
+
 class RequestRateDriver:
     """Open-loop load (perf_analyzer --request-rate-range): a request
     schedule at ``rate`` req/s with constant or Poisson gaps, executed
@@ -567,6 +638,35 @@ class PerfAnalyzer:
                     self._teardown_hipshm(client)
             finally:
                 client.close()
+        return results
+
+    def run_periodic(self, start, end, step, period_s=2.0, warmup_s=1.0):
+        """Ramped single-run concurrency profile (perf_analyzer
+        --periodic-concurrency-range start:end:step with a time-based
+        request period); returns one result dict per stage."""
+        client, mod = self._make_client(end)
+        results = []
+        try:
+            issue = self._build_issue(client, mod, end)
+            driver = PeriodicConcurrencyDriver(issue, start, end, step,
+                                               period_s)
+            stats_before = self._server_stats_snapshot(client)
+            stages = driver.run(warmup_s=warmup_s)
+            server = self._server_breakdown(
+                stats_before, self._server_stats_snapshot(client))
+            for st in stages:
+                result = self._result_dict(
+                    st["throughput"], st["latencies_ns"], st["errors"], 1,
+                    concurrency=st["concurrency"], ramped=True,
+                    **({"server": server} if server else {}),
+                )
+                results.append(result)
+                if self.verbose:
+                    print(result)
+            if self.shared_memory in ("cuda", "hip"):
+                self._teardown_hipshm(client)
+        finally:
+            client.close()
         return results
 
     def run_request_rate(self, rate_list, warmup_s=1.0, window_s=2.0,

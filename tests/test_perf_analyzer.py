@@ -180,3 +180,42 @@ def test_perf_server_breakdown(grpc_fixture_server):
     for k in ("avg_queue_us", "avg_compute_input_us",
               "avg_compute_infer_us", "avg_compute_output_us"):
         assert srv[k] >= 0
+
+
+def test_perf_periodic_ramp(grpc_fixture_server):
+    """--periodic-concurrency-range analog: one run, concurrency ramps
+    1 -> 3 by 1 per period, one result stage per level."""
+    host, port, _ = grpc_fixture_server
+    pa = PerfAnalyzer(
+        url=f"{host}:{port}", protocol="grpc", model_name="simple",
+        batch_size=1,
+    )
+    results = pa.run_periodic(1, 3, 1, period_s=0.3, warmup_s=0.1)
+    assert [r["concurrency"] for r in results] == [1, 2, 3]
+    for r in results:
+        assert r["errors"] == 0
+        assert r["ramped"] is True
+        assert r["request_rate_per_sec"] > 0
+
+
+def test_perf_cli_periodic(grpc_fixture_server, tmp_path, capsys):
+    import json as _json
+
+    import client_amd.perf.__main__ as cli
+
+    host, port, _ = grpc_fixture_server
+    out = tmp_path / "ramp.json"
+    csv = tmp_path / "ramp.csv"
+    cli.main([
+        "-m", "simple", "-u", f"{host}:{port}", "-i", "grpc",
+        "--periodic-concurrency-range", "1:2:1",
+        "--request-period", "0.3", "--warmup", "0.1",
+        "--json", str(out), "-f", str(csv),
+    ])
+    data = _json.loads(out.read_text())
+    assert [r["concurrency"] for r in data] == [1, 2]
+    lines = csv.read_text().strip().splitlines()
+    assert lines[0].startswith("Concurrency,")
+    assert len(lines) == 3
+    captured = capsys.readouterr()
+    assert "Ramp concurrency: 1" in captured.out
