@@ -22,3 +22,14 @@ Subpackages
 """
 
 __version__ = "0.1.0"
+
+from ._builder import InferRequestBuilder
+from ._endpoint import FixedEndpoint, MultiEndpointClient, RoundRobinEndpoint
+
+__all__ = [
+    "InferRequestBuilder",
+    "FixedEndpoint",
+    "MultiEndpointClient",
+    "RoundRobinEndpoint",
+    "__version__",
+]
