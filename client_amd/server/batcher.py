@@ -76,11 +76,8 @@ class DynamicBatcher:
             self._cv.notify()
 
     def _run(self):
-        import contextlib
-
         import torch
 
-        stream = None  # lazily created once we see a CUDA tensor
         while True:
             with self._cv:
                 while self._alive and not self._queue:
@@ -106,37 +103,34 @@ class DynamicBatcher:
                     break
                 time.sleep(min(remaining, 0.0002))
 
+            # NOTE: everything runs on the worker thread's DEFAULT
+            # stream. A redesign that put batches on a dedicated stream
+            # with event-gated output copies MEASURED SLOWER on MI355X
+            # (ResNet 8569 -> 6167 inf/s, DenseNet 4648 -> 1638): the
+            # single-stream pipeline is what the hardware runs best
+            # here, and the per-request completion event below is
+            # enough to avoid full-device syncs downstream.
             try:
                 cuda = items[0].tensors and items[0].tensors[0].is_cuda
-                if cuda and stream is None:
-                    stream = torch.cuda.Stream()
-                ctx = (torch.cuda.stream(stream) if cuda
-                       else contextlib.nullcontext())
-                with ctx:
-                    if cuda:
-                        # batch inputs were produced on the default
-                        # stream (or by the client process, host-ordered)
-                        stream.wait_stream(torch.cuda.default_stream())
-                    if len(items) == 1:
-                        outputs = self._model._execute_direct(
-                            items[0].tensors)
-                        items[0].outputs = outputs
-                    else:
-                        n_inputs = len(items[0].tensors)
-                        merged = [
-                            torch.cat([it.tensors[i] for it in items], dim=0)
-                            for i in range(n_inputs)
-                        ]
-                        outputs = self._model._execute_direct(merged)
-                        off = 0
-                        for it in items:
-                            it.outputs = [o[off : off + it.batch]
-                                          for o in outputs]
-                            off += it.batch
-                    done_ev = None
-                    if cuda:
-                        done_ev = torch.cuda.Event()
-                        done_ev.record(stream)
+                if len(items) == 1:
+                    outputs = self._model._execute_direct(items[0].tensors)
+                    items[0].outputs = outputs
+                else:
+                    n_inputs = len(items[0].tensors)
+                    merged = [
+                        torch.cat([it.tensors[i] for it in items], dim=0)
+                        for i in range(n_inputs)
+                    ]
+                    outputs = self._model._execute_direct(merged)
+                    off = 0
+                    for it in items:
+                        it.outputs = [o[off : off + it.batch]
+                                      for o in outputs]
+                        off += it.batch
+                done_ev = None
+                if cuda:
+                    done_ev = torch.cuda.Event()
+                    done_ev.record()
                 for it in items:
                     it.done_ev = done_ev
                     it.event.set()

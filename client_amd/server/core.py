@@ -214,22 +214,6 @@ class InferenceCore:
         self.ready = True
         self.config_overrides = {}
         self.file_overrides = {}
-        self._copy_streams = threading.local()
-
-    def _get_copy_stream(self, torch):
-        """Per-THREAD copy stream for output-region copies. A single
-        shared stream deadlocks throughput: worker threads enqueue
-        [wait_event(their batch), copy] in arbitrary order, so a
-        later batch's event-wait can land ahead of an earlier batch's
-        copies and head-of-line block them for a full batch time
-        (measured: DenseNet c8 halved, 24 ms avg_compute_output).
-        Requests within one thread are sequential, so per-thread
-        streams have no such inversion."""
-        s = getattr(self._copy_streams, "stream", None)
-        if s is None:
-            s = torch.cuda.Stream()
-            self._copy_streams.stream = s
-        return s
 
     # ---- repository ----
     def add_model(self, model, ready=True):
@@ -545,55 +529,44 @@ class InferenceCore:
             "UINT8": torch.uint8, "FP64": torch.float64, "BOOL": torch.bool,
         }
         by_name = {n: r for (n, _, _), r in zip(model.outputs, results)}
-        # output copies ride a dedicated copy stream ordered on the
-        # model's completion event, and the request host-waits only its
-        # OWN copy event — a full torch.cuda.synchronize() here queued
-        # every request's ~16 KB output copy behind the next batches'
-        # forwards on the shared default stream (5-6 ms of
-        # avg_compute_output at c8, DenseNet — ROUND2_NOTES item 7)
-        copy_stream = None
-        if done_ev is not None:
-            copy_stream = self._get_copy_stream(torch)
-        import contextlib
-
-        ctx = (torch.cuda.stream(copy_stream) if copy_stream is not None
-               else contextlib.nullcontext())
-        with ctx:
-            if copy_stream is not None:
-                copy_stream.wait_event(done_ev)
-            for io in req_outputs:
-                name = io["name"]
-                if name not in by_name:
-                    raise InferenceError(
-                        f"unexpected inference output '{name}' for model "
-                        f"'{model.name}'"
-                    )
-                result = by_name[name]
-                datatype = model_dtypes[name]
-                out_view = region_tensor(
-                    {"parameters": io["parameters"]}, datatype,
-                    list(result.shape)
+        # Everything stays on the default stream (a dedicated-copy-
+        # stream design measured SLOWER on MI355X — see
+        # DynamicBatcher._run). The request host-waits a per-request
+        # event recorded after ITS copies instead of a full device
+        # synchronize, so it never blocks on other models' side-stream
+        # work (e.g. the LLM decode loop).
+        for io in req_outputs:
+            name = io["name"]
+            if name not in by_name:
+                raise InferenceError(
+                    f"unexpected inference output '{name}' for model "
+                    f"'{model.name}'"
                 )
-                out_view.copy_(result.to(torch_dt[datatype]))
-                params = io["parameters"]
-                response["outputs"].append({
-                    "name": name,
-                    "datatype": datatype,
-                    "shape": list(result.shape),
-                    "parameters": {
-                        "shared_memory_region":
-                            params["shared_memory_region"],
-                        "shared_memory_byte_size":
-                            params["shared_memory_byte_size"],
-                        **({"shared_memory_offset":
-                            params["shared_memory_offset"]}
-                           if params.get("shared_memory_offset") else {}),
-                    },
-                })
-            if copy_stream is not None:
-                copied = torch.cuda.Event()
-                copied.record(copy_stream)
-        if copy_stream is not None:
+            result = by_name[name]
+            datatype = model_dtypes[name]
+            out_view = region_tensor(
+                {"parameters": io["parameters"]}, datatype,
+                list(result.shape)
+            )
+            out_view.copy_(result.to(torch_dt[datatype]))
+            params = io["parameters"]
+            response["outputs"].append({
+                "name": name,
+                "datatype": datatype,
+                "shape": list(result.shape),
+                "parameters": {
+                    "shared_memory_region":
+                        params["shared_memory_region"],
+                    "shared_memory_byte_size":
+                        params["shared_memory_byte_size"],
+                    **({"shared_memory_offset":
+                        params["shared_memory_offset"]}
+                       if params.get("shared_memory_offset") else {}),
+                },
+            })
+        if results and hasattr(results[0], "is_cuda") and results[0].is_cuda:
+            copied = torch.cuda.Event()
+            copied.record()
             copied.synchronize()
         else:
             torch.cuda.synchronize()

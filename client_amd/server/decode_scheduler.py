@@ -373,6 +373,11 @@ class DecodeScheduler:
         return True
 
     def _run(self):
+        import os
+        import time as _time
+
+        trace = os.environ.get("CLIENT_AMD_DECODE_TRACE") == "1"
+        acc = [0.0, 0.0, 0.0, 0]  # admit+wait, prefill, decode, iters
         while True:
             with self._cv:
                 while (self._alive and self._pending.empty()
@@ -380,6 +385,26 @@ class DecodeScheduler:
                     self._cv.wait()
                 if not self._alive:
                     return
+            if not trace:
+                self._admit()
+                self._prefill_step()
+                self._decode_step()
+                continue
+            t0 = _time.monotonic_ns()
             self._admit()
+            t1 = _time.monotonic_ns()
             self._prefill_step()
+            t2 = _time.monotonic_ns()
             self._decode_step()
+            t3 = _time.monotonic_ns()
+            acc[0] += t1 - t0
+            acc[1] += t2 - t1
+            acc[2] += t3 - t2
+            acc[3] += 1
+            if acc[3] % 200 == 0:
+                n = acc[3]
+                print(f"[decode-trace] iters={n} admit={acc[0]/n/1e6:.3f}ms "
+                      f"prefill={acc[1]/n/1e6:.3f}ms "
+                      f"decode={acc[2]/n/1e6:.3f}ms", flush=True)
+                acc[0] = acc[1] = acc[2] = 0.0
+                acc[3] = 0
