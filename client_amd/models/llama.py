@@ -9,11 +9,33 @@ inputs synthetic, which is exactly what the decode-path benchmark
 needs — the compute per token is the real thing.
 """
 
+import os
 from dataclasses import dataclass
 
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
+
+# Decode attention via grouped bmm instead of materializing K/V per
+# query head: repeat_interleave writes (and re-reads) rep copies of the
+# whole K/V window EVERY layer EVERY step — ~100 MB/layer of pure
+# HBM3E traffic at 1k context. The bmm path reads K/V once
+# ([b,kv,rep,d] @ [b,kv,d,L] broadcast matmul), fp32 softmax like
+# sdpa's math backend. A/B-gated: CLIENT_AMD_GQA_BMM=0 restores sdpa.
+_GQA_BMM = os.environ.get("CLIENT_AMD_GQA_BMM", "1") != "0"
+
+
+def _gqa_decode_attention(q, k_all, v_all, mask, rep):
+    """q [b, H, 1, d] (H = kv*rep, head h = g*rep + r); k_all/v_all
+    [b, kv, L, d]; mask additive [b, 1, 1, L]. Returns [b, H, 1, d]."""
+    b, H, _, d = q.shape
+    kv = H // rep
+    qg = q.view(b, kv, rep, d)
+    scores = torch.matmul(qg, k_all.transpose(-1, -2))  # [b,kv,rep,L]
+    scores = scores * (d ** -0.5) + mask.to(scores.dtype)  # [b,1,1,L] bc
+    attn = torch.softmax(scores.float(), dim=-1).to(q.dtype)
+    out = torch.matmul(attn, v_all)  # [b, kv, rep, d]
+    return out.view(b, H, 1, d)
 
 
 @dataclass
@@ -274,13 +296,19 @@ class LlamaModel(nn.Module):
             k_all = ck[:, :, :max_len]
             v_all = cv[:, :, :max_len]
             rep = self.cfg.n_heads // self.cfg.n_kv_heads
-            # materialize K/V heads: measured faster than enable_gqa on
-            # ROCm for decode shapes (see docs/PERFORMANCE.md)
-            k_all = k_all.repeat_interleave(rep, dim=1)
-            v_all = v_all.repeat_interleave(rep, dim=1)
-            attn = F.scaled_dot_product_attention(
-                q, k_all, v_all, attn_mask=mask.to(q.dtype)
-            )
+            if _GQA_BMM and q.is_cuda:
+                # grouped bmm reads K/V once (no per-query-head
+                # materialization; see _gqa_decode_attention)
+                attn = _gqa_decode_attention(q, k_all, v_all, mask, rep)
+            else:
+                # materialize K/V heads: measured faster than
+                # enable_gqa on ROCm (docs/PERFORMANCE.md)
+                attn = F.scaled_dot_product_attention(
+                    q,
+                    k_all.repeat_interleave(rep, dim=1),
+                    v_all.repeat_interleave(rep, dim=1),
+                    attn_mask=mask.to(q.dtype),
+                )
             attn = attn.transpose(1, 2).reshape(b, 1, -1)
             x = x + block.wo(attn)
             h = block.ffn_norm(x)

@@ -590,3 +590,39 @@ def test_image_preprocess_batched_matches_single(hipshm):
         hr.free(src)
         hr.free(dst_b)
         hr.free(dst_1)
+
+
+@pytest.mark.gpu
+def test_gqa_bmm_decode_matches_sdpa():
+    """The grouped-bmm decode attention (no K/V materialization) must
+    produce the same generations as the repeat_interleave+sdpa path."""
+    import torch
+
+    from client_amd.models import llama as L
+
+    torch.manual_seed(11)
+    cfg = L.llama_tiny_config()
+    m = L.LlamaModel(cfg).eval().to("cuda", torch.bfloat16)
+    tokens = torch.randint(0, cfg.vocab_size, (3, 1), device="cuda")
+    pos = torch.tensor([5, 9, 2], device="cuda")
+    kv = m.make_kv_cache(3, "cuda", torch.bfloat16)
+    for ck, cv in kv:
+        ck.normal_()
+        cv.normal_()
+    kv2 = [(ck.clone(), cv.clone()) for ck, cv in kv]
+
+    old = L._GQA_BMM
+    try:
+        with torch.inference_mode():
+            L._GQA_BMM = True
+            a = m.forward_decode_batch(tokens, pos, kv, max_len=16)
+            L._GQA_BMM = False
+            b = m.forward_decode_batch(tokens, pos, kv2, max_len=16)
+    finally:
+        L._GQA_BMM = old
+    torch.cuda.synchronize()
+    assert torch.allclose(a.float(), b.float(), atol=0.25, rtol=0.05), (
+        (a.float() - b.float()).abs().max().item()
+    )
+    # the decision that matters: identical argmax tokens
+    assert torch.equal(a.argmax(-1), b.argmax(-1))
