@@ -129,7 +129,17 @@ class DynamicBatcher:
                         off += it.batch
                 done_ev = None
                 if cuda:
-                    done_ev = torch.cuda.Event()
+                    # ring of reused events: creating one per batch
+                    # churns HSA interrupt signals on ROCm (each
+                    # hipEventCreate takes one); 64 deep is far beyond
+                    # any consumer's lag
+                    ring = getattr(self, "_ev_ring", None)
+                    if ring is None:
+                        ring = [torch.cuda.Event() for _ in range(64)]
+                        self._ev_ring = ring
+                        self._ev_idx = 0
+                    done_ev = ring[self._ev_idx]
+                    self._ev_idx = (self._ev_idx + 1) % len(ring)
                     done_ev.record()
                 for it in items:
                     it.done_ev = done_ev

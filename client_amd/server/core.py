@@ -214,6 +214,7 @@ class InferenceCore:
         self.ready = True
         self.config_overrides = {}
         self.file_overrides = {}
+        self._tls = threading.local()
 
     # ---- repository ----
     def add_model(self, model, ready=True):
@@ -564,13 +565,31 @@ class InferenceCore:
                        if params.get("shared_memory_offset") else {}),
                 },
             })
-        if results and hasattr(results[0], "is_cuda") and results[0].is_cuda:
-            copied = torch.cuda.Event()
-            copied.record()
-            copied.synchronize()
-        else:
-            torch.cuda.synchronize()
+        self._sync_outputs(torch)
         return response, [], t1 - t0, t2 - t1
+
+    _SYNC_MODE = None
+
+    def _sync_outputs(self, torch):
+        """Wait for this request's output copies. Uses a REUSED
+        per-thread event (rapid hipEventCreate churn is a ROCm
+        pathology — each event takes an HSA interrupt signal);
+        CLIENT_AMD_SYNC_MODE=device falls back to a full device
+        synchronize for A/B."""
+        if InferenceCore._SYNC_MODE is None:
+            import os
+
+            InferenceCore._SYNC_MODE = os.environ.get(
+                "CLIENT_AMD_SYNC_MODE", "event")
+        if InferenceCore._SYNC_MODE == "device":
+            torch.cuda.synchronize()
+            return
+        ev = getattr(self._tls, "sync_event", None)
+        if ev is None:
+            ev = torch.cuda.Event()
+            self._tls.sync_event = ev
+        ev.record()
+        ev.synchronize()
 
     def _build_response(self, model, request, result, parameters):
         requested = request.get("outputs")

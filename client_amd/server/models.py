@@ -512,7 +512,18 @@ class TorchModel(Model):
         outs = self._execute_direct(device_tensors)
         ev = None
         if outs and outs[0].is_cuda:
-            ev = self._torch.cuda.Event()
+            # reused per-thread event (hipEventCreate churn is costly
+            # on ROCm — see DynamicBatcher ring)
+            tls = getattr(self, "_ev_tls", None)
+            if tls is None:
+                import threading
+
+                tls = threading.local()
+                self._ev_tls = tls
+            ev = getattr(tls, "event", None)
+            if ev is None:
+                ev = self._torch.cuda.Event()
+                tls.event = ev
             ev.record()
         return outs, ev
 
