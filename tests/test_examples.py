@@ -23,6 +23,8 @@ HTTP_EXAMPLES = [
     "simple_http_sequence_sync_infer_client.py",
     "simple_http_aio_infer_client.py",
     "reuse_infer_objects_client.py",
+    "builder_infer_client.py",
+    "rotating_endpoint_client.py",
 ]
 
 GRPC_EXAMPLES = [
