@@ -571,16 +571,18 @@ class InferenceCore:
     _SYNC_MODE = None
 
     def _sync_outputs(self, torch):
-        """Wait for this request's output copies. Uses a REUSED
-        per-thread event (rapid hipEventCreate churn is a ROCm
-        pathology — each event takes an HSA interrupt signal);
-        CLIENT_AMD_SYNC_MODE=device falls back to a full device
-        synchronize for A/B."""
+        """Wait for this request's output copies. Full device
+        synchronize by DEFAULT: host event-synchronize measured
+        catastrophically slower on MI355X under thread concurrency
+        (DenseNet c8: 1928 inf/s with per-request hipEventSynchronize —
+        even with reused events — vs 4419 with hipDeviceSynchronize;
+        gpurun_out/r02v6). CLIENT_AMD_SYNC_MODE=event keeps the event
+        path selectable for re-measurement on future ROCm builds."""
         if InferenceCore._SYNC_MODE is None:
             import os
 
             InferenceCore._SYNC_MODE = os.environ.get(
-                "CLIENT_AMD_SYNC_MODE", "event")
+                "CLIENT_AMD_SYNC_MODE", "device")
         if InferenceCore._SYNC_MODE == "device":
             torch.cuda.synchronize()
             return

@@ -332,32 +332,44 @@ class DecodeScheduler:
         self._graphs[bucket] = entry
         return entry
 
+    def _host_staging(self):
+        st = getattr(self, "_staging", None)
+        if st is None:
+            pin = str(self.device).startswith("cuda")
+            th = torch.zeros(self.max_batch, 1, dtype=torch.int64,
+                             pin_memory=pin)
+            ph = torch.zeros(self.max_batch, dtype=torch.int64,
+                             pin_memory=pin)
+            st = (th, ph, th.view(-1).numpy(), ph.numpy())
+            self._staging = st
+        return st
+
     def _decode_step(self):
         active = [i for i, s in enumerate(self.slots) if s.active]
         if not active:
             return False
         # run the FULL preallocated batch: decode cost is weight-read
-        # bound, so inactive rows are free; masks keep rows independent
-        tokens = torch.tensor(
-            [s.last_token for s in self.slots], dtype=torch.int64,
-            device=self.device,
-        )[:, None]
-        pos_rows = torch.tensor(
-            [max(s.pos, 1) if s.active else self._scratch
-             for s in self.slots],
-            dtype=torch.int64, device=self.device,
-        )
+        # bound, so inactive rows are free; masks keep rows independent.
+        # Inputs stage through REUSED pinned host tensors + one async
+        # H2D each (torch.tensor(..., device=cuda) per step was two
+        # blocking H2D round-trips plus allocations).
+        th, ph, tnp, pnp = self._host_staging()
+        for i, s in enumerate(self.slots):
+            tnp[i] = s.last_token
+            pnp[i] = max(s.pos, 1) if s.active else self._scratch
         if self.use_graph:
-            self._tokens_dev.copy_(tokens)
-            self._pos_dev.copy_(pos_rows)
+            self._tokens_dev.copy_(th, non_blocking=True)
+            self._pos_dev.copy_(ph, non_blocking=True)
             max_pos = max(s.pos for s in self.slots if s.active)
             graph, next_out = self._get_graph(self._bucket(max_pos + 1))
             graph.replay()
             next_tokens = next_out.tolist()
         else:
+            self._tokens_dev.copy_(th)
+            self._pos_dev.copy_(ph)
             with torch.inference_mode():
                 logits = self.model.forward_decode_batch(
-                    tokens, pos_rows, self.kv_cache
+                    self._tokens_dev, self._pos_dev, self.kv_cache
                 )
                 next_tokens = logits.argmax(-1).tolist()
         for i in active:
