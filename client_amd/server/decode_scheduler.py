@@ -358,12 +358,43 @@ class DecodeScheduler:
             tnp[i] = s.last_token
             pnp[i] = max(s.pos, 1) if s.active else self._scratch
         if self.use_graph:
+            trace2 = getattr(self, "_trace2", None)
+            if trace2 is None:
+                import os as _os
+
+                trace2 = _os.environ.get(
+                    "CLIENT_AMD_DECODE_TRACE") == "2"
+                self._trace2 = trace2
+                if trace2:
+                    self._t2_acc = [0.0, 0.0, 0.0, 0]
+            if trace2:
+                import time as _t
+
+                a = _t.monotonic_ns()
             self._tokens_dev.copy_(th, non_blocking=True)
             self._pos_dev.copy_(ph, non_blocking=True)
             max_pos = max(s.pos for s in self.slots if s.active)
             graph, next_out = self._get_graph(self._bucket(max_pos + 1))
+            if trace2:
+                b = _t.monotonic_ns()
             graph.replay()
+            if trace2:
+                c = _t.monotonic_ns()
             next_tokens = next_out.tolist()
+            if trace2:
+                d = _t.monotonic_ns()
+                acc = self._t2_acc
+                acc[0] += b - a   # staging copies + graph lookup
+                acc[1] += c - b   # replay submit
+                acc[2] += d - c   # tolist = GPU wait + D2H
+                acc[3] += 1
+                if acc[3] % 200 == 0:
+                    n = acc[3]
+                    print(f"[decode-trace2] stage={acc[0]/n/1e6:.3f}ms "
+                          f"submit={acc[1]/n/1e6:.3f}ms "
+                          f"wait={acc[2]/n/1e6:.3f}ms", flush=True)
+                    acc[0] = acc[1] = acc[2] = 0.0
+                    acc[3] = 0
         else:
             self._tokens_dev.copy_(th)
             self._pos_dev.copy_(ph)
