@@ -18,6 +18,7 @@
 #include "client_amd/common.h"
 #include "client_amd/grpc_client.h"
 #include "client_amd/http_client.h"
+#include "client_amd/infer_builder.h"
 #include "client_amd/json.h"
 
 using namespace client_amd;
@@ -344,6 +345,35 @@ static void RunSuite(Client* client, const char* proto) {
     CHECK(errored);
     delete short0;
     delete short1;
+    delete result;
+  }
+
+  // -- Fluent builder drives both clients (the Rust
+  //    InferRequestBuilder surface, infer.rs:548)
+  {
+    std::vector<int32_t> d0(16), d1(16);
+    for (int i = 0; i < 16; ++i) {
+      d0[i] = i;
+      d1[i] = 3 * i;
+    }
+    InferRequestBuilder b("simple");
+    b.RequestId("builder-1")
+        .AddInput<int32_t>("INPUT0", {1, 16}, d0)
+        .AddInput<int32_t>("INPUT1", {1, 16}, d1)
+        .AddOutput("OUTPUT0")
+        .AddOutput("OUTPUT1");
+    InferResult* result = nullptr;
+    CHECK_OK(client->Infer(&result, b.Options(), b.Inputs(), b.Outputs()));
+    CHECK(result != nullptr && result->RequestStatus().IsOk());
+    std::string id;
+    CHECK(result->Id(&id).IsOk() && id == "builder-1");
+    const uint8_t* buf;
+    size_t n;
+    CHECK(result->RawData("OUTPUT0", &buf, &n).IsOk() && n == 64);
+    const int32_t* v = reinterpret_cast<const int32_t*>(buf);
+    bool ok = true;
+    for (int i = 0; i < 16; ++i) ok = ok && v[i] == d0[i] + d1[i];
+    CHECK(ok);
     delete result;
   }
 
