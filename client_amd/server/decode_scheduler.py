@@ -119,6 +119,12 @@ class DecodeScheduler:
             while g <= max(1, self.prefill_rows_per_step):
                 self._get_prefill_graph(g, bucket)
                 g *= 2
+            if i <= 2:
+                # idle-burst shape: with nothing decoding the cap lifts
+                # to max_batch, so the first bursts replay (max_batch,
+                # low-bucket) graphs
+                self._get_prefill_graph(self._group_size(self.max_batch),
+                                        bucket)
             if bucket >= self.model.cfg.max_seq:
                 break
 
@@ -226,10 +232,15 @@ class DecodeScheduler:
         if not pf:
             return
         c = self.prefill_chunk
-        # advance at most prefill_rows_per_step rows THIS iteration;
-        # the rest progress on subsequent loop iterations (each
-        # interleaved with a decode step)
-        todo = pf[:max(1, self.prefill_rows_per_step)]
+        # advance at most prefill_rows_per_step rows THIS iteration so
+        # in-flight decodes never stall more than ~one decode step; the
+        # rest progress on later iterations. When NOTHING is decoding
+        # (e.g. a start burst) there is no ITL to protect — run every
+        # prefilling row at once for the best TTFT.
+        any_active = any(s.active for s in self.slots)
+        cap = (max(1, self.prefill_rows_per_step) if any_active
+               else self.max_batch)
+        todo = pf[:cap]
         done = 0
         while done < len(todo):
             group_slots = todo[done:done + self.max_batch]
