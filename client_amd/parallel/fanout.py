@@ -153,10 +153,13 @@ class OverlappedBroadcaster:
             dist.broadcast(t, src=self.src)
 
     def wait_ready(self):
-        """Host-gate on the last broadcast's completion event (only);
-        records its hipEvent-measured duration."""
+        """Host-gate on the last broadcast's completion (only the side
+        stream — not the device); records the hipEvent-measured bcast
+        duration. Uses hipStreamSynchronize rather than
+        hipEventSynchronize: host event-waits measured pathologically
+        slow on this ROCm stack (profiles/sync_mode_ab_r02.md)."""
         if self._inflight:
-            self._ev_end.synchronize()
+            self._side.synchronize()
             self.bcast_ms.append(self._ev_start.elapsed_time(self._ev_end))
             self._inflight = False
 
