@@ -399,7 +399,7 @@ class DecodeScheduler:
                 acc[1] += c - b   # replay submit
                 acc[2] += d - c   # tolist = GPU wait + D2H
                 acc[3] += 1
-                if acc[3] % 200 == 0:
+                if acc[3] % 50 == 0:
                     n = acc[3]
                     print(f"[decode-trace2] stage={acc[0]/n/1e6:.3f}ms "
                           f"submit={acc[1]/n/1e6:.3f}ms "
