@@ -93,8 +93,7 @@ class BiasAct(nn.Module):
     def forward(self, x):
         import torch
 
-        if (x.is_cuda and x.dtype == torch.bfloat16
-                and x.is_contiguous()):
+        if x.is_cuda and x.dtype == torch.bfloat16:
             from ..ops import hip_runtime as hr
 
             if self.bias.dtype != torch.float32:
@@ -103,13 +102,25 @@ class BiasAct(nn.Module):
                 # eager warmup hits this before any hipGraph capture)
                 self.bias.data = self.bias.data.float()
 
-            n, c = x.shape[0], x.shape[1]
-            plane = x.numel() // (n * c)
-            hr.bias_act_bf16(
-                x.data_ptr(), self.bias.data_ptr(), x.data_ptr(),
-                n * c, plane, c, self.relu_flag,
-                torch.cuda.current_stream().cuda_stream)
-            return x
+            c = x.shape[1]
+            if x.is_contiguous():
+                n = x.shape[0]
+                plane = x.numel() // (n * c)
+                hr.bias_act_bf16(
+                    x.data_ptr(), self.bias.data_ptr(), x.data_ptr(),
+                    n * c, plane, c, self.relu_flag,
+                    torch.cuda.current_stream().cuda_stream)
+                return x
+            if (x.dim() == 4 and c % 8 == 0
+                    and x.is_contiguous(
+                        memory_format=torch.channels_last)):
+                # NHWC memory: channel is the fastest dim — the cl
+                # kernel keeps the fusion on MIOpen's native layout
+                hr.bias_res_act_cl_bf16(
+                    x.data_ptr(), 0, self.bias.data_ptr(), x.data_ptr(),
+                    x.numel(), c, self.relu_flag,
+                    torch.cuda.current_stream().cuda_stream)
+                return x
         out = x + self.bias.view(1, -1, *([1] * (x.dim() - 2))).to(x.dtype)
         return out.relu_() if self.relu_flag else out
 
@@ -129,21 +140,31 @@ class BiasResAct(nn.Module):
     def forward(self, x, residual):
         import torch
 
-        if (x.is_cuda and x.dtype == torch.bfloat16 and x.is_contiguous()
-                and residual.dtype == torch.bfloat16
-                and residual.is_contiguous()):
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and residual.dtype == torch.bfloat16):
             from ..ops import hip_runtime as hr
 
             if self.bias.dtype != torch.float32:
                 # TorchModel's .to(bf16) sweeps buffers; kernel reads fp32
                 self.bias.data = self.bias.data.float()
-            n, c = x.shape[0], x.shape[1]
-            plane = x.numel() // (n * c)
-            hr.bias_res_act_bf16(
-                x.data_ptr(), residual.data_ptr(), self.bias.data_ptr(),
-                x.data_ptr(), n * c, plane, c, True,
-                torch.cuda.current_stream().cuda_stream)
-            return x
+            c = x.shape[1]
+            if x.is_contiguous() and residual.is_contiguous():
+                n = x.shape[0]
+                plane = x.numel() // (n * c)
+                hr.bias_res_act_bf16(
+                    x.data_ptr(), residual.data_ptr(),
+                    self.bias.data_ptr(), x.data_ptr(), n * c, plane, c,
+                    True, torch.cuda.current_stream().cuda_stream)
+                return x
+            cl = torch.channels_last
+            if (x.dim() == 4 and c % 8 == 0
+                    and x.is_contiguous(memory_format=cl)
+                    and residual.is_contiguous(memory_format=cl)):
+                hr.bias_res_act_cl_bf16(
+                    x.data_ptr(), residual.data_ptr(),
+                    self.bias.data_ptr(), x.data_ptr(), x.numel(), c,
+                    True, torch.cuda.current_stream().cuda_stream)
+                return x
         out = x + residual + self.bias.view(
             1, -1, *([1] * (x.dim() - 2))
         ).to(x.dtype)

@@ -331,6 +331,16 @@ static void bias_act_bf16(uintptr_t x, uintptr_t bias, uintptr_t out,
                              reinterpret_cast<hipStream_t>(stream_handle)));
 }
 
+static void bias_res_act_cl_bf16(uintptr_t x, uintptr_t res, uintptr_t bias,
+                                 uintptr_t out, long n, int channels,
+                                 bool relu, uintptr_t stream_handle) {
+  HIP_CHECK(ca_bias_res_act_cl_bf16(
+      reinterpret_cast<const void*>(x), reinterpret_cast<const void*>(res),
+      reinterpret_cast<const void*>(bias), reinterpret_cast<void*>(out), n,
+      channels, relu ? 1 : 0,
+      reinterpret_cast<hipStream_t>(stream_handle)));
+}
+
 static void bias_res_act_bf16(uintptr_t x, uintptr_t res, uintptr_t bias,
                               uintptr_t out, long n_planes, long plane,
                               int channels, bool relu,
@@ -448,6 +458,9 @@ PYBIND11_MODULE(_hip_c, m) {
         py::arg("channels"), py::arg("relu"), py::arg("stream_handle"));
   m.def("bias_res_act_bf16", &bias_res_act_bf16, py::arg("x"), py::arg("res"),
         py::arg("bias"), py::arg("out"), py::arg("n_planes"), py::arg("plane"),
+        py::arg("channels"), py::arg("relu"), py::arg("stream_handle"));
+  m.def("bias_res_act_cl_bf16", &bias_res_act_cl_bf16, py::arg("x"),
+        py::arg("res"), py::arg("bias"), py::arg("out"), py::arg("n"),
         py::arg("channels"), py::arg("relu"), py::arg("stream_handle"));
   m.def("rmsnorm_bf16", &rmsnorm_bf16, py::arg("x"), py::arg("w"),
         py::arg("out"), py::arg("rows"), py::arg("dim"), py::arg("eps"),

@@ -828,6 +828,77 @@ extern "C" __global__ void bias_res_act_bf16_kernel(
   }
 }
 
+// Channels-LAST (NHWC-contiguous) variant: channel is the fastest
+// dim, so each lane's 8 consecutive elements are 8 consecutive
+// channels — loads of x, residual AND bias all coalesce. This is what
+// lets the fused epilogues ride MIOpen's native NHWC conv path
+// (channels_last measured catastrophic in r01 precisely because the
+// NCHW-plane kernels fell back to eager torch ops on NHWC tensors).
+// Requires channels % 8 == 0 (ResNet: 64..2048, all satisfy).
+extern "C" __global__ void bias_res_act_cl_bf16_kernel(
+    const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
+    const float* __restrict__ bias, uint16_t* __restrict__ out, long n,
+    int channels, int do_relu) {
+  auto fin = [&](float f) -> uint16_t {
+    if (do_relu && f < 0.f) f = 0.f;
+    uint32_t u = __float_as_uint(f);
+    u += 0x7FFF + ((u >> 16) & 1);  // RNE to bf16
+    return (uint16_t)(u >> 16);
+  };
+  long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  long stride = (long)gridDim.x * blockDim.x * 8;
+  for (long i = i0; i + 8 <= n; i += stride) {
+    const int c0 = (int)(i % channels);
+    uint4 v = *reinterpret_cast<const uint4*>(x + i);
+    uint4 r = res ? *reinterpret_cast<const uint4*>(res + i)
+                  : uint4{0, 0, 0, 0};
+    const float4 b0 = *reinterpret_cast<const float4*>(bias + c0);
+    const float4 b1 = *reinterpret_cast<const float4*>(bias + c0 + 4);
+    const float bv[8] = {b0.x, b0.y, b0.z, b0.w, b1.x, b1.y, b1.z, b1.w};
+    uint32_t* w = (uint32_t*)&v;
+    const uint32_t* rw = (const uint32_t*)&r;
+    uint16_t o[8];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float lo = __uint_as_float((w[j] & 0xFFFFu) << 16);
+      float hi = __uint_as_float(w[j] & 0xFFFF0000u);
+      if (res) {
+        lo += __uint_as_float((rw[j] & 0xFFFFu) << 16);
+        hi += __uint_as_float(rw[j] & 0xFFFF0000u);
+      }
+      o[2 * j] = fin(lo + bv[2 * j]);
+      o[2 * j + 1] = fin(hi + bv[2 * j + 1]);
+    }
+    *reinterpret_cast<uint4*>(out + i) = *reinterpret_cast<uint4*>(o);
+  }
+  // tail (n % 8 == 0 whenever channels % 8 == 0, but keep it safe)
+  long tail = (n / 8) * 8;
+  long ti = tail + (blockIdx.x * blockDim.x + threadIdx.x);
+  if (ti < n && (blockIdx.x * blockDim.x + threadIdx.x) < 8) {
+    float f = __uint_as_float(((uint32_t)x[ti]) << 16);
+    if (res) f += __uint_as_float(((uint32_t)res[ti]) << 16);
+    out[ti] = fin(f + bias[ti % channels]);
+  }
+}
+
+extern "C" hipError_t ca_bias_res_act_cl_bf16(const void* x, const void* res,
+                                              const void* bias, void* out,
+                                              long n, int channels,
+                                              int do_relu,
+                                              hipStream_t stream) {
+  if (channels % 8 != 0) return hipErrorInvalidValue;
+  if (((uintptr_t)x & 15) || ((uintptr_t)out & 15) ||
+      ((uintptr_t)res & 15)) {
+    return hipErrorInvalidValue;
+  }
+  hipLaunchKernelGGL(bias_res_act_cl_bf16_kernel,
+                     dim3(ca_grid_for((n + 7) / 8)), dim3(256), 0, stream,
+                     (const uint16_t*)x, (const uint16_t*)res,
+                     (const float*)bias, (uint16_t*)out, n, channels,
+                     do_relu);
+  return hipGetLastError();
+}
+
 extern "C" hipError_t ca_bias_res_act_bf16(const void* x, const void* res,
                                            const void* bias, void* out,
                                            long n_planes, long plane,

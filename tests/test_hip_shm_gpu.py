@@ -626,3 +626,37 @@ def test_gqa_bmm_decode_matches_sdpa():
     )
     # the decision that matters: identical argmax tokens
     assert torch.equal(a.argmax(-1), b.argmax(-1))
+
+
+@pytest.mark.gpu
+def test_bias_res_act_channels_last_numerics():
+    """Channels-last fused bias(+residual)+ReLU must be bit-exact with
+    the NCHW kernel result order ((x+res)+bias, RNE) on NHWC tensors."""
+    import torch
+
+    from client_amd.ops import hip_runtime as hr
+
+    torch.manual_seed(13)
+    for (c, h, w) in [(64, 56, 56), (2048, 7, 7), (8, 5, 3)]:
+        cl = torch.channels_last
+        x = torch.randn(4, c, h, w, device="cuda", dtype=torch.bfloat16
+                        ).contiguous(memory_format=cl)
+        r = torch.randn(4, c, h, w, device="cuda", dtype=torch.bfloat16
+                        ).contiguous(memory_format=cl)
+        bias = torch.randn(c, device="cuda", dtype=torch.float32)
+        ref = (x.float() + r.float() + bias.view(1, -1, 1, 1)).relu()
+        ref = ref.to(torch.bfloat16)
+        y = x.clone()  # preserves channels_last
+        hr.bias_res_act_cl_bf16(
+            y.data_ptr(), r.data_ptr(), bias.data_ptr(), y.data_ptr(),
+            y.numel(), c, True, torch.cuda.current_stream().cuda_stream)
+        torch.cuda.synchronize()
+        assert torch.equal(y, ref), (c, h, w)
+        # bias-only + relu path (res = null)
+        y2 = x.clone()
+        ref2 = (x.float() + bias.view(1, -1, 1, 1)).relu().to(torch.bfloat16)
+        hr.bias_res_act_cl_bf16(
+            y2.data_ptr(), 0, bias.data_ptr(), y2.data_ptr(),
+            y2.numel(), c, True, torch.cuda.current_stream().cuda_stream)
+        torch.cuda.synchronize()
+        assert torch.equal(y2, ref2), (c, h, w)
