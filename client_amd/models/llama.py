@@ -24,6 +24,30 @@ import torch.nn.functional as F
 # sdpa's math backend. A/B-gated: CLIENT_AMD_GQA_BMM=0 restores sdpa.
 _GQA_BMM = os.environ.get("CLIENT_AMD_GQA_BMM", "1") != "0"
 
+# Hand-written skinny decode GEMM (weights-stationary streaming; see
+# kernels.hip decode_gemm_bf16_kernel): the batch-8 decode GEMMs run at
+# 2.1-4.2 TB/s through hipBLASLt's tiles (profiles/decode_rocprof_r02.txt)
+# against an ~8 TB/s weight-read ceiling. A/B-gated until measured.
+_DECODE_GEMM = os.environ.get("CLIENT_AMD_DECODE_GEMM", "0") == "1"
+
+
+def _decode_linear(lin, h):
+    """h [8, 1, K] -> [8, 1, N] via the skinny decode GEMM kernel when
+    eligible, else the torch linear."""
+    if (_DECODE_GEMM and h.is_cuda and h.dtype == torch.bfloat16
+            and h.shape[0] == 8 and h.shape[1] == 1
+            and lin.weight.shape[1] % 512 == 0 and lin.bias is None):
+        from ..ops import hip_runtime as hr
+
+        x2 = h.reshape(8, -1).contiguous()
+        n, k = lin.weight.shape
+        y = torch.empty(8, n, device=h.device, dtype=torch.bfloat16)
+        hr.decode_gemm_bf16(x2.data_ptr(), lin.weight.data_ptr(),
+                            y.data_ptr(), n, k,
+                            torch.cuda.current_stream().cuda_stream)
+        return y.view(8, 1, n)
+    return lin(h)
+
 
 def _gqa_decode_attention(q, k_all, v_all, mask, rep):
     """q [b, H, 1, d] (H = kv*rep, head h = g*rep + r); k_all/v_all
@@ -261,12 +285,12 @@ class LlamaModel(nn.Module):
         ar = torch.arange(b, device=tokens.device)
         for block, (ck, cv) in zip(self.blocks, kv_cache):
             h = block.attn_norm(x)
-            q = block.wq(h).view(b, 1, self.cfg.n_heads, block.head_dim
-                                 ).transpose(1, 2)
-            k = block.wk(h).view(b, 1, self.cfg.n_kv_heads, block.head_dim
-                                 ).transpose(1, 2)
-            v = block.wv(h).view(b, 1, self.cfg.n_kv_heads, block.head_dim
-                                 ).transpose(1, 2)
+            q = _decode_linear(block.wq, h).view(
+                b, 1, self.cfg.n_heads, block.head_dim).transpose(1, 2)
+            k = _decode_linear(block.wk, h).view(
+                b, 1, self.cfg.n_kv_heads, block.head_dim).transpose(1, 2)
+            v = _decode_linear(block.wv, h).view(
+                b, 1, self.cfg.n_kv_heads, block.head_dim).transpose(1, 2)
 
             if q.is_cuda and q.dtype == torch.bfloat16:
                 # fused decode RoPE (q+k, per-row positions, one launch)
@@ -310,11 +334,15 @@ class LlamaModel(nn.Module):
                     attn_mask=mask.to(q.dtype),
                 )
             attn = attn.transpose(1, 2).reshape(b, 1, -1)
-            x = x + block.wo(attn)
+            x = x + _decode_linear(block.wo, attn)
             h = block.ffn_norm(x)
-            x = x + block.w2(F.silu(block.w1(h)) * block.w3(h))
+            x = x + _decode_linear(
+                block.w2,
+                F.silu(_decode_linear(block.w1, h))
+                * _decode_linear(block.w3, h),
+            )
         x = self.norm(x)
-        return self.lm_head(x)[:, 0]
+        return _decode_linear(self.lm_head, x)[:, 0]
 
     def forward_prefill_chunk(self, tokens, pos_rows, row_lens, last_idx,
                               kv_cache, bucket, row_map=None):

@@ -331,6 +331,14 @@ static void bias_act_bf16(uintptr_t x, uintptr_t bias, uintptr_t out,
                              reinterpret_cast<hipStream_t>(stream_handle)));
 }
 
+static void decode_gemm_bf16(uintptr_t x, uintptr_t w, uintptr_t y, int n,
+                             int k, uintptr_t stream_handle) {
+  HIP_CHECK(ca_decode_gemm_bf16(
+      reinterpret_cast<const void*>(x), reinterpret_cast<const void*>(w),
+      reinterpret_cast<void*>(y), n, k,
+      reinterpret_cast<hipStream_t>(stream_handle)));
+}
+
 static void bias_res_act_cl_bf16(uintptr_t x, uintptr_t res, uintptr_t bias,
                                  uintptr_t out, long n, int channels,
                                  bool relu, uintptr_t stream_handle) {
@@ -459,6 +467,8 @@ PYBIND11_MODULE(_hip_c, m) {
   m.def("bias_res_act_bf16", &bias_res_act_bf16, py::arg("x"), py::arg("res"),
         py::arg("bias"), py::arg("out"), py::arg("n_planes"), py::arg("plane"),
         py::arg("channels"), py::arg("relu"), py::arg("stream_handle"));
+  m.def("decode_gemm_bf16", &decode_gemm_bf16, py::arg("x"), py::arg("w"),
+        py::arg("y"), py::arg("n"), py::arg("k"), py::arg("stream_handle"));
   m.def("bias_res_act_cl_bf16", &bias_res_act_cl_bf16, py::arg("x"),
         py::arg("res"), py::arg("bias"), py::arg("out"), py::arg("n"),
         py::arg("channels"), py::arg("relu"), py::arg("stream_handle"));

@@ -660,3 +660,63 @@ def test_bias_res_act_channels_last_numerics():
             y2.numel(), c, True, torch.cuda.current_stream().cuda_stream)
         torch.cuda.synchronize()
         assert torch.equal(y2, ref2), (c, h, w)
+
+
+@pytest.mark.gpu
+def test_decode_gemm_numerics():
+    """Skinny decode GEMM vs torch linear (fp32-accum reference) on the
+    real decode shapes; bf16-ulp agreement."""
+    import torch
+
+    from client_amd.ops import hip_runtime as hr
+
+    torch.manual_seed(17)
+    for (n, k) in [(4096, 4096), (1024, 4096), (14336, 4096),
+                   (4096, 14336), (128256, 4096), (512, 512)]:
+        x = torch.randn(8, k, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16) * 0.02
+        y = torch.empty(8, n, device="cuda", dtype=torch.bfloat16)
+        hr.decode_gemm_bf16(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                            n, k, torch.cuda.current_stream().cuda_stream)
+        torch.cuda.synchronize()
+        ref = (x.float() @ w.float().t())
+        got = y.float()
+        # both accumulate in fp32; reduction ORDER differs, so allow a
+        # few bf16 ulps relative to the magnitude
+        tol = ref.abs().max().item() * 2 ** -7 + 1e-2
+        assert (got - ref).abs().max().item() < tol, (
+            n, k, (got - ref).abs().max().item()
+        )
+
+
+@pytest.mark.gpu
+def test_decode_gemm_in_model_matches_torch():
+    """forward_decode_batch with the kernel GEMMs produces the same
+    argmax tokens as the torch-linear path."""
+    import torch
+
+    from client_amd.models import llama as L
+
+    torch.manual_seed(19)
+    cfg = L.LlamaConfig(vocab_size=2048, dim=512, n_layers=2, n_heads=8,
+                        n_kv_heads=4, ffn_dim=1024, max_seq=128)
+    m = L.LlamaModel(cfg).eval().to("cuda", torch.bfloat16)
+    tokens = torch.randint(0, cfg.vocab_size, (8, 1), device="cuda")
+    pos = torch.randint(1, 64, (8,), device="cuda")
+    kv = m.make_kv_cache(8, "cuda", torch.bfloat16)
+    for ck, cv in kv:
+        ck.normal_(std=0.1)
+        cv.normal_(std=0.1)
+    kv2 = [(ck.clone(), cv.clone()) for ck, cv in kv]
+    old = L._DECODE_GEMM
+    try:
+        with torch.inference_mode():
+            L._DECODE_GEMM = True
+            a = m.forward_decode_batch(tokens, pos, kv, max_len=64)
+            L._DECODE_GEMM = False
+            b = m.forward_decode_batch(tokens, pos, kv2, max_len=64)
+    finally:
+        L._DECODE_GEMM = old
+    torch.cuda.synchronize()
+    assert torch.allclose(a.float(), b.float(), atol=0.3, rtol=0.05)
+    assert (a.argmax(-1) == b.argmax(-1)).float().mean() > 0.9
