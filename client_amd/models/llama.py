@@ -51,12 +51,13 @@ def _decode_linear(lin, h):
 
 def _gqa_decode_attention(q, k_all, v_all, mask, rep):
     """q [b, H, 1, d] (H = kv*rep, head h = g*rep + r); k_all/v_all
-    [b, kv, L, d]; mask additive [b, 1, 1, L]. Returns [b, H, 1, d]."""
+    [b, kv, L, d]; mask additive [b, 1, 1, L], PRE-CAST to q.dtype
+    (the per-layer cast was 32 extra launches per decode step)."""
     b, H, _, d = q.shape
     kv = H // rep
     qg = q.view(b, kv, rep, d)
     scores = torch.matmul(qg, k_all.transpose(-1, -2))  # [b,kv,rep,L]
-    scores = scores * (d ** -0.5) + mask.to(scores.dtype)  # [b,1,1,L] bc
+    scores = scores * (d ** -0.5) + mask  # [b,1,1,L] broadcast
     attn = torch.softmax(scores.float(), dim=-1).to(q.dtype)
     out = torch.matmul(attn, v_all)  # [b, kv, rep, d]
     return out.view(b, H, 1, d)
@@ -282,6 +283,7 @@ class LlamaModel(nn.Module):
         c_rows = cos[pos_rows][:, None, None, :]  # [b,1,1,hd/2]
         s_rows = sin[pos_rows][:, None, None, :]
         x = self.tok(tokens)  # [b,1,dim]
+        mask_x = mask.to(x.dtype)  # cast ONCE, not per layer
         ar = torch.arange(b, device=tokens.device)
         for block, (ck, cv) in zip(self.blocks, kv_cache):
             h = block.attn_norm(x)
@@ -293,15 +295,21 @@ class LlamaModel(nn.Module):
                 b, 1, self.cfg.n_kv_heads, block.head_dim).transpose(1, 2)
 
             if q.is_cuda and q.dtype == torch.bfloat16:
-                # fused decode RoPE (q+k, per-row positions, one launch)
+                # fused decode RoPE + KV-cache scatter: q rotated in
+                # place, rotated k and copied v land directly in the
+                # cache at each row's position — one launch instead of
+                # rope + two index_put scatters per layer
                 from ..ops import hip_runtime as hr
 
                 q = q.contiguous()
                 k = k.contiguous()
-                hr.rope_decode_bf16(
-                    q.data_ptr(), k.data_ptr(), cos.data_ptr(),
+                v = v.contiguous()
+                hr.rope_scatter_decode_bf16(
+                    q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                    ck.data_ptr(), cv.data_ptr(), cos.data_ptr(),
                     sin.data_ptr(), pos_rows.data_ptr(), b,
-                    self.cfg.n_heads, self.cfg.n_kv_heads, block.head_dim,
+                    self.cfg.n_heads, self.cfg.n_kv_heads,
+                    block.head_dim, ck.shape[2],
                     torch.cuda.current_stream().cuda_stream,
                 )
             else:
@@ -314,16 +322,16 @@ class LlamaModel(nn.Module):
 
                 q = rope_rows(q)
                 k = rope_rows(k)
-            # scatter this step's k/v at each row's own position
-            ck[ar, :, pos_rows] = k[:, :, 0]
-            cv[ar, :, pos_rows] = v[:, :, 0]
+                # scatter this step's k/v at each row's own position
+                ck[ar, :, pos_rows] = k[:, :, 0]
+                cv[ar, :, pos_rows] = v[:, :, 0]
             k_all = ck[:, :, :max_len]
             v_all = cv[:, :, :max_len]
             rep = self.cfg.n_heads // self.cfg.n_kv_heads
             if _GQA_BMM and q.is_cuda:
                 # grouped bmm reads K/V once (no per-query-head
                 # materialization; see _gqa_decode_attention)
-                attn = _gqa_decode_attention(q, k_all, v_all, mask, rep)
+                attn = _gqa_decode_attention(q, k_all, v_all, mask_x, rep)
             else:
                 # materialize K/V heads: measured faster than
                 # enable_gqa on ROCm (docs/PERFORMANCE.md)
@@ -331,7 +339,7 @@ class LlamaModel(nn.Module):
                     q,
                     k_all.repeat_interleave(rep, dim=1),
                     v_all.repeat_interleave(rep, dim=1),
-                    attn_mask=mask.to(q.dtype),
+                    attn_mask=mask_x,
                 )
             attn = attn.transpose(1, 2).reshape(b, 1, -1)
             x = x + _decode_linear(block.wo, attn)

@@ -320,6 +320,63 @@ extern "C" __global__ void rope_decode_bf16_kernel(
   }
 }
 
+// RoPE + KV-cache scatter in ONE launch: rotates q (in place) and k
+// (rotated values written straight into the cache at each row's
+// position), and copies v into the cache — removing the two
+// torch index_put launches per layer per decode step (64/step at 32
+// layers; measured 28.9 ms / 5120 calls in the decode profile).
+// ck/cv: [b, hk, cache_len, d]; k/v: [b, hk, d] contiguous.
+extern "C" __global__ void rope_scatter_decode_bf16_kernel(
+    uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
+    const uint16_t* __restrict__ v, uint16_t* __restrict__ ck,
+    uint16_t* __restrict__ cv, const float* __restrict__ cos_tab,
+    const float* __restrict__ sin_tab, const long* __restrict__ pos, int b,
+    int hq, int hk, int d, long cache_len) {
+  const int half = d / 2;
+  const long total = (long)b * (hq + 2 * hk) * half;
+  long i = (long)(blockIdx.x * blockDim.x + threadIdx.x);
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    long rem = i;
+    const int j = (int)(rem % half);
+    rem /= half;
+    const int head = (int)(rem % (hq + 2 * hk));
+    const int row = (int)(rem / (hq + 2 * hk));
+    const long p = pos[row];
+    if (head < hq) {  // q: rotate in place
+      uint16_t* base = q + ((long)row * hq + head) * d;
+      const float c = cos_tab[p * half + j];
+      const float s = sin_tab[p * half + j];
+      const float x1 = __uint_as_float((uint32_t)base[2 * j] << 16);
+      const float x2 = __uint_as_float((uint32_t)base[2 * j + 1] << 16);
+      base[2 * j] =
+          __bfloat16_as_ushort(__float2bfloat16(x1 * c - x2 * s));
+      base[2 * j + 1] =
+          __bfloat16_as_ushort(__float2bfloat16(x1 * s + x2 * c));
+    } else if (head < hq + hk) {  // k: rotate -> cache
+      const int h = head - hq;
+      const uint16_t* base = k + ((long)row * hk + h) * d;
+      uint16_t* dst =
+          ck + (((long)row * hk + h) * cache_len + p) * d;
+      const float c = cos_tab[p * half + j];
+      const float s = sin_tab[p * half + j];
+      const float x1 = __uint_as_float((uint32_t)base[2 * j] << 16);
+      const float x2 = __uint_as_float((uint32_t)base[2 * j + 1] << 16);
+      dst[2 * j] =
+          __bfloat16_as_ushort(__float2bfloat16(x1 * c - x2 * s));
+      dst[2 * j + 1] =
+          __bfloat16_as_ushort(__float2bfloat16(x1 * s + x2 * c));
+    } else {  // v: straight copy -> cache
+      const int h = head - hq - hk;
+      const uint16_t* base = v + ((long)row * hk + h) * d;
+      uint16_t* dst =
+          cv + (((long)row * hk + h) * cache_len + p) * d;
+      dst[2 * j] = base[2 * j];
+      dst[2 * j + 1] = base[2 * j + 1];
+    }
+  }
+}
+
 // Scalar fallbacks for pointers not 16-byte aligned (region offsets are
 // caller-controlled; hipMalloc bases are 256-B aligned so the vector
 // path is the common case).
@@ -1039,5 +1096,19 @@ extern "C" hipError_t ca_rope_decode_bf16(void* q, void* k,
                      dim3(256), 0, stream, (uint16_t*)q, (uint16_t*)k,
                      (const float*)cos_tab, (const float*)sin_tab,
                      (const long*)pos, b, hq, hk, d);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t ca_rope_scatter_decode_bf16(
+    void* q, const void* k, const void* v, void* ck, void* cv,
+    const void* cos_tab, const void* sin_tab, const void* pos, int b,
+    int hq, int hk, int d, long cache_len, hipStream_t stream) {
+  long total = (long)b * (hq + 2 * hk) * (d / 2);
+  hipLaunchKernelGGL(rope_scatter_decode_bf16_kernel,
+                     dim3(ca_grid_for(total)), dim3(256), 0, stream,
+                     (uint16_t*)q, (const uint16_t*)k, (const uint16_t*)v,
+                     (uint16_t*)ck, (uint16_t*)cv, (const float*)cos_tab,
+                     (const float*)sin_tab, (const long*)pos, b, hq, hk, d,
+                     cache_len);
   return hipGetLastError();
 }

@@ -720,3 +720,53 @@ def test_decode_gemm_in_model_matches_torch():
     torch.cuda.synchronize()
     assert torch.allclose(a.float(), b.float(), atol=0.3, rtol=0.05)
     assert (a.argmax(-1) == b.argmax(-1)).float().mean() > 0.9
+
+
+@pytest.mark.gpu
+def test_rope_scatter_decode_kernel():
+    """Fused RoPE+KV-scatter: q rotated in place, rotated k / copied v
+    land at each row's cache position, rest of the cache untouched."""
+    import torch
+
+    from client_amd.models.llama import precompute_rope
+    from client_amd.ops import hip_runtime as hr
+
+    torch.manual_seed(23)
+    b, hq, hk, d, clen = 8, 8, 4, 64, 33
+    cos, sin = precompute_rope(d, clen, 10000.0, "cuda")
+    q = torch.randn(b, hq, 1, d, device="cuda", dtype=torch.bfloat16
+                    ).contiguous()
+    k = torch.randn(b, hk, 1, d, device="cuda", dtype=torch.bfloat16
+                    ).contiguous()
+    v = torch.randn(b, hk, 1, d, device="cuda", dtype=torch.bfloat16
+                    ).contiguous()
+    ck = torch.zeros(b, hk, clen, d, device="cuda", dtype=torch.bfloat16)
+    cv = torch.zeros_like(ck)
+    pos = torch.randint(0, clen, (b,), device="cuda", dtype=torch.int64)
+
+    def rope_ref(t):
+        c = cos[pos][:, None, None, :].float()
+        s = sin[pos][:, None, None, :].float()
+        tf = t.float()
+        t1, t2 = tf[..., 0::2], tf[..., 1::2]
+        out = torch.empty_like(tf)
+        out[..., 0::2] = t1 * c - t2 * s
+        out[..., 1::2] = t1 * s + t2 * c
+        return out.to(torch.bfloat16)
+
+    q_ref = rope_ref(q)
+    k_ref = rope_ref(k)
+    hr.rope_scatter_decode_bf16(
+        q.data_ptr(), k.data_ptr(), v.data_ptr(), ck.data_ptr(),
+        cv.data_ptr(), cos.data_ptr(), sin.data_ptr(), pos.data_ptr(),
+        b, hq, hk, d, clen, torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    assert torch.equal(q, q_ref)
+    ar = torch.arange(b, device="cuda")
+    assert torch.equal(ck[ar, :, pos], k_ref[:, :, 0])
+    assert torch.equal(cv[ar, :, pos], v[:, :, 0])
+    # everything else stays zero
+    mask = torch.ones(b, hk, clen, d, device="cuda", dtype=torch.bool)
+    mask[ar, :, pos] = False
+    assert ck[mask].abs().sum() == 0
+    assert cv[mask].abs().sum() == 0
