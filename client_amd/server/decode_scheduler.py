@@ -184,6 +184,46 @@ class DecodeScheduler:
             self._pf_bufs[group] = bufs
         return bufs
 
+    _EAGER = "eager"  # sentinel: capture failed, run this shape eager
+
+    def _capture_graph(self, forward):
+        """Capture forward().argmax(-1) into a hipGraph; returns
+        (graph, out_tensor) or None if capture cannot complete.
+
+        A capture can be INVALIDATED by concurrent work from other
+        serving threads (co-served models replaying/copying while this
+        model lazily captures a new shape —
+        hipErrorStreamCaptureInvalidated, seen in the r02 mixed soak,
+        where the raw exception killed the decode worker). One retry
+        after a full sync, then the caller falls back to eager for
+        that shape."""
+        from .models import GRAPH_CAPTURE_LOCK
+
+        for _ in range(2):
+            try:
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    with torch.inference_mode():
+                        for _ in range(2):  # warmup (idempotent writes)
+                            warm = forward().argmax(-1)
+                torch.cuda.current_stream().wait_stream(side)
+                torch.cuda.synchronize()
+                graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
+                with GRAPH_CAPTURE_LOCK:
+                    with torch.inference_mode():
+                        with torch.cuda.graph(
+                                graph,
+                                capture_error_mode="thread_local"):
+                            out = forward().argmax(-1)
+                return (graph, out)
+            except Exception:
+                try:
+                    torch.cuda.synchronize()
+                except Exception:
+                    pass
+        return None
+
     def _get_prefill_graph(self, group, bucket):
         entry = self._pf_graphs.get((group, bucket))
         if entry is not None:
@@ -191,25 +231,9 @@ class DecodeScheduler:
         bufs = self._pf_buffers(group)
         args = (bufs["tokens"], bufs["pos"], bufs["lens"], bufs["last"],
                 self.kv_cache, bucket, bufs["rows"])
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            with torch.inference_mode():
-                for _ in range(2):  # warmup (idempotent cache writes)
-                    warm = self.model.forward_prefill_chunk(
-                        *args).argmax(-1)
-        torch.cuda.current_stream().wait_stream(side)
-        torch.cuda.synchronize()
-        from .models import GRAPH_CAPTURE_LOCK
-
-        graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
-        with GRAPH_CAPTURE_LOCK:
-            with torch.inference_mode():
-                with torch.cuda.graph(graph,
-                                      capture_error_mode="thread_local"):
-                    first_out = self.model.forward_prefill_chunk(
-                        *args).argmax(-1)
-        entry = (graph, first_out)
+        entry = self._capture_graph(
+            lambda: self.model.forward_prefill_chunk(*args)
+        ) or self._EAGER
         self._pf_graphs[(group, bucket)] = entry
         return entry
 
@@ -276,8 +300,10 @@ class DecodeScheduler:
             bufs["lens"].copy_(lens)
             bufs["last"].copy_(last)
             bufs["rows"].copy_(rows)
-            if self.use_graph:
-                graph, first_out = self._get_prefill_graph(group, bucket)
+            entry = (self._get_prefill_graph(group, bucket)
+                     if self.use_graph else self._EAGER)
+            if entry is not self._EAGER:
+                graph, first_out = entry
                 graph.replay()
                 firsts = first_out.tolist()
             else:
@@ -315,31 +341,12 @@ class DecodeScheduler:
         entry = self._graphs.get(bucket)
         if entry is not None:
             return entry
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            with torch.inference_mode():
-                for _ in range(2):  # warmup (idempotent cache writes)
-                    logits = self.model.forward_decode_batch(
-                        self._tokens_dev, self._pos_dev, self.kv_cache,
-                        max_len=bucket,
-                    )
-                    warm = logits.argmax(-1)
-        torch.cuda.current_stream().wait_stream(side)
-        torch.cuda.synchronize()
-        from .models import GRAPH_CAPTURE_LOCK
-
-        graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
-        with GRAPH_CAPTURE_LOCK:
-            with torch.inference_mode():
-                with torch.cuda.graph(graph,
-                                      capture_error_mode="thread_local"):
-                    logits = self.model.forward_decode_batch(
-                        self._tokens_dev, self._pos_dev, self.kv_cache,
-                        max_len=bucket,
-                    )
-                    next_out = logits.argmax(-1)
-        entry = (graph, next_out)
+        entry = self._capture_graph(
+            lambda: self.model.forward_decode_batch(
+                self._tokens_dev, self._pos_dev, self.kv_cache,
+                max_len=bucket,
+            )
+        ) or self._EAGER
         self._graphs[bucket] = entry
         return entry
 
@@ -385,13 +392,24 @@ class DecodeScheduler:
             self._tokens_dev.copy_(th, non_blocking=True)
             self._pos_dev.copy_(ph, non_blocking=True)
             max_pos = max(s.pos for s in self.slots if s.active)
-            graph, next_out = self._get_graph(self._bucket(max_pos + 1))
+            entry = self._get_graph(self._bucket(max_pos + 1))
             if trace2:
                 b = _t.monotonic_ns()
-            graph.replay()
-            if trace2:
-                c = _t.monotonic_ns()
-            next_tokens = next_out.tolist()
+            if entry is not self._EAGER:
+                graph, next_out = entry
+                graph.replay()
+                if trace2:
+                    c = _t.monotonic_ns()
+                next_tokens = next_out.tolist()
+            else:
+                with torch.inference_mode():
+                    logits = self.model.forward_decode_batch(
+                        self._tokens_dev, self._pos_dev, self.kv_cache,
+                        max_len=self._bucket(max_pos + 1),
+                    )
+                if trace2:
+                    c = _t.monotonic_ns()
+                next_tokens = logits.argmax(-1).tolist()
             if trace2:
                 d = _t.monotonic_ns()
                 acc = self._t2_acc
@@ -426,6 +444,27 @@ class DecodeScheduler:
                 slot.out_queue.put(self.END)
         return True
 
+    def _fail_all_streams(self, exc):
+        """Recovery path for an unexpected error inside a scheduler
+        step: terminate every in-flight stream (consumers see END) and
+        reset the slots so new requests keep being served."""
+        import sys
+        import traceback
+
+        print(f"[decode-scheduler] step failed, resetting slots: {exc}",
+              file=sys.stderr, flush=True)
+        traceback.print_exc()
+        for slot in self.slots:
+            if slot.state != _Slot.FREE and slot.out_queue is not None:
+                slot.out_queue.put(self.END)
+            slot.state = _Slot.FREE
+            slot.prefill_ids = None
+            slot.out_queue = None
+        try:
+            torch.cuda.synchronize()
+        except Exception:
+            pass
+
     def _run(self):
         import os
         import time as _time
@@ -440,9 +479,16 @@ class DecodeScheduler:
                 if not self._alive:
                     return
             if not trace:
-                self._admit()
-                self._prefill_step()
-                self._decode_step()
+                try:
+                    self._admit()
+                    self._prefill_step()
+                    self._decode_step()
+                except Exception as e:
+                    # the worker must OUTLIVE any transient GPU error
+                    # (a capture invalidation once killed this thread
+                    # and hung every stream — r02 mixed soak): fail the
+                    # in-flight streams, reset the slots, keep serving
+                    self._fail_all_streams(e)
                 continue
             t0 = _time.monotonic_ns()
             self._admit()

@@ -551,8 +551,27 @@ class TorchModel(Model):
         with self._graph_lock:
             entry = self._graphs.get(key)
             if entry is None:
-                entry = self._capture(device_tensors)
+                # a capture can be invalidated by concurrent work from
+                # co-served models (hipErrorStreamCaptureInvalidated,
+                # r02 mixed soak); retry once, then serve this
+                # signature eager instead of erroring the request
+                for _ in range(2):
+                    try:
+                        entry = self._capture(device_tensors)
+                        break
+                    except Exception:
+                        entry = "eager"
+                        try:
+                            self._torch.cuda.synchronize()
+                        except Exception:
+                            pass
                 self._graphs[key] = entry
+            if entry == "eager":
+                with self._torch.inference_mode():
+                    result = self.module(*device_tensors)
+                    if not isinstance(result, (tuple, list)):
+                        result = (result,)
+                    return list(result)
             static_ins, graph, static_outs = entry
             for si, t in zip(static_ins, device_tensors):
                 si.copy_(t)
