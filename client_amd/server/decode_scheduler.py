@@ -201,16 +201,18 @@ class DecodeScheduler:
 
         for _ in range(2):
             try:
-                side = torch.cuda.Stream()
-                side.wait_stream(torch.cuda.current_stream())
-                with torch.cuda.stream(side):
-                    with torch.inference_mode():
-                        for _ in range(2):  # warmup (idempotent writes)
-                            warm = forward().argmax(-1)
-                torch.cuda.current_stream().wait_stream(side)
-                torch.cuda.synchronize()
-                graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
+                # EXCLUSIVE vs every model execution for warmup AND
+                # capture (see models._RWLock rationale)
                 with GRAPH_CAPTURE_LOCK:
+                    side = torch.cuda.Stream()
+                    side.wait_stream(torch.cuda.current_stream())
+                    with torch.cuda.stream(side):
+                        with torch.inference_mode():
+                            for _ in range(2):  # warmup (idempotent)
+                                warm = forward().argmax(-1)
+                    torch.cuda.current_stream().wait_stream(side)
+                    torch.cuda.synchronize()
+                    graph = torch.cuda.CUDAGraph()  # hipGraph on ROCm
                     with torch.inference_mode():
                         with torch.cuda.graph(
                                 graph,
@@ -295,25 +297,30 @@ class DecodeScheduler:
                 max_end = max(max_end, end)
             bucket = self._bucket(max_end)
             bufs = self._pf_buffers(group)
-            bufs["tokens"].copy_(tokens)
-            bufs["pos"].copy_(pos)
-            bufs["lens"].copy_(lens)
-            bufs["last"].copy_(last)
-            bufs["rows"].copy_(rows)
+            # resolve (or capture) the graph BEFORE taking the shared
+            # execution guard — capture takes the exclusive side
             entry = (self._get_prefill_graph(group, bucket)
                      if self.use_graph else self._EAGER)
-            if entry is not self._EAGER:
-                graph, first_out = entry
-                graph.replay()
-                firsts = first_out.tolist()
-            else:
-                with torch.inference_mode():
-                    logits = self.model.forward_prefill_chunk(
-                        bufs["tokens"], bufs["pos"], bufs["lens"],
-                        bufs["last"], self.kv_cache, bucket,
-                        bufs["rows"],
-                    )
-                    firsts = logits.argmax(-1).tolist()
+            from .models import GRAPH_EXEC_SHARED
+
+            with GRAPH_EXEC_SHARED:
+                bufs["tokens"].copy_(tokens)
+                bufs["pos"].copy_(pos)
+                bufs["lens"].copy_(lens)
+                bufs["last"].copy_(last)
+                bufs["rows"].copy_(rows)
+                if entry is not self._EAGER:
+                    graph, first_out = entry
+                    graph.replay()
+                    firsts = first_out.tolist()
+                else:
+                    with torch.inference_mode():
+                        logits = self.model.forward_prefill_chunk(
+                            bufs["tokens"], bufs["pos"], bufs["lens"],
+                            bufs["last"], self.kv_cache, bucket,
+                            bufs["rows"],
+                        )
+                        firsts = logits.argmax(-1).tolist()
             for g, i in enumerate(group_slots):
                 slot = self.slots[i]
                 end, total = ends[i]
@@ -389,27 +396,33 @@ class DecodeScheduler:
                 import time as _t
 
                 a = _t.monotonic_ns()
-            self._tokens_dev.copy_(th, non_blocking=True)
-            self._pos_dev.copy_(ph, non_blocking=True)
             max_pos = max(s.pos for s in self.slots if s.active)
+            # graph resolution (possible capture = exclusive) happens
+            # BEFORE the shared execution guard
             entry = self._get_graph(self._bucket(max_pos + 1))
+            from .models import GRAPH_EXEC_SHARED
+
             if trace2:
                 b = _t.monotonic_ns()
-            if entry is not self._EAGER:
-                graph, next_out = entry
-                graph.replay()
-                if trace2:
-                    c = _t.monotonic_ns()
-                next_tokens = next_out.tolist()
-            else:
-                with torch.inference_mode():
-                    logits = self.model.forward_decode_batch(
-                        self._tokens_dev, self._pos_dev, self.kv_cache,
-                        max_len=self._bucket(max_pos + 1),
-                    )
-                if trace2:
-                    c = _t.monotonic_ns()
-                next_tokens = logits.argmax(-1).tolist()
+            with GRAPH_EXEC_SHARED:
+                self._tokens_dev.copy_(th, non_blocking=True)
+                self._pos_dev.copy_(ph, non_blocking=True)
+                if entry is not self._EAGER:
+                    graph, next_out = entry
+                    graph.replay()
+                    if trace2:
+                        c = _t.monotonic_ns()
+                    next_tokens = next_out.tolist()
+                else:
+                    with torch.inference_mode():
+                        logits = self.model.forward_decode_batch(
+                            self._tokens_dev, self._pos_dev,
+                            self.kv_cache,
+                            max_len=self._bucket(max_pos + 1),
+                        )
+                    if trace2:
+                        c = _t.monotonic_ns()
+                    next_tokens = logits.argmax(-1).tolist()
             if trace2:
                 d = _t.monotonic_ns()
                 acc = self._t2_acc

@@ -26,7 +26,76 @@ from ..utils import triton_to_np_dtype
 # on the same GPU) are not poisoned by capture state. Found by the
 # mixed-load soak: global-mode capture made every other model fail with
 # hipErrorStreamCaptureUnsupported.
-GRAPH_CAPTURE_LOCK = threading.Lock()
+class _RWLock:
+    """Writer-priority readers/writer lock. Model EXECUTIONS hold it
+    shared; hipGraph CAPTURES hold it exclusive. Rationale (r02 mixed
+    co-serving soak): capturing a new shape while another model's
+    MIOpen kernels launch concurrently invalidates the capture
+    (thread_local capture mode does not protect MIOpen's launch path
+    on this ROCm build), and torch's capture-failure cleanup then
+    aborts the whole process from a noexcept context. Exclusivity at
+    capture time prevents the poisoning outright; writer priority
+    keeps a capture from starving under request traffic."""
+
+    def __init__(self):
+        self._cond = threading.Condition()
+        self._readers = 0
+        self._writer = False
+        self._pending_writers = 0
+
+    def acquire_read(self):
+        with self._cond:
+            while self._writer or self._pending_writers:
+                self._cond.wait()
+            self._readers += 1
+
+    def release_read(self):
+        with self._cond:
+            self._readers -= 1
+            if self._readers == 0:
+                self._cond.notify_all()
+
+    def acquire_write(self):
+        with self._cond:
+            self._pending_writers += 1
+            while self._writer or self._readers:
+                self._cond.wait()
+            self._pending_writers -= 1
+            self._writer = True
+
+    def release_write(self):
+        with self._cond:
+            self._writer = False
+            self._cond.notify_all()
+
+
+class _WriteGuard:
+    def __init__(self, lock):
+        self._lock = lock
+
+    def __enter__(self):
+        self._lock.acquire_write()
+
+    def __exit__(self, *exc):
+        self._lock.release_write()
+
+
+class _ReadGuard:
+    def __init__(self, lock):
+        self._lock = lock
+
+    def __enter__(self):
+        self._lock.acquire_read()
+
+    def __exit__(self, *exc):
+        self._lock.release_read()
+
+
+_GRAPH_RW = _RWLock()
+# capture sites: exclusive (name kept from r01 so call sites read the same)
+GRAPH_CAPTURE_LOCK = _WriteGuard(_GRAPH_RW)
+# model-execution sites: shared
+GRAPH_EXEC_SHARED = _ReadGuard(_GRAPH_RW)
 
 
 class Model:
@@ -474,17 +543,19 @@ class TorchModel(Model):
 
     def _capture(self, device_tensors):
         torch = self._torch
-        static_ins = [t.clone() for t in device_tensors]
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            with torch.inference_mode():
-                for _ in range(3):
-                    warm = self.module(*static_ins)
-        torch.cuda.current_stream().wait_stream(side)
-        torch.cuda.synchronize()
-        graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
+        # EXCLUSIVE for warmup + capture: no other model may launch
+        # kernels while a capture is open (see _RWLock rationale)
         with GRAPH_CAPTURE_LOCK:
+            static_ins = [t.clone() for t in device_tensors]
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                with torch.inference_mode():
+                    for _ in range(3):
+                        warm = self.module(*static_ins)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()  # hipGraph under ROCm
             with torch.inference_mode():
                 with torch.cuda.graph(graph,
                                       capture_error_mode="thread_local"):
@@ -536,11 +607,12 @@ class TorchModel(Model):
         launch submission — GPU work stays pipelined on the stream."""
         torch = self._torch
         if not self.use_graph:
-            with torch.inference_mode():
-                result = self.module(*device_tensors)
-                if not isinstance(result, (tuple, list)):
-                    result = (result,)
-                return list(result)
+            with GRAPH_EXEC_SHARED:
+                with torch.inference_mode():
+                    result = self.module(*device_tensors)
+                    if not isinstance(result, (tuple, list)):
+                        result = (result,)
+                    return list(result)
         if self._channels_last:
             device_tensors = [
                 t.to(memory_format=self._torch.channels_last)
@@ -567,13 +639,15 @@ class TorchModel(Model):
                             pass
                 self._graphs[key] = entry
             if entry == "eager":
-                with self._torch.inference_mode():
-                    result = self.module(*device_tensors)
-                    if not isinstance(result, (tuple, list)):
-                        result = (result,)
-                    return list(result)
+                with GRAPH_EXEC_SHARED:
+                    with self._torch.inference_mode():
+                        result = self.module(*device_tensors)
+                        if not isinstance(result, (tuple, list)):
+                            result = (result,)
+                        return list(result)
             static_ins, graph, static_outs = entry
-            for si, t in zip(static_ins, device_tensors):
-                si.copy_(t)
-            graph.replay()
-            return [o.clone() for o in static_outs]
+            with GRAPH_EXEC_SHARED:
+                for si, t in zip(static_ins, device_tensors):
+                    si.copy_(t)
+                graph.replay()
+                return [o.clone() for o in static_outs]
