@@ -770,3 +770,65 @@ def test_rope_scatter_decode_kernel():
     mask[ar, :, pos] = False
     assert ck[mask].abs().sum() == 0
     assert cv[mask].abs().sum() == 0
+
+
+@pytest.mark.gpu
+def test_overlapped_broadcaster_pipeline_nccl_world1(hipshm):
+    """The full bench.py world>1 pipeline machinery on CUDA under an
+    nccl world=1 group: ping-pong HIP-shm staging regions, pack kernel
+    on the HIP-runtime stream, ExternalStream event ordering, the
+    side-stream collective with work.wait(), hipEvent bcast timing and
+    the stream-synchronize host gate. (The 8-GPU run is driver-side;
+    this pins every API the multi-rank path exercises.)"""
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no torch GPU")
+    import os
+
+    import torch.distributed as dist
+
+    from client_amd.parallel import OverlappedBroadcaster
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29613")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        n = 1 << 14
+        regions = [
+            hipshm.create_shared_memory_region(f"ovl{b}", n * 2, 0)
+            for b in range(2)
+        ]
+        try:
+            bc = OverlappedBroadcaster(regions, (n,), "BF16", src=0)
+            assert bc._cuda
+
+            data = [np.random.rand(n).astype(np.float32)
+                    for _ in range(7)]
+
+            def pack_for(step):
+                def pack_fn(buf_idx, v=data[step]):
+                    hipshm.set_shared_memory_region_cast(
+                        regions[buf_idx], v, "BF16", sync=False)
+                return pack_fn
+
+            # prologue + 6 pipelined steps, exactly bench.py's loop
+            bc.stage_and_broadcast(0, pack_for(0))
+            for step in range(6):
+                bc.wait_ready()
+                # serve step: buffer step%2 must hold data[step]
+                got = hipshm.get_contents_cast(regions[step % 2],
+                                               "BF16", [n])
+                np.testing.assert_allclose(got, data[step], rtol=2e-2,
+                                           atol=2e-2)
+                nxt = step + 1
+                if nxt < 7:
+                    bc.stage_and_broadcast(nxt % 2, pack_for(nxt))
+            bc.wait_ready()
+            # hipEvent timings were recorded for every broadcast
+            assert len(bc.bcast_ms) == 7
+            assert all(t >= 0.0 for t in bc.bcast_ms)
+        finally:
+            for r in regions:
+                hipshm.destroy_shared_memory_region(r)
+    finally:
+        dist.destroy_process_group()
