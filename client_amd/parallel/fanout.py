@@ -128,7 +128,13 @@ class OverlappedBroadcaster:
 
     @property
     def _active(self):
-        return dist.is_initialized() and dist.get_world_size() > 1
+        # any initialized group, INCLUDING world=1: a single-rank
+        # broadcast is a valid (cheap) collective, and running it keeps
+        # the whole event/side-stream machinery exercised by the
+        # world=1 GPU test instead of silently skipped (bench.py only
+        # constructs a broadcaster when world > 1, so the N=1 headline
+        # path never pays for this)
+        return dist.is_initialized()
 
     def stage_and_broadcast(self, buf_idx, pack_fn=None):
         """Enqueue pack (caller's fn, async on the pack stream) and the
