@@ -366,8 +366,22 @@ class PerfAnalyzer:
                 return [self.batch_size] + list(self.shapes[name]) \
                     if len(self.shapes[name]) == len(shape) - 1 \
                     else list(self.shapes[name])
-            return [d if d > 0 else (self.batch_size if i == 0 else 16)
-                    for i, d in enumerate(shape)]
+            out = [d if d > 0 else (self.batch_size if i == 0 else 16)
+                   for i, d in enumerate(shape)]
+            if any(d <= 0 for d in shape[1:]):
+                # a silently-defaulted dim makes results incomparable
+                # across runs (a BERT "seq128" sweep without
+                # --shape input_ids:128 actually measured seq16 once) —
+                # ALWAYS say what was measured
+                import sys
+
+                print(
+                    f"perf: input '{name}' has wildcard dims {list(shape)}"
+                    f" -> measuring {out}; pass --shape {name}:... to"
+                    " control it",
+                    file=sys.stderr,
+                )
+            return out
 
         inputs = [(n, d, concrete(n, s)) for n, d, s in inputs]
         outputs = [(n, d, concrete(n, s)) for n, d, s in outputs]
