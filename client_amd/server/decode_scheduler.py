@@ -209,7 +209,7 @@ class DecodeScheduler:
                     with torch.cuda.stream(side):
                         with torch.inference_mode():
                             for _ in range(2):  # warmup (idempotent)
-                                warm = forward().argmax(-1)
+                                warm = forward()
                     torch.cuda.current_stream().wait_stream(side)
                     torch.cuda.synchronize()
                     graph = torch.cuda.CUDAGraph()  # hipGraph on ROCm
@@ -217,7 +217,7 @@ class DecodeScheduler:
                         with torch.cuda.graph(
                                 graph,
                                 capture_error_mode="thread_local"):
-                            out = forward().argmax(-1)
+                            out = forward()
                 return (graph, out)
             except Exception:
                 try:
@@ -234,7 +234,7 @@ class DecodeScheduler:
         args = (bufs["tokens"], bufs["pos"], bufs["lens"], bufs["last"],
                 self.kv_cache, bucket, bufs["rows"])
         entry = self._capture_graph(
-            lambda: self.model.forward_prefill_chunk(*args)
+            lambda: self.model.forward_prefill_chunk(*args).argmax(-1)
         ) or self._EAGER
         self._pf_graphs[(group, bucket)] = entry
         return entry
@@ -344,16 +344,63 @@ class DecodeScheduler:
              ) * self.len_bucket
         return min(b, self.model.cfg.max_seq)
 
+    @property
+    def _wide_graph(self):
+        """CLIENT_AMD_WIDE_GRAPH=1: the decode graph also contains the
+        H2D staging copies (reading the pinned host tensors, whose
+        CONTENTS are re-read at every replay) and the D2H copy of the
+        argmax tokens into a pinned output — per step the host only
+        fills the pinned staging, replays, stream-syncs and reads the
+        pinned result (no per-step copy launches, no tolist transfer).
+        A/B-gated until measured."""
+        w = getattr(self, "_wide", None)
+        if w is None:
+            import os
+
+            w = (os.environ.get("CLIENT_AMD_WIDE_GRAPH") == "1"
+                 and self.use_graph)
+            self._wide = w
+        return w
+
+    def _wide_out(self):
+        out = getattr(self, "_out_pinned", None)
+        if out is None:
+            out = torch.zeros(self.max_batch, dtype=torch.int64,
+                              pin_memory=True)
+            self._out_pinned = out
+            self._out_np = out.numpy()
+        return out
+
     def _get_graph(self, bucket):
         entry = self._graphs.get(bucket)
         if entry is not None:
             return entry
-        entry = self._capture_graph(
-            lambda: self.model.forward_decode_batch(
-                self._tokens_dev, self._pos_dev, self.kv_cache,
-                max_len=bucket,
-            )
-        ) or self._EAGER
+        if self._wide_graph:
+            th, ph, _, _ = self._host_staging()
+            out_pinned = self._wide_out()
+
+            def fwd():
+                self._tokens_dev.copy_(th, non_blocking=True)
+                self._pos_dev.copy_(ph, non_blocking=True)
+                out = self.model.forward_decode_batch(
+                    self._tokens_dev, self._pos_dev, self.kv_cache,
+                    max_len=bucket,
+                ).argmax(-1)
+                out_pinned.copy_(out, non_blocking=True)
+                return out
+        else:
+            def fwd():
+                return self.model.forward_decode_batch(
+                    self._tokens_dev, self._pos_dev, self.kv_cache,
+                    max_len=bucket,
+                ).argmax(-1)
+        entry = self._capture_graph(fwd) or self._EAGER
+        if entry is self._EAGER and self._wide_graph:
+            import sys
+
+            print("[decode-scheduler] wide-graph capture failed for "
+                  f"bucket {bucket}; running eager", file=sys.stderr,
+                  flush=True)
         self._graphs[bucket] = entry
         return entry
 
@@ -405,9 +452,20 @@ class DecodeScheduler:
             if trace2:
                 b = _t.monotonic_ns()
             with GRAPH_EXEC_SHARED:
-                self._tokens_dev.copy_(th, non_blocking=True)
-                self._pos_dev.copy_(ph, non_blocking=True)
-                if entry is not self._EAGER:
+                wide = self._wide_graph and entry is not self._EAGER
+                if not wide:
+                    self._tokens_dev.copy_(th, non_blocking=True)
+                    self._pos_dev.copy_(ph, non_blocking=True)
+                if wide:
+                    # everything (H2D, forward, argmax, D2H) is inside
+                    # the graph; the pinned staging is already filled
+                    graph, _ = entry
+                    graph.replay()
+                    if trace2:
+                        c = _t.monotonic_ns()
+                    torch.cuda.current_stream().synchronize()
+                    next_tokens = self._out_np.tolist()
+                elif entry is not self._EAGER:
                     graph, next_out = entry
                     graph.replay()
                     if trace2:
