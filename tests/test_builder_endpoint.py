@@ -102,3 +102,60 @@ def test_multi_endpoint_exhausted_raises():
             client.is_server_live()
     finally:
         client.close()
+
+
+def test_multi_endpoint_grpc(grpc_fixture_server):
+    host, port, _ = grpc_fixture_server
+    client = MultiEndpointClient([f"{host}:{port}"], protocol="grpc")
+    try:
+        assert client.is_server_live() is True
+        x0 = np.arange(16, dtype=np.int32).reshape(1, 16)
+        x1 = np.full((1, 16), 2, dtype=np.int32)
+        result = (InferRequestBuilder("simple")
+                  .input_from_numpy("INPUT0", x0)
+                  .input_from_numpy("INPUT1", x1)
+                  .infer(client))
+        np.testing.assert_array_equal(result.as_numpy("OUTPUT1"), x0 - x1)
+    finally:
+        client.close()
+
+
+def test_builder_resolves_aio_client_classes():
+    """The builder's protocol resolution also covers the aio clients
+    (module path contains .http/.grpc)."""
+    import asyncio
+
+    from client_amd._builder import _io_classes_for
+    from client_amd.http import InferInput as HttpInput
+    from client_amd.http.aio import InferenceServerClient as AioHttp
+
+    async def make():
+        return AioHttp("127.0.0.1:1")
+
+    client = asyncio.new_event_loop().run_until_complete(make())
+    try:
+        input_cls, _ = _io_classes_for(client)
+        assert input_cls is HttpInput
+    finally:
+        pass
+
+
+def test_builder_shared_memory_bindings():
+    b = (InferRequestBuilder("m")
+         .shared_memory_input("IN", "region_a", 1024, [4, 64], "FP32",
+                              offset=128)
+         .shared_memory_output("OUT", "region_b", 2048))
+    from client_amd import http as httpclient
+
+    class FakeHttpClient(httpclient.InferenceServerClient):
+        protocol = "http"  # builder resolution hint (module is the
+        #                    test file, not client_amd.http)
+
+        def __init__(self):  # no network
+            pass
+
+    inputs, outputs, kwargs = b.build(FakeHttpClient())
+    t = inputs[0]._get_tensor()
+    assert t["parameters"]["shared_memory_region"] == "region_a"
+    assert t["parameters"]["shared_memory_offset"] == 128
+    assert t["shape"] == [4, 64]

@@ -219,3 +219,22 @@ def test_perf_cli_periodic(grpc_fixture_server, tmp_path, capsys):
     assert len(lines) == 3
     captured = capsys.readouterr()
     assert "Ramp concurrency: 1" in captured.out
+
+
+def test_perf_wildcard_dim_warning(grpc_fixture_server, capsys):
+    """Defaulted NON-batch wildcard dims must be announced (a silent
+    seq16 once masqueraded as a seq128 BERT measurement)."""
+    from client_amd.server.models import IdentityModel
+
+    host, port, core = grpc_fixture_server
+    wild = IdentityModel(name="identity_wild")
+    wild.inputs = [("INPUT0", "FP32", [-1, -1])]
+    wild.outputs = [("OUTPUT0", "FP32", [-1, -1])]
+    core.add_model(wild)
+    pa = PerfAnalyzer(
+        url=f"{host}:{port}", protocol="grpc", model_name="identity_wild",
+        batch_size=2,
+    )
+    pa.run([1], warmup_s=0.05, window_s=0.1, max_windows=1)
+    err = capsys.readouterr().err
+    assert "wildcard dims" in err and "--shape" in err
