@@ -49,16 +49,22 @@ def _decode_linear(lin, h):
     return lin(h)
 
 
-def _gqa_decode_attention(q, k_all, v_all, mask, rep):
+def _gqa_decode_attention(q, k_all, v_all, mask, rep, scaled=False):
     """q [b, H, 1, d] (H = kv*rep, head h = g*rep + r); k_all/v_all
-    [b, kv, L, d]; mask additive [b, 1, 1, L], PRE-CAST to q.dtype
-    (the per-layer cast was 32 extra launches per decode step)."""
+    [b, kv, L, d] (strided cache views — the 4-D matmul batches over
+    them WITHOUT materializing); mask additive [b, 1, 1, L] in
+    q.dtype. With ``scaled`` the 1/sqrt(d) factor was folded into q by
+    the RoPE kernel (free FLOPs during the rotation), and softmax runs
+    directly on bf16 (fp32 internal accumulation in ATen) — per layer
+    4 kernels (matmul, add, softmax, matmul) instead of 7."""
     b, H, _, d = q.shape
     kv = H // rep
     qg = q.view(b, kv, rep, d)
     scores = torch.matmul(qg, k_all.transpose(-1, -2))  # [b,kv,rep,L]
-    scores = scores * (d ** -0.5) + mask  # [b,1,1,L] broadcast
-    attn = torch.softmax(scores.float(), dim=-1).to(q.dtype)
+    if not scaled:
+        scores = scores * (d ** -0.5)
+    scores = scores + mask  # [b,1,1,L] broadcast
+    attn = torch.softmax(scores, dim=-1)
     out = torch.matmul(attn, v_all)  # [b, kv, rep, d]
     return out.view(b, H, 1, d)
 
@@ -310,8 +316,11 @@ class LlamaModel(nn.Module):
                     sin.data_ptr(), pos_rows.data_ptr(), b,
                     self.cfg.n_heads, self.cfg.n_kv_heads,
                     block.head_dim, ck.shape[2],
-                    torch.cuda.current_stream().cuda_stream,
+                    q_scale=block.head_dim ** -0.5,
+                    stream_handle=(
+                        torch.cuda.current_stream().cuda_stream),
                 )
+                q_scaled = True
             else:
                 def rope_rows(t):
                     t1, t2 = t[..., 0::2], t[..., 1::2]
@@ -322,6 +331,7 @@ class LlamaModel(nn.Module):
 
                 q = rope_rows(q)
                 k = rope_rows(k)
+                q_scaled = False
                 # scatter this step's k/v at each row's own position
                 ck[ar, :, pos_rows] = k[:, :, 0]
                 cv[ar, :, pos_rows] = v[:, :, 0]
@@ -331,7 +341,8 @@ class LlamaModel(nn.Module):
             if _GQA_BMM and q.is_cuda:
                 # grouped bmm reads K/V once (no per-query-head
                 # materialization; see _gqa_decode_attention)
-                attn = _gqa_decode_attention(q, k_all, v_all, mask_x, rep)
+                attn = _gqa_decode_attention(q, k_all, v_all, mask_x,
+                                             rep, scaled=q_scaled)
             else:
                 # materialize K/V heads: measured faster than
                 # enable_gqa on ROCm (docs/PERFORMANCE.md)
@@ -340,6 +351,7 @@ class LlamaModel(nn.Module):
                     k_all.repeat_interleave(rep, dim=1),
                     v_all.repeat_interleave(rep, dim=1),
                     attn_mask=mask_x,
+                    scale=1.0 if q_scaled else None,
                 )
             attn = attn.transpose(1, 2).reshape(b, 1, -1)
             x = x + _decode_linear(block.wo, attn)

@@ -388,14 +388,14 @@ static void rope_scatter_decode_bf16(uintptr_t q, uintptr_t k, uintptr_t v,
                                      uintptr_t ck, uintptr_t cv,
                                      uintptr_t cos_tab, uintptr_t sin_tab,
                                      uintptr_t pos, int b, int hq, int hk,
-                                     int d, long cache_len,
+                                     int d, long cache_len, float q_scale,
                                      uintptr_t stream_handle) {
   HIP_CHECK(ca_rope_scatter_decode_bf16(
       reinterpret_cast<void*>(q), reinterpret_cast<const void*>(k),
       reinterpret_cast<const void*>(v), reinterpret_cast<void*>(ck),
       reinterpret_cast<void*>(cv), reinterpret_cast<const void*>(cos_tab),
       reinterpret_cast<const void*>(sin_tab),
-      reinterpret_cast<const void*>(pos), b, hq, hk, d, cache_len,
+      reinterpret_cast<const void*>(pos), b, hq, hk, d, cache_len, q_scale,
       reinterpret_cast<hipStream_t>(stream_handle)));
 }
 
@@ -498,7 +498,7 @@ PYBIND11_MODULE(_hip_c, m) {
         py::arg("k"), py::arg("v"), py::arg("ck"), py::arg("cv"),
         py::arg("cos_tab"), py::arg("sin_tab"), py::arg("pos"), py::arg("b"),
         py::arg("hq"), py::arg("hk"), py::arg("d"), py::arg("cache_len"),
-        py::arg("stream_handle"));
+        py::arg("q_scale") = 1.0f, py::arg("stream_handle") = 0);
   m.def("gather_pack", &gather_pack, py::arg("src"), py::arg("dst"),
         py::arg("elem_size"), py::arg("shape"), py::arg("strides"),
         py::arg("device") = 0, py::arg("sync") = true);

@@ -331,7 +331,7 @@ extern "C" __global__ void rope_scatter_decode_bf16_kernel(
     const uint16_t* __restrict__ v, uint16_t* __restrict__ ck,
     uint16_t* __restrict__ cv, const float* __restrict__ cos_tab,
     const float* __restrict__ sin_tab, const long* __restrict__ pos, int b,
-    int hq, int hk, int d, long cache_len) {
+    int hq, int hk, int d, long cache_len, float q_scale) {
   const int half = d / 2;
   const long total = (long)b * (hq + 2 * hk) * half;
   long i = (long)(blockIdx.x * blockDim.x + threadIdx.x);
@@ -343,16 +343,16 @@ extern "C" __global__ void rope_scatter_decode_bf16_kernel(
     const int head = (int)(rem % (hq + 2 * hk));
     const int row = (int)(rem / (hq + 2 * hk));
     const long p = pos[row];
-    if (head < hq) {  // q: rotate in place
+    if (head < hq) {  // q: rotate in place (+ folded 1/sqrt(d) scale)
       uint16_t* base = q + ((long)row * hq + head) * d;
       const float c = cos_tab[p * half + j];
       const float s = sin_tab[p * half + j];
       const float x1 = __uint_as_float((uint32_t)base[2 * j] << 16);
       const float x2 = __uint_as_float((uint32_t)base[2 * j + 1] << 16);
-      base[2 * j] =
-          __bfloat16_as_ushort(__float2bfloat16(x1 * c - x2 * s));
-      base[2 * j + 1] =
-          __bfloat16_as_ushort(__float2bfloat16(x1 * s + x2 * c));
+      base[2 * j] = __bfloat16_as_ushort(
+          __float2bfloat16((x1 * c - x2 * s) * q_scale));
+      base[2 * j + 1] = __bfloat16_as_ushort(
+          __float2bfloat16((x1 * s + x2 * c) * q_scale));
     } else if (head < hq + hk) {  // k: rotate -> cache
       const int h = head - hq;
       const uint16_t* base = k + ((long)row * hk + h) * d;
@@ -1157,13 +1157,14 @@ extern "C" hipError_t ca_rope_decode_bf16(void* q, void* k,
 extern "C" hipError_t ca_rope_scatter_decode_bf16(
     void* q, const void* k, const void* v, void* ck, void* cv,
     const void* cos_tab, const void* sin_tab, const void* pos, int b,
-    int hq, int hk, int d, long cache_len, hipStream_t stream) {
+    int hq, int hk, int d, long cache_len, float q_scale,
+    hipStream_t stream) {
   long total = (long)b * (hq + 2 * hk) * (d / 2);
   hipLaunchKernelGGL(rope_scatter_decode_bf16_kernel,
                      dim3(ca_grid_for(total)), dim3(256), 0, stream,
                      (uint16_t*)q, (const uint16_t*)k, (const uint16_t*)v,
                      (uint16_t*)ck, (uint16_t*)cv, (const float*)cos_tab,
                      (const float*)sin_tab, (const long*)pos, b, hq, hk, d,
-                     cache_len);
+                     cache_len, q_scale);
   return hipGetLastError();
 }

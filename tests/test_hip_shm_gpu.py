@@ -754,12 +754,13 @@ def test_rope_scatter_decode_kernel():
         out[..., 1::2] = t1 * s + t2 * c
         return out.to(torch.bfloat16)
 
-    q_ref = rope_ref(q)
+    q_ref = (rope_ref(q).float() * 0.5).to(torch.bfloat16)
     k_ref = rope_ref(k)
     hr.rope_scatter_decode_bf16(
         q.data_ptr(), k.data_ptr(), v.data_ptr(), ck.data_ptr(),
         cv.data_ptr(), cos.data_ptr(), sin.data_ptr(), pos.data_ptr(),
-        b, hq, hk, d, clen, torch.cuda.current_stream().cuda_stream)
+        b, hq, hk, d, clen, q_scale=0.5,
+        stream_handle=torch.cuda.current_stream().cuda_stream)
     torch.cuda.synchronize()
     assert torch.equal(q, q_ref)
     ar = torch.arange(b, device="cuda")
